@@ -1,0 +1,35 @@
+"""geomx_amd.kvstore — GeoMX-compatible kvstore factory.
+
+`create("dist_sync")` / `create("dist_async")` mirrors
+python/mxnet/kvstore.py:663 + src/kvstore/kvstore.cc:41.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+from ..config import Config
+from ..topology import Topology
+from .base import KVStoreBase
+from .dist import KVStoreDist
+from .optimizer import OptimizerSpec, ServerOptimizer
+from .wan import TokenBucket, cross_party_bytes
+
+__all__ = ["create", "KVStoreBase", "KVStoreDist", "OptimizerSpec",
+           "ServerOptimizer", "TokenBucket", "cross_party_bytes"]
+
+
+def create(name: str = "dist_sync", cfg: Optional[Config] = None,
+           topo: Optional[Topology] = None, **overrides) -> KVStoreDist:
+    """Create a kvstore.
+
+    name: "dist_sync" (FSA), "dist_async" (MixedSync), or "local".
+    Extra keyword arguments override Config fields (which themselves
+    default from the GeoMX-compatible environment variables).
+    """
+    global_mode = overrides.pop("global_mode", "sharded")
+    if cfg is None:
+        cfg = Config.from_env(**overrides)
+    cfg.mode = name if name != "local" else "dist_sync"
+    cfg.validate()
+    return KVStoreDist(cfg, topo=topo, global_mode=global_mode)

@@ -1,0 +1,477 @@
+"""KVStoreDist: hierarchical (HiPS) parameter server over RCCL/xGMI.
+
+Re-expresses GeoMX's two-level server message flow
+(src/kvstore/kvstore_dist_server.h:1213-1366 DataHandleSyncDefault;
+src/kvstore/kvstore_dist.h:460-720 worker Push_/PullImpl) as nested
+torch.distributed collectives:
+
+  worker push  -> intra-party reduce(SUM) to the party leader
+                  (= N workers' pushes aggregated by the local server)
+  leader tier  -> inter-party exchange on the leader group, optionally
+                  compressed (Bi-Sparse / FP16 / MPQ / DGT) and paced by
+                  the WAN token bucket
+                  (= local servers re-pushing sums to the global server)
+  owner/global -> fused HIP optimizer update on the authoritative fp32
+                  copy (= global server ApplyUpdates), or plain store of
+                  the aggregated gradient when no optimizer is set
+                  (update-on-worker mode, examples/cnn_bsc.py)
+  worker pull  <- leader-group broadcast (optionally pull-compressed,
+                  BSCPullCompress) + intra-party broadcast
+
+Global-tier strategies:
+  * "sharded"    — each key has an owner leader (round-robin = MultiGPS,
+                   kvstore_dist_server.h:1770-1810): gather to owner,
+                   update there, broadcast back on pull.
+  * "replicated" — leaders all_gather the (compressed) party sums and
+                   each replays the identical deterministic update, so
+                   pull needs no WAN traffic. This is the xGMI-native
+                   incast-free strategy (the role TSEngine's relay trees
+                   play in the reference, van.cc:1312-1458).
+
+Synchronization modes:
+  * dist_sync  (FSA)       — both tiers synchronous.
+  * dist_async (MixedSync) — the global tier applies each party's
+    contribution as a SEPARATE sequential optimizer step (the async
+    global server applies every leader push on arrival,
+    kvstore_dist_server.h:1519-1611); DCASGD compensates staleness.
+  * HFA — leaders store party aggregates locally and only every K2-th
+    push syncs globally, transmitting the milestone DELTA
+    (stored - milestone)/P (kvstore_dist_server.h:959-972,1324-1343).
+"""
+
+from __future__ import annotations
+
+import math
+import pickle
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from .. import ops
+from ..config import Config
+from ..topology import Topology, init_topology
+from .base import KVStoreBase
+from .optimizer import OptimizerSpec, ServerOptimizer
+from .wan import TokenBucket, cross_party_bytes
+
+
+class _KeyState:
+    __slots__ = ("shape", "numel", "dtype", "stored", "update_buf",
+                 "bsc_u", "bsc_v", "residual_2bit", "milestone",
+                 "push_count", "owner_party", "dgt_contrib", "dgt_residual")
+
+    def __init__(self, shape, numel, dtype):
+        self.shape = shape
+        self.numel = numel
+        self.dtype = dtype
+        self.stored: Optional[torch.Tensor] = None       # fp32 flat
+        self.update_buf: Optional[torch.Tensor] = None   # fp32 flat (leaders)
+        self.bsc_u: Optional[torch.Tensor] = None        # leader momentum corr
+        self.bsc_v: Optional[torch.Tensor] = None        # leader error accum
+        self.residual_2bit: Optional[torch.Tensor] = None  # worker residual
+        self.milestone: Optional[torch.Tensor] = None    # HFA milestone (leader)
+        self.dgt_contrib: Optional[torch.Tensor] = None  # EWMA chunk contribution
+        self.dgt_residual: Optional[torch.Tensor] = None
+        self.push_count = 0
+        self.owner_party = 0
+
+
+class KVStoreDist(KVStoreBase):
+    def __init__(self, cfg: Config, topo: Optional[Topology] = None,
+                 global_mode: str = "sharded"):
+        cfg.validate()
+        self.cfg = cfg
+        self.topo = topo if topo is not None else init_topology(
+            cfg.num_parties, cfg.party_sizes, cfg.backend, cfg.device)
+        if global_mode not in ("sharded", "replicated"):
+            raise ValueError(global_mode)
+        self.global_mode = global_mode
+        self.keys: Dict[object, _KeyState] = {}
+        self._key_order: List[object] = []
+        self.optimizer: Optional[ServerOptimizer] = None
+        self.compression: Optional[Dict] = None
+        self.wan = TokenBucket(cfg.wan_gbps)
+        self._device = self.topo.device
+
+    # ------------------------------------------------------------------
+    # properties (GeoMX API parity: kvstore.py:501-565)
+    # ------------------------------------------------------------------
+    @property
+    def type(self) -> str:
+        return self.cfg.mode
+
+    @property
+    def rank(self) -> int:
+        return self.topo.rank
+
+    @property
+    def num_workers(self) -> int:
+        return self.topo.num_workers
+
+    @property
+    def num_all_workers(self) -> int:
+        return self.topo.num_all_workers
+
+    @property
+    def is_master_worker(self) -> bool:
+        return self.topo.is_master_worker
+
+    @property
+    def num_parties(self) -> int:
+        return self.topo.num_parties
+
+    def barrier(self):
+        if dist.is_initialized():
+            dist.barrier()
+
+    _barrier = barrier
+
+    # ------------------------------------------------------------------
+    # configuration commands (the reference sends these as in-band server
+    # commands: CommandType, kvstore_dist_server.h:49-52)
+    # ------------------------------------------------------------------
+    def set_optimizer(self, optimizer) -> None:
+        if isinstance(optimizer, OptimizerSpec):
+            spec = optimizer
+        elif isinstance(optimizer, dict):
+            spec = OptimizerSpec(**optimizer)
+        else:
+            raise TypeError("set_optimizer expects OptimizerSpec or dict")
+        # every rank creates it; only leaders/owners apply it (state is lazy)
+        self.optimizer = ServerOptimizer(spec)
+
+    def set_gradient_compression(self, compression_params: Dict) -> None:
+        params = dict(compression_params)
+        ctype = params.get("type")
+        if ctype not in ("2bit", "bsc", "fp16", "mpq", "dgt"):
+            raise ValueError(f"unknown compression type {ctype!r}")
+        if ctype == "2bit":
+            params.setdefault("threshold", self.cfg.threshold)
+        if ctype in ("bsc", "mpq"):
+            params.setdefault("threshold", self.cfg.bsc_ratio)
+            params.setdefault("size_lower_bound", self.cfg.size_lower_bound)
+        self.compression = params
+
+    # ------------------------------------------------------------------
+    # init
+    # ------------------------------------------------------------------
+    def init(self, key, value: torch.Tensor) -> None:
+        if key in self.keys:
+            raise ValueError(f"key {key!r} already initialised")
+        st = _KeyState(tuple(value.shape), value.numel(), value.dtype)
+        st.owner_party = len(self._key_order) % self.topo.num_parties
+        self.keys[key] = st
+        self._key_order.append(key)
+        flat = value.detach().reshape(-1).float().to(self._device)
+        # rank 0's value is authoritative at init (reference: first init wins)
+        if dist.is_initialized() and self.topo.world_size > 1:
+            dist.broadcast(flat, src=0)
+        st.stored = flat.clone()
+
+    def _state(self, key) -> _KeyState:
+        st = self.keys.get(key)
+        if st is None:
+            raise KeyError(f"key {key!r} not initialised")
+        return st
+
+    # ------------------------------------------------------------------
+    # push
+    # ------------------------------------------------------------------
+    def push(self, key, value: torch.Tensor, priority: int = 0) -> None:
+        st = self._state(key)
+        if value.numel() != st.numel:
+            raise ValueError(f"push size mismatch for {key!r}")
+        topo = self.topo
+        st.push_count += 1
+
+        party_sum = self._party_aggregate(st, value)
+
+        if topo.num_parties == 1:
+            if topo.is_leader:
+                agg = party_sum
+                self._apply_global(key, st, [agg])
+            self._leader_cache_sync(st)
+            return
+
+        if self.cfg.use_hfa:
+            self._push_hfa(key, st, party_sum)
+            return
+
+        # inter-party (WAN) tier — leaders only
+        if topo.is_leader:
+            contribs = self._global_exchange_push(key, st, party_sum)
+            if contribs is not None:
+                self._apply_global(key, st, contribs)
+
+    # -- intra-party tier ------------------------------------------------
+    def _party_aggregate(self, st: _KeyState, value: torch.Tensor) -> torch.Tensor:
+        """N workers' pushes -> leader-held sum (local server aggregation,
+        kvstore_dist_server.h:1286-1296)."""
+        topo = self.topo
+        buf = value.detach().reshape(-1).float().to(self._device)
+        if topo.world_size == 1 or topo.num_workers == 1:
+            return buf.clone()
+        ctype = self.compression.get("type") if self.compression else None
+        if ctype == "2bit":
+            # worker-side quantize; leader gathers packed words and
+            # dequantize-sums each (CommDevice::ReduceCompressed analog,
+            # comm.h:545-590)
+            thr = float(self.compression.get("threshold", 0.5))
+            if st.residual_2bit is None:
+                st.residual_2bit = torch.zeros(st.numel, device=self._device)
+            packed = ops.quantize_2bit(buf, st.residual_2bit, thr)
+            gathered = self._party_gather(packed)
+            if topo.is_leader:
+                acc = torch.zeros(st.numel, device=self._device)
+                tmp = torch.empty(st.numel, device=self._device)
+                for p in gathered:
+                    ops.dequantize_2bit(p, st.numel, thr, out=tmp)
+                    acc += tmp
+                return acc
+            return buf  # non-leaders' value unused
+        out = buf.clone()
+        dist.reduce(out, dst=topo.leader_rank, op=dist.ReduceOp.SUM,
+                    group=topo.party_group)
+        return out
+
+    def _party_gather(self, t: torch.Tensor) -> List[torch.Tensor]:
+        topo = self.topo
+        out = [torch.empty_like(t) for _ in range(topo.num_workers)] \
+            if topo.is_leader else None
+        if self.topo.backend == "nccl":
+            # all_gather is universally supported; party links are xGMI
+            out_all = [torch.empty_like(t) for _ in range(topo.num_workers)]
+            dist.all_gather(out_all, t, group=topo.party_group)
+            return out_all
+        dist.gather(t, gather_list=out, dst=topo.leader_rank,
+                    group=topo.party_group)
+        return out if out is not None else []
+
+    # -- global (WAN) tier ----------------------------------------------
+    def _effective_ctype(self, st: _KeyState) -> Optional[str]:
+        if not self.compression:
+            return None
+        ctype = self.compression["type"]
+        if ctype == "mpq":
+            # MPQ: small tensors -> fp16, large -> bsc
+            # (size gate kvstore_dist_server.h:841,879)
+            bound = int(self.compression.get("size_lower_bound",
+                                             self.cfg.size_lower_bound))
+            return "fp16" if st.numel < bound else "bsc"
+        if ctype == "2bit":
+            return None  # 2bit applies to the intra-party tier only
+        return ctype
+
+    def _global_exchange_push(self, key, st: _KeyState,
+                              party_sum: torch.Tensor) -> Optional[List[torch.Tensor]]:
+        """Leader-tier exchange. Returns the list of per-party
+        contributions on ranks that must apply the update (owner in
+        sharded mode / all leaders in replicated), else None."""
+        topo = self.topo
+        P = topo.num_parties
+        group = topo.leader_group
+        ctype = self._effective_ctype(st)
+        ratio = float(self.compression.get("threshold", self.cfg.bsc_ratio)) \
+            if self.compression else self.cfg.bsc_ratio
+
+        if ctype == "bsc":
+            if st.bsc_u is None:
+                st.bsc_u = torch.zeros(st.numel, device=self._device)
+                st.bsc_v = torch.zeros(st.numel, device=self._device)
+            vals, idx = ops.bsc_compress(party_sum, st.bsc_u, st.bsc_v, ratio)
+            payload = vals.numel() * 4 + idx.numel() * 4
+            vlist = [torch.empty_like(vals) for _ in range(P)]
+            ilist = [torch.empty_like(idx) for _ in range(P)]
+            dist.all_gather(vlist, vals, group=group)
+            dist.all_gather(ilist, idx, group=group)
+            self.wan.charge(cross_party_bytes("all_gather", payload, P))
+            dense = []
+            for v_, i_ in zip(vlist, ilist):
+                dense.append(ops.bsc_decompress(v_, i_, st.numel))
+            return dense
+
+        if ctype == "fp16":
+            h = party_sum.to(torch.float16)
+            hlist = [torch.empty_like(h) for _ in range(P)]
+            dist.all_gather(hlist, h, group=group)
+            self.wan.charge(cross_party_bytes("all_gather", h.numel() * 2, P))
+            return [x.float() for x in hlist]
+
+        if ctype == "dgt":
+            contrib_now = self._dgt_transform(st, party_sum)
+            hlist = [torch.empty_like(contrib_now) for _ in range(P)]
+            dist.all_gather(hlist, contrib_now, group=group)
+            nbytes = self._dgt_wire_bytes(st)
+            self.wan.charge(cross_party_bytes("all_gather", nbytes, P))
+            return list(hlist)
+
+        # dense fp32
+        if self.global_mode == "replicated" or self.cfg.mode == "dist_async":
+            flist = [torch.empty_like(party_sum) for _ in range(P)]
+            dist.all_gather(flist, party_sum, group=group)
+            self.wan.charge(cross_party_bytes("all_gather", st.numel * 4, P))
+            return list(flist)
+        # sharded dense: reduce to owner (sum), single contribution
+        out = party_sum.clone()
+        owner_leader = topo.leader_ranks[st.owner_party]
+        dist.reduce(out, dst=owner_leader, op=dist.ReduceOp.SUM, group=group)
+        self.wan.charge(cross_party_bytes("reduce", st.numel * 4, P))
+        if topo.leader_rank == owner_leader and topo.is_leader \
+                and topo.party_id == st.owner_party:
+            return [out]
+        return None
+
+    def _apply_global(self, key, st: _KeyState, contribs: List[torch.Tensor]):
+        """Global-server update (ApplyUpdates, kvstore_dist_server.h:535-559).
+
+        dist_sync: one update on the summed contribution.
+        dist_async: one sequential update PER party contribution (the
+        async server applies each leader push on arrival)."""
+        if self.cfg.mode == "dist_async" and self.optimizer is not None \
+                and len(contribs) > 1:
+            for c in contribs:
+                self.optimizer.update(key, st.stored, c)
+            st.update_buf = None
+            return
+        agg = contribs[0]
+        for c in contribs[1:]:
+            agg = agg + c
+        if self.optimizer is not None:
+            self.optimizer.update(key, st.stored, agg)
+        else:
+            # update-on-worker mode: stored holds the aggregated gradient
+            st.stored = agg if agg.device == self._device else agg.to(self._device)
+
+    def _push_hfa(self, key, st: _KeyState, party_sum: torch.Tensor):
+        """HFA: store party aggregate locally; every K2-th push, exchange
+        milestone deltas (stored-milestone)/P and rebase
+        (kvstore_dist_server.h:959-972,1324-1343)."""
+        topo = self.topo
+        if not topo.is_leader:
+            return
+        st.stored = party_sum
+        if st.milestone is None:
+            st.milestone = torch.zeros_like(st.stored)
+        if st.push_count % self.cfg.hfa_k2 != 0:
+            return
+        P = topo.num_parties
+        delta = (st.stored - st.milestone) / P
+        dist.all_reduce(delta, op=dist.ReduceOp.SUM, group=topo.leader_group)
+        self.wan.charge(cross_party_bytes("all_reduce", st.numel * 4, P))
+        st.stored = st.milestone + delta
+        st.milestone = st.stored.clone()
+
+    # ------------------------------------------------------------------
+    # pull
+    # ------------------------------------------------------------------
+    def pull(self, key, out: torch.Tensor, priority: int = 0) -> None:
+        st = self._state(key)
+        topo = self.topo
+        self._global_exchange_pull(key, st)
+        # intra-party: leader broadcasts authoritative value to its workers
+        if topo.world_size > 1 and topo.num_workers > 1:
+            dist.broadcast(st.stored, src=topo.leader_rank,
+                           group=topo.party_group)
+        result = st.stored.reshape(st.shape).to(out.dtype)
+        with torch.no_grad():
+            out.reshape(st.shape).copy_(result)
+
+    def _global_exchange_pull(self, key, st: _KeyState):
+        """Owner -> leaders distribution of the updated value (pull from
+        global servers, kvstore_dist_server.h:899-1094). No-op in
+        replicated mode / single party / HFA (already rebased)."""
+        topo = self.topo
+        P = topo.num_parties
+        if P == 1 or self.cfg.use_hfa:
+            return
+        ctype = self._effective_ctype(st)
+        need_wire = self.global_mode == "sharded" and not (
+            ctype in ("bsc", "fp16", "dgt") or self.cfg.mode == "dist_async")
+        if not need_wire:
+            return  # all leaders already hold the result (replayed update)
+        if not topo.is_leader:
+            return
+        group = topo.leader_group
+        owner_leader = topo.leader_ranks[st.owner_party]
+        if ctype == "bsc" and self.optimizer is None:
+            # pull-side re-sparsification (BSCPullCompress,
+            # gradient_compression.cc:271-308) — only meaningful for the
+            # sharded aggregated-gradient pull; replicated mode skips the
+            # wire entirely. Kept for the sharded path.
+            cap = ops.ref.bsc_capacity(st.numel,
+                                       float(self.compression["threshold"]), P)
+            if topo.party_id == st.owner_party:
+                vals, idx = ops.bsc_pull_compress(st.stored, cap)
+            else:
+                vals = torch.empty(cap, device=self._device)
+                idx = torch.empty(cap, dtype=torch.int32, device=self._device)
+            dist.broadcast(vals, src=owner_leader, group=group)
+            dist.broadcast(idx, src=owner_leader, group=group)
+            self.wan.charge(cross_party_bytes("broadcast", cap * 8, P))
+            if topo.party_id != st.owner_party:
+                ops.bsc_decompress(vals, idx, st.numel, out=st.stored)
+            return
+        dist.broadcast(st.stored, src=owner_leader, group=group)
+        self.wan.charge(cross_party_bytes("broadcast", st.numel * 4, P))
+
+    def _leader_cache_sync(self, st: _KeyState):
+        # placeholder for party-internal consistency; pull() broadcasts.
+        pass
+
+    # ------------------------------------------------------------------
+    # checkpointing (layout parity: named-param dict + separate optimizer
+    # state blob — gluon/block.py:315,356 + kvstore.py:566-592)
+    # ------------------------------------------------------------------
+    def save_optimizer_states(self, fname: str, dump_optimizer: bool = False):
+        if self.optimizer is None:
+            raise RuntimeError("no optimizer set")
+        blob = self.optimizer.state_dict()
+        if not dump_optimizer:
+            blob = {k: v for k, v in blob.items() if k != "spec"} | \
+                {"spec": blob["spec"]}
+        with open(fname, "wb") as f:
+            pickle.dump(blob, f)
+
+    def load_optimizer_states(self, fname: str):
+        with open(fname, "rb") as f:
+            blob = pickle.load(f)
+        if self.optimizer is None:
+            self.optimizer = ServerOptimizer(OptimizerSpec(**blob["spec"]))
+        self.optimizer.load_state_dict(blob, device=self._device)
+
+    # ------------------------------------------------------------------
+    # DGT transform (priority chunks; lossy low-priority chunks)
+    # ------------------------------------------------------------------
+    def _dgt_transform(self, st: _KeyState, x: torch.Tensor) -> torch.Tensor:
+        """Split into chunks, EWMA contribution, keep top DMLC_K fraction
+        exact; 4-bit-quantize the rest (enable_dgt==3 semantics,
+        van.cc:750-824 + kv_app.h:853-894). Returns the reconstructed
+        (lossy) tensor; _dgt_wire_bytes gives its wire cost."""
+        chunk = max(64, self.cfg.dgt_block_size // 4)
+        contrib = ops.dgt_contribution(x, chunk)
+        if st.dgt_contrib is None or st.dgt_contrib.numel() != contrib.numel():
+            st.dgt_contrib = contrib
+            st.dgt_residual = torch.zeros(st.numel, device=x.device)
+        else:
+            a = self.cfg.dgt_alpha
+            st.dgt_contrib = a * contrib + (1 - a) * st.dgt_contrib
+        nchunks = contrib.numel()
+        n_keep = max(1, int(math.ceil(self.cfg.dgt_k * nchunks)))
+        keep = torch.topk(st.dgt_contrib, n_keep).indices
+        keep_mask = torch.zeros(nchunks, dtype=torch.bool, device=x.device)
+        keep_mask[keep] = True
+        self._dgt_last = (nchunks, n_keep, chunk)
+        out = x.clone()
+        # quantize unimportant chunks (with residual feedback)
+        packed, minmax = ops.quantize_4bit_chunked(x, chunk, st.dgt_residual)
+        deq = ops.dequantize_4bit_chunked(packed, minmax, st.numel, chunk)
+        elem_mask = keep_mask.repeat_interleave(chunk)[:st.numel]
+        out[~elem_mask] = deq.to(out.device)[~elem_mask]
+        return out
+
+    def _dgt_wire_bytes(self, st: _KeyState) -> int:
+        nchunks, n_keep, chunk = self._dgt_last
+        exact = n_keep * chunk * 4
+        lossy = (nchunks - n_keep) * (chunk // 2 + 8)
+        return min(st.numel * 4, exact) + max(0, lossy)

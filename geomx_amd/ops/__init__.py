@@ -1,0 +1,244 @@
+"""Op dispatch: hand-written gfx950 HIP kernels on GPU, torch reference on CPU.
+
+Policy (matches the project contract):
+  - CUDA (ROCm) tensors MUST go through the native `_geops` extension;
+    if the extension is missing on a GPU machine the op RAISES — there
+    is no silent eager fallback on the GPU path.
+  - CPU tensors use the pure-torch reference implementations
+    (`geomx_amd.ops.reference`), which also serve as the golden model
+    for the HIP kernels in tests.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference as ref
+
+_geops = None
+_geops_err: Optional[str] = None
+try:
+    from geomx_amd import _geops as _geops_mod  # built in-tree by setup.py
+    _geops = _geops_mod
+except Exception as e:  # pragma: no cover - exercised only when ext missing
+    _geops_err = repr(e)
+
+
+def native_available() -> bool:
+    return _geops is not None
+
+
+def native_error() -> Optional[str]:
+    return _geops_err
+
+
+def _require_native():
+    if _geops is None:
+        raise RuntimeError(
+            "geomx_amd native extension (_geops) is required for GPU tensors "
+            "but failed to import: %s. Build it with "
+            "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950)."
+            % _geops_err)
+    return _geops
+
+
+def _on_gpu(*tensors) -> bool:
+    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+
+
+# ---------------------------------------------------------------------------
+# 2bit
+# ---------------------------------------------------------------------------
+
+def quantize_2bit(grad: torch.Tensor, residual: torch.Tensor, threshold: float,
+                  out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    n = grad.numel()
+    nw = ref.quantized_words(n)
+    if _on_gpu(grad):
+        g = _require_native()
+        if out is None:
+            out = torch.empty(nw, dtype=torch.int32, device=grad.device)
+        g.quantize_2bit(grad.reshape(-1), residual.reshape(-1), out, threshold)
+        return out
+    packed = ref.quantize_2bit(grad, residual, threshold)
+    if out is not None:
+        out.copy_(packed)
+        return out
+    return packed
+
+
+def dequantize_2bit(packed: torch.Tensor, n: int, threshold: float,
+                    out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if _on_gpu(packed):
+        g = _require_native()
+        if out is None:
+            out = torch.empty(n, dtype=torch.float32, device=packed.device)
+        g.dequantize_2bit(packed, out.reshape(-1), threshold)
+        return out
+    return ref.dequantize_2bit(packed, n, threshold, out=out)
+
+
+# ---------------------------------------------------------------------------
+# Bi-Sparse
+# ---------------------------------------------------------------------------
+
+def bsc_compress(grad, u, v, ratio, momentum=ref.BSC_MOMENTUM, seed=42
+                 ) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _on_gpu(grad):
+        g = _require_native()
+        n = grad.numel()
+        k = ref.bsc_capacity(n, ratio)
+        vals = torch.empty(k, dtype=torch.float32, device=grad.device)
+        idx = torch.empty(k, dtype=torch.int32, device=grad.device)
+        # pass 1: fused momentum correction (u = mu*u + g ; v += u)
+        g.bsc_momentum(grad.reshape(-1), u, v, momentum)
+        # boundary: sampled top-k over |v| (small torch.topk on the sample)
+        boundary = _bsc_boundary_gpu(v, ratio, seed)
+        # pass 2: capacity-bounded select + pack + zero u,v at selected
+        g.bsc_pack(v, u, vals, idx, boundary, ref.BSC_PLACEHOLDER)
+        return vals, idx
+    return ref.bsc_compress(grad, u, v, ratio, momentum, seed)
+
+
+def _bsc_boundary_gpu(v: torch.Tensor, ratio: float, seed: int) -> float:
+    n = v.numel()
+    sample_size = min(ref.bsc_sample_size(n, ratio), n)
+    top_k = max(1, int(sample_size * ratio))
+    # strided sample (deterministic, cheap, avoids a randperm H2D copy)
+    stride = max(1, n // sample_size)
+    sample = v[:: stride][:sample_size].abs()
+    k = min(top_k, sample.numel())
+    return torch.topk(sample, k).values[-1].item()
+
+
+def bsc_pull_compress(x: torch.Tensor, capacity: int
+                      ) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _on_gpu(x):
+        g = _require_native()
+        vals = torch.empty(capacity, dtype=torch.float32, device=x.device)
+        idx = torch.empty(capacity, dtype=torch.int32, device=x.device)
+        g.bsc_pull_pack(x.reshape(-1), vals, idx, ref.BSC_PLACEHOLDER)
+        return vals, idx
+    return ref.bsc_pull_compress(x, capacity)
+
+
+def bsc_decompress(vals, idx, n, out=None, accumulate=False) -> torch.Tensor:
+    if _on_gpu(vals):
+        g = _require_native()
+        if out is None:
+            out = torch.zeros(n, dtype=torch.float32, device=vals.device)
+            g.bsc_unpack(vals, idx, out.reshape(-1), True)
+            return out
+        g.bsc_unpack(vals, idx, out.reshape(-1), accumulate)
+        return out
+    return ref.bsc_decompress(vals, idx, n, out=out, accumulate=accumulate)
+
+
+# ---------------------------------------------------------------------------
+# DGT
+# ---------------------------------------------------------------------------
+
+def dgt_contribution(grad: torch.Tensor, chunk_elems: int) -> torch.Tensor:
+    if _on_gpu(grad):
+        g = _require_native()
+        n = grad.numel()
+        nchunks = (n + chunk_elems - 1) // chunk_elems
+        out = torch.empty(nchunks, dtype=torch.float32, device=grad.device)
+        g.dgt_contribution(grad.reshape(-1), out, chunk_elems)
+        return out
+    return ref.dgt_contribution(grad, chunk_elems)
+
+
+def quantize_4bit_chunked(x: torch.Tensor, chunk_elems: int,
+                          residual: Optional[torch.Tensor] = None):
+    """Per-chunk min/max 4-bit quantization. Returns (packed u8, minmax f32[nchunks,2])."""
+    n = x.numel()
+    nchunks = (n + chunk_elems - 1) // chunk_elems
+    if _on_gpu(x):
+        g = _require_native()
+        packed = torch.empty((n + 1) // 2, dtype=torch.uint8, device=x.device)
+        minmax = torch.empty((nchunks, 2), dtype=torch.float32, device=x.device)
+        res = residual.reshape(-1) if residual is not None else torch.Tensor()
+        g.quantize_4bit(x.reshape(-1), res, packed, minmax, chunk_elems)
+        return packed, minmax
+    packs, mins, maxs = [], [], []
+    f = x.reshape(-1)
+    for c in range(nchunks):
+        sl = f[c * chunk_elems:(c + 1) * chunk_elems]
+        rsl = residual.reshape(-1)[c * chunk_elems:(c + 1) * chunk_elems] \
+            if residual is not None else None
+        p, lo, hi = ref.quantize_4bit(sl, rsl)
+        packs.append(p)
+        mins.append(lo)
+        maxs.append(hi)
+    minmax = torch.tensor([[lo, hi] for lo, hi in zip(mins, maxs)],
+                          dtype=torch.float32)
+    return torch.cat(packs), minmax
+
+
+def dequantize_4bit_chunked(packed: torch.Tensor, minmax: torch.Tensor,
+                            n: int, chunk_elems: int,
+                            out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    nchunks = minmax.shape[0]
+    if _on_gpu(packed):
+        g = _require_native()
+        if out is None:
+            out = torch.empty(n, dtype=torch.float32, device=packed.device)
+        g.dequantize_4bit(packed, minmax, out.reshape(-1), chunk_elems)
+        return out
+    res = []
+    # CPU reference packs each chunk independently (chunk_elems even except last)
+    half = (chunk_elems + 1) // 2
+    for c in range(nchunks):
+        lo, hi = minmax[c, 0].item(), minmax[c, 1].item()
+        m = min(chunk_elems, n - c * chunk_elems)
+        res.append(ref.dequantize_4bit(packed[c * half:c * half + (m + 1) // 2],
+                                       m, lo, hi))
+    full = torch.cat(res)
+    if out is not None:
+        out.reshape(-1).copy_(full)
+        return out
+    return full
+
+
+# ---------------------------------------------------------------------------
+# Fused optimizer updates
+# ---------------------------------------------------------------------------
+
+def sgd_update(w, g, lr, wd=0.0, rescale=1.0):
+    if _on_gpu(w):
+        _require_native().sgd_update(w.reshape(-1), g.reshape(-1), lr, wd, rescale)
+        return
+    ref.sgd_update(w, g, lr, wd, rescale)
+
+
+def sgd_mom_update(w, g, mom, lr, momentum=0.9, wd=0.0, rescale=1.0):
+    if _on_gpu(w):
+        _require_native().sgd_mom_update(w.reshape(-1), g.reshape(-1),
+                                         mom.reshape(-1), lr, momentum, wd, rescale)
+        return
+    ref.sgd_mom_update(w, g, mom, lr, momentum, wd, rescale)
+
+
+def adam_update(w, g, m, v, t, lr, beta1=0.9, beta2=0.999, eps=1e-8, wd=0.0,
+                rescale=1.0):
+    if _on_gpu(w):
+        _require_native().adam_update(w.reshape(-1), g.reshape(-1),
+                                      m.reshape(-1), v.reshape(-1),
+                                      t, lr, beta1, beta2, eps, wd, rescale)
+        return
+    ref.adam_update(w, g, m, v, t, lr, beta1, beta2, eps, wd, rescale)
+
+
+def dcasgd_update(w, g, prev_w, mom, lr, lamda=0.04, momentum=0.0, wd=0.0,
+                  rescale=1.0):
+    if _on_gpu(w):
+        _require_native().dcasgd_update(
+            w.reshape(-1), g.reshape(-1), prev_w.reshape(-1),
+            mom.reshape(-1) if mom is not None else torch.Tensor(),
+            lr, lamda, momentum, wd, rescale)
+        return
+    ref.dcasgd_update(w, g, prev_w, mom, lr, lamda, momentum, wd, rescale)

@@ -1,0 +1,285 @@
+"""Multi-process CPU (gloo) tests of the hierarchical kvstore.
+
+These cover the distributed semantics the driver can check without a
+GPU: FSA two-tier aggregation, update-on-server vs update-on-worker,
+HFA, Bi-Sparse / FP16 / MPQ global-tier compression, MixedSync
+sequential async updates, and the WAN token-bucket accounting.
+"""
+
+import pytest
+import torch
+
+from dist_helpers import run_dist
+
+from geomx_amd import Config
+from geomx_amd.kvstore import create
+from geomx_amd.kvstore.optimizer import OptimizerSpec
+from geomx_amd.ops import reference as ref
+from geomx_amd.topology import init_topology
+
+
+def _mk(mode="dist_sync", num_parties=1, **over):
+    cfg = Config.from_env(num_parties=num_parties, backend="gloo",
+                          device="cpu", **over)
+    topo = init_topology(cfg.num_parties, cfg.party_sizes, "gloo", "cpu")
+    return create(mode, cfg=cfg, topo=topo)
+
+
+# ---------------------------------------------------------------------------
+# FSA: flat sum semantics (1 party)
+# ---------------------------------------------------------------------------
+
+def _fsa_flat(rank, world):
+    kv = _mk(num_parties=1)
+    torch.manual_seed(0)  # same init on all ranks
+    w = torch.randn(8)
+    kv.init("w", w)
+    g = torch.full((8,), float(rank + 1))
+    kv.push("w", g)
+    out = torch.empty(8)
+    kv.pull("w", out)
+    expected = sum(r + 1 for r in range(world))
+    assert torch.allclose(out, torch.full((8,), float(expected))), out
+
+
+def test_fsa_flat_sum_ws2():
+    run_dist(2, _fsa_flat)
+
+
+# ---------------------------------------------------------------------------
+# FSA: hierarchical 2 parties x 2 workers
+# ---------------------------------------------------------------------------
+
+def _fsa_hier(rank, world, global_mode):
+    kv = _mk(num_parties=2, **{})
+    kv.global_mode = global_mode
+    w = torch.zeros(6)
+    kv.init("w", w)
+    g = torch.full((6,), float(rank + 1))
+    kv.push("w", g)
+    out = torch.empty(6)
+    kv.pull("w", out)
+    expected = sum(r + 1 for r in range(world))  # sum over both tiers
+    assert torch.allclose(out, torch.full((6,), float(expected))), \
+        (rank, out, expected)
+
+
+@pytest.mark.parametrize("global_mode", ["sharded", "replicated"])
+def test_fsa_hierarchical_ws4(global_mode):
+    run_dist(4, _fsa_hier, global_mode)
+
+
+# ---------------------------------------------------------------------------
+# update-on-server: global-server optimizer, workers pull params
+# ---------------------------------------------------------------------------
+
+def _server_sgd(rank, world):
+    kv = _mk(num_parties=2)
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    kv.init("w", torch.ones(5))
+    kv.push("w", torch.full((5,), 1.0))   # sum over 4 workers = 4
+    out = torch.empty(5)
+    kv.pull("w", out)
+    # w = 1 - 0.1*4 = 0.6 on every worker in every party
+    assert torch.allclose(out, torch.full((5,), 0.6), atol=1e-6), (rank, out)
+    # second step from the updated value
+    kv.push("w", torch.full((5,), 0.5))   # sum = 2
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((5,), 0.4), atol=1e-6), (rank, out)
+
+
+def test_update_on_server_sharded_ws4():
+    run_dist(4, _server_sgd)
+
+
+# ---------------------------------------------------------------------------
+# Bi-Sparse over the WAN tier (update-on-worker)
+# ---------------------------------------------------------------------------
+
+def _bsc_push_pull(rank, world):
+    kv = _mk(num_parties=2)
+    kv.set_gradient_compression({"type": "bsc", "threshold": 0.05})
+    n = 2000
+    kv.init("w", torch.zeros(n))
+    torch.manual_seed(7)  # identical grads on all ranks for predictability
+    g = torch.randn(n)
+    kv.push("w", g)
+    out = torch.empty(n)
+    kv.pull("w", out)
+    # Aggregated gradient: each party's sum (2g) is bsc-compressed; with
+    # identical inputs both parties send the same top-k values, so the
+    # result is 2*sum_of_parties for sent coords = 4g at sent coords.
+    sent = out != 0
+    assert sent.sum() > 0
+    assert torch.allclose(out[sent], 4 * g[sent], atol=1e-5)
+    # sparsity: roughly capacity-bound (2 parties x k each, overlapping)
+    k = ref.bsc_capacity(n, 0.05)
+    assert sent.sum() <= 2 * k
+
+
+def test_bsc_ws4():
+    run_dist(4, _bsc_push_pull)
+
+
+# ---------------------------------------------------------------------------
+# FP16 global tier
+# ---------------------------------------------------------------------------
+
+def _fp16(rank, world):
+    kv = _mk(num_parties=2)
+    kv.set_gradient_compression({"type": "fp16"})
+    kv.init("w", torch.zeros(64))
+    g = torch.full((64,), 0.25)  # exactly representable in fp16
+    kv.push("w", g)
+    out = torch.empty(64)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((64,), 0.25 * world), atol=1e-3)
+
+
+def test_fp16_ws4():
+    run_dist(4, _fp16)
+
+
+# ---------------------------------------------------------------------------
+# MPQ: size gate routes small->fp16, large->bsc
+# ---------------------------------------------------------------------------
+
+def _mpq(rank, world):
+    kv = _mk(num_parties=2)
+    kv.set_gradient_compression(
+        {"type": "mpq", "threshold": 0.05, "size_lower_bound": 100})
+    kv.init("small", torch.zeros(10))
+    kv.init("large", torch.zeros(4000))
+    kv.push("small", torch.full((10,), 0.5))
+    torch.manual_seed(3)
+    g = torch.randn(4000)
+    kv.push("large", g)
+    small = torch.empty(10)
+    large = torch.empty(4000)
+    kv.pull("small", small)
+    kv.pull("large", large)
+    # small went fp16 dense: exact sum
+    assert torch.allclose(small, torch.full((10,), 0.5 * world), atol=1e-3)
+    # large went bsc: sparse
+    assert (large == 0).sum() > 2000
+
+
+def test_mpq_ws4():
+    run_dist(4, _mpq)
+
+
+# ---------------------------------------------------------------------------
+# HFA: K2 gating + milestone delta rebase (model averaging)
+# ---------------------------------------------------------------------------
+
+def _hfa(rank, world):
+    kv = _mk(num_parties=2, use_hfa=True, hfa_k2=2)
+    kv.init("w", torch.zeros(4))
+    # HFA pushes param/num_local_workers (model averaging). Party 0
+    # workers hold 2.0, party 1 workers hold 6.0.
+    party = 0 if rank < 2 else 1
+    local_param = torch.full((4,), 2.0 if party == 0 else 6.0)
+    nloc = 2
+    out = torch.empty(4)
+
+    # push 1: no global sync (1 % K2 != 0) -> pull returns party average
+    kv.push("w", local_param / nloc)
+    kv.pull("w", out)
+    assert torch.allclose(out, local_param), (rank, out)
+
+    # push 2: global sync -> average across parties = 4.0
+    kv.push("w", local_param / nloc)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((4,), 4.0)), (rank, out)
+
+
+def test_hfa_ws4():
+    run_dist(4, _hfa)
+
+
+# ---------------------------------------------------------------------------
+# MixedSync (dist_async): sequential per-party optimizer updates
+# ---------------------------------------------------------------------------
+
+def _mixed_async(rank, world):
+    kv = _mk("dist_async", num_parties=2)
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    kv.init("w", torch.ones(3))
+    kv.push("w", torch.full((3,), 1.0))
+    out = torch.empty(3)
+    kv.pull("w", out)
+    # two sequential updates of party sums (2.0 each): w = 1 - .1*2 - .1*2
+    assert torch.allclose(out, torch.full((3,), 0.6), atol=1e-6), (rank, out)
+
+
+def test_mixed_async_ws4():
+    run_dist(4, _mixed_async)
+
+
+# ---------------------------------------------------------------------------
+# DCASGD delay compensation runs under dist_async
+# ---------------------------------------------------------------------------
+
+def _dcasgd(rank, world):
+    kv = _mk("dist_async", num_parties=2)
+    kv.set_optimizer(OptimizerSpec(name="dcasgd", lr=0.01))
+    kv.init("w", torch.ones(4))
+    for _ in range(3):
+        kv.push("w", torch.full((4,), 0.5))
+        out = torch.empty(4)
+        kv.pull("w", out)
+    assert torch.isfinite(out).all()
+    # all ranks agree
+    import torch.distributed as dist
+    ref_out = out.clone()
+    dist.broadcast(ref_out, src=0)
+    assert torch.allclose(out, ref_out)
+
+
+def test_dcasgd_async_ws4():
+    run_dist(4, _dcasgd)
+
+
+# ---------------------------------------------------------------------------
+# WAN token bucket accounting
+# ---------------------------------------------------------------------------
+
+def _wan_accounting(rank, world):
+    kv = _mk(num_parties=2, wan_gbps=100.0)
+    n = 1000
+    kv.init("w", torch.zeros(n))
+    kv.push("w", torch.ones(n))
+    out = torch.empty(n)
+    kv.pull("w", out)
+    if kv.topo.is_leader:
+        assert kv.wan.total_bytes > 0
+    # dense fsa sharded: push reduce (n*4*(1/2)) + pull broadcast (n*4*1)
+    assert torch.allclose(out, torch.full((n,), 4.0))
+
+
+def test_wan_accounting_ws4():
+    run_dist(4, _wan_accounting)
+
+
+# ---------------------------------------------------------------------------
+# 2bit intra-party compression
+# ---------------------------------------------------------------------------
+
+def _2bit(rank, world):
+    kv = _mk(num_parties=1)
+    kv.set_gradient_compression({"type": "2bit", "threshold": 0.5})
+    n = 100
+    kv.init("w", torch.zeros(n))
+    g = torch.full((n,), 0.7)  # above threshold -> each worker emits +0.5
+    kv.push("w", g)
+    out = torch.empty(n)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((n,), 0.5 * world)), (rank, out)
+    # residual keeps the remainder (0.2) per worker
+    kv.push("w", torch.full((n,), 0.4))  # residual .2+.4=.6 -> emit .5
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((n,), 0.5 * world)), (rank, out)
+
+
+def test_2bit_ws2():
+    run_dist(2, _2bit)

@@ -1,0 +1,112 @@
+"""Single-process kvstore semantics (world_size=1, no process group)."""
+
+import pytest
+import torch
+
+import geomx_amd
+from geomx_amd import Config
+from geomx_amd.kvstore import create
+from geomx_amd.kvstore.optimizer import OptimizerSpec
+from geomx_amd.ops import reference as ref
+
+
+def make_kv(mode="dist_sync", **over):
+    cfg = Config.from_env(**over)
+    return create(mode, cfg=cfg)
+
+
+def test_init_push_pull_grad_mode():
+    kv = make_kv()
+    w = torch.randn(4, 5)
+    kv.init("w", w)
+    g = torch.randn(4, 5)
+    kv.push("w", g)
+    out = torch.empty(4, 5)
+    kv.pull("w", out)
+    # no optimizer: pull returns the aggregated gradient
+    assert torch.allclose(out, g, atol=1e-6)
+
+
+def test_update_on_server_sgd():
+    kv = make_kv()
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    w0 = torch.ones(10)
+    kv.init(0, w0)
+    g = torch.full((10,), 2.0)
+    kv.push(0, g)
+    out = torch.empty(10)
+    kv.pull(0, out)
+    assert torch.allclose(out, torch.full((10,), 0.8))
+
+
+def test_update_on_server_adam_matches_reference():
+    kv = make_kv()
+    kv.set_optimizer(OptimizerSpec(name="adam", lr=0.01))
+    torch.manual_seed(0)
+    w0 = torch.randn(33)
+    kv.init("k", w0)
+    w = w0.clone()
+    m = torch.zeros(33)
+    v = torch.zeros(33)
+    for t in range(1, 5):
+        g = torch.randn(33)
+        kv.push("k", g)
+        ref.adam_update(w, g, m, v, t, lr=0.01)
+    out = torch.empty(33)
+    kv.pull("k", out)
+    assert torch.allclose(out, w, atol=1e-6)
+
+
+def test_api_properties():
+    kv = make_kv()
+    assert kv.rank == 0
+    assert kv.num_workers == 1
+    assert kv.num_all_workers == 1
+    assert kv.is_master_worker
+    assert kv.type == "dist_sync"
+    kv.barrier()  # no-op single process
+
+
+def test_duplicate_init_raises():
+    kv = make_kv()
+    kv.init("a", torch.ones(3))
+    with pytest.raises(ValueError):
+        kv.init("a", torch.ones(3))
+
+
+def test_push_unknown_key_raises():
+    kv = make_kv()
+    with pytest.raises(KeyError):
+        kv.push("nope", torch.ones(3))
+
+
+def test_optimizer_state_checkpoint(tmp_path):
+    kv = make_kv()
+    kv.set_optimizer(OptimizerSpec(name="adam", lr=0.01))
+    kv.init("k", torch.randn(16))
+    for _ in range(3):
+        kv.push("k", torch.randn(16))
+    f = tmp_path / "opt.bin"
+    kv.save_optimizer_states(str(f), dump_optimizer=True)
+
+    kv2 = make_kv()
+    kv2.load_optimizer_states(str(f))
+    assert kv2.optimizer.spec.name == "adam"
+    assert kv2.optimizer.step_count["k"] == 3
+    st1 = kv.optimizer.state["k"]
+    st2 = kv2.optimizer.state["k"]
+    for name in st1:
+        assert torch.allclose(st1[name].cpu(), st2[name].cpu())
+
+
+def test_set_gradient_compression_validation():
+    kv = make_kv()
+    kv.set_gradient_compression({"type": "bsc", "threshold": 0.02})
+    assert kv.compression["threshold"] == 0.02
+    with pytest.raises(ValueError):
+        kv.set_gradient_compression({"type": "nope"})
+
+
+def test_mx_kv_alias():
+    kv = geomx_amd.kv.create("dist_sync")
+    assert kv.type == "dist_sync"
