@@ -1,0 +1,179 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: CNN training samples/sec on 1..8 MI355X GPUs.
+
+Measures the BASELINE.json headline metric — samples/sec (whole node)
+for the GeoMX example CNN (examples/cnn.py:56-63 architecture) on
+synthetic 3x224x224 data, random-init weights, bf16 autocast compute.
+
+Modes:
+  --mode flat  (default) : bucketed all_reduce data parallelism — the
+                           baseline the HiPS speedup is measured against
+  --mode hips            : two-tier HiPS (party tier + leader/WAN tier),
+                           with --parties, --compress, --wan-gbps
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W           (single GPU)
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as distmod
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from geomx_amd import Config  # noqa: E402
+from geomx_amd.kvstore.optimizer import OptimizerSpec  # noqa: E402
+from geomx_amd.models import create_model  # noqa: E402
+from geomx_amd.parallel import GeoTrainer  # noqa: E402
+from geomx_amd.topology import init_topology  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=256,
+                   help="per-GPU batch (weak scaling)")
+    p.add_argument("--model", type=str, default="geomx_cnn")
+    p.add_argument("--image-size", type=int, default=224)
+    p.add_argument("--num-classes", type=int, default=10)
+    p.add_argument("--mode", type=str, default="flat", choices=["flat", "hips"])
+    p.add_argument("--parties", type=int, default=0,
+                   help="HiPS party count (default: world_size//2, min 2)")
+    p.add_argument("--compress", type=str, default=None,
+                   choices=[None, "bsc", "fp16", "mpq", "2bit", "dgt"])
+    p.add_argument("--bsc-ratio", type=float, default=0.01)
+    p.add_argument("--wan-gbps", type=float, default=0.0)
+    p.add_argument("--bucket-mb", type=int, default=25)
+    p.add_argument("--optimizer", type=str, default="sgd_mom")
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--json-out", type=str, default=None)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+
+    use_cuda = torch.cuda.is_available()
+    device_str = "cuda" if use_cuda else "cpu"
+    backend = "nccl" if use_cuda else "gloo"
+
+    if args.mode == "hips" and world > 1:
+        parties = args.parties or max(2, world // 4)
+        if world % parties:
+            parties = 1
+    else:
+        parties = 1
+
+    cfg = Config.from_env(
+        num_parties=parties, backend=backend,
+        compression=args.compress, bsc_ratio=args.bsc_ratio,
+        wan_gbps=args.wan_gbps, bucket_mb=args.bucket_mb)
+    topo = init_topology(parties, None, backend)
+    device = topo.device
+
+    # fail loudly if the native extension is missing on a GPU machine
+    if use_cuda:
+        from geomx_amd import ops
+        if not ops.native_available():
+            raise RuntimeError("GPU run without native _geops extension: "
+                               + str(ops.native_error()))
+
+    torch.manual_seed(1234)  # same random init on all ranks
+    model = create_model(args.model, image_size=args.image_size,
+                         num_classes=args.num_classes).to(device)
+    spec = OptimizerSpec(name=args.optimizer, lr=0.01, momentum=0.9)
+    trainer = GeoTrainer(model, cfg, topo, spec, mode=args.mode)
+
+    bs = args.batch_size
+    x = torch.randn(bs, 3, args.image_size, args.image_size, device=device)
+    y = torch.randint(0, args.num_classes, (bs,), device=device)
+
+    amp_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    amp = args.dtype == "bf16"
+
+    def one_step():
+        with torch.autocast(device_type=device_str, dtype=amp_dtype,
+                            enabled=amp):
+            out = model(x)
+            loss = torch.nn.functional.cross_entropy(out, y)
+        trainer.zero_grad()
+        loss.backward()
+        trainer.step()
+        return loss
+
+    for _ in range(args.warmup):
+        one_step()
+
+    if distmod.is_initialized():
+        distmod.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if distmod.is_initialized():
+        distmod.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if distmod.is_initialized() and world > 1:
+        t = torch.tensor([elapsed], device=device if backend == "nccl" else "cpu")
+        distmod.all_reduce(t, op=distmod.ReduceOp.MAX)
+        elapsed = t.item()
+
+    samples = world * bs * args.steps
+    sps = samples / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        result = {
+            "metric": "samples/sec (whole node)",
+            "value": round(sps, 2),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "input": f"3x{args.image_size}x{args.image_size}",
+                "global_batch": world * bs,
+                "per_gpu_batch": bs,
+                "parallelism": (f"hips{topo.num_parties}x"
+                                f"{world // max(1, topo.num_parties)}"
+                                if args.mode == "hips" else f"dp{world}"),
+                "compression": args.compress,
+                "wan_gbps": args.wan_gbps,
+                "optimizer": args.optimizer,
+            },
+        }
+        line = json.dumps(result)
+        print(line)
+        if args.json_out:
+            with open(args.json_out, "w") as f:
+                f.write(line + "\n")
+
+    if distmod.is_initialized():
+        distmod.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

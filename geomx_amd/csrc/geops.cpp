@@ -1,0 +1,228 @@
+// Torch bindings for the geomx_amd gfx950 kernel library (kernels.hip).
+//
+// Thin checked wrappers: dtype/contiguity/device checks, stream plumbing
+// (kernels run on the current torch stream so they order correctly with
+// autograd/backward and RCCL collectives), and workspace management for
+// the Bi-Sparse pack pipeline.
+
+#include <torch/extension.h>
+
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+// C-ABI launchers from kernels.hip
+extern "C" {
+void geops_quantize_2bit(const float*, float*, uint32_t*, long long, float,
+                         hipStream_t);
+void geops_dequantize_2bit(const uint32_t*, float*, long long, float,
+                           hipStream_t);
+void geops_bsc_momentum(const float*, float*, float*, float, long long,
+                        hipStream_t);
+void geops_bsc_pack(const float*, float*, float*, float*, int*, long long*,
+                    float, long long, long long, float, bool, hipStream_t);
+void geops_bsc_pull_pack(const float*, float*, int*, long long*, long long,
+                         long long, float, hipStream_t);
+void geops_bsc_unpack(const float*, const int*, float*, long long, bool,
+                      hipStream_t);
+void geops_dgt_contribution(const float*, float*, long long, int, int,
+                            hipStream_t);
+void geops_quantize_4bit(const float*, float*, uint8_t*, float*, long long,
+                         int, int, bool, hipStream_t);
+void geops_dequantize_4bit(const uint8_t*, const float*, float*, long long,
+                           int, hipStream_t);
+void geops_sgd_update(float*, const float*, float, float, float, long long,
+                      hipStream_t);
+void geops_sgd_mom_update(float*, const float*, float*, float, float, float,
+                          float, long long, hipStream_t);
+void geops_adam_update(float*, const float*, float*, float*, float, float,
+                       float, float, float, float, long long, hipStream_t);
+void geops_dcasgd_update(float*, const float*, float*, float*, float, float,
+                         float, float, float, long long, bool, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(((uintptr_t)t.data_ptr() & 15) == 0, name,
+              " must be 16-byte aligned");
+}
+
+void quantize_2bit(torch::Tensor grad, torch::Tensor residual,
+                   torch::Tensor out, double thr) {
+  check_f32(grad, "grad");
+  check_f32(residual, "residual");
+  TORCH_CHECK(out.scalar_type() == torch::kInt32 && out.is_contiguous());
+  const long long n = grad.numel();
+  TORCH_CHECK(residual.numel() == n);
+  TORCH_CHECK(out.numel() == (n + 15) / 16);
+  geops_quantize_2bit(grad.data_ptr<float>(), residual.data_ptr<float>(),
+                      (uint32_t*)out.data_ptr<int32_t>(), n, (float)thr,
+                      cur_stream());
+}
+
+void dequantize_2bit(torch::Tensor packed, torch::Tensor out, double thr) {
+  check_f32(out, "out");
+  TORCH_CHECK(packed.scalar_type() == torch::kInt32 && packed.is_contiguous());
+  const long long n = out.numel();
+  TORCH_CHECK(packed.numel() >= (n + 15) / 16);
+  geops_dequantize_2bit((const uint32_t*)packed.data_ptr<int32_t>(),
+                        out.data_ptr<float>(), n, (float)thr, cur_stream());
+}
+
+void bsc_momentum(torch::Tensor g, torch::Tensor u, torch::Tensor v,
+                  double mu) {
+  check_f32(g, "g"); check_f32(u, "u"); check_f32(v, "v");
+  const long long n = g.numel();
+  TORCH_CHECK(u.numel() == n && v.numel() == n);
+  geops_bsc_momentum(g.data_ptr<float>(), u.data_ptr<float>(),
+                     v.data_ptr<float>(), (float)mu, n, cur_stream());
+}
+
+torch::Tensor make_workspace(const torch::Tensor& like) {
+  return torch::empty({2049}, torch::TensorOptions()
+                                  .dtype(torch::kInt64)
+                                  .device(like.device()));
+}
+
+void bsc_pack(torch::Tensor v, torch::Tensor u, torch::Tensor vals,
+              torch::Tensor idx, double boundary, double placeholder) {
+  check_f32(v, "v"); check_f32(u, "u"); check_f32(vals, "vals");
+  TORCH_CHECK(idx.scalar_type() == torch::kInt32 && idx.is_contiguous());
+  TORCH_CHECK(vals.numel() == idx.numel());
+  auto ws = make_workspace(v);
+  geops_bsc_pack(v.data_ptr<float>(), v.data_ptr<float>(),
+                 u.data_ptr<float>(), vals.data_ptr<float>(),
+                 idx.data_ptr<int32_t>(), (long long*)ws.data_ptr<int64_t>(),
+                 (float)boundary, v.numel(), vals.numel(),
+                 (float)placeholder, /*zero_uv=*/true, cur_stream());
+}
+
+void bsc_pull_pack(torch::Tensor x, torch::Tensor vals, torch::Tensor idx,
+                   double placeholder) {
+  check_f32(x, "x"); check_f32(vals, "vals");
+  TORCH_CHECK(idx.scalar_type() == torch::kInt32 && idx.is_contiguous());
+  auto ws = make_workspace(x);
+  geops_bsc_pull_pack(x.data_ptr<float>(), vals.data_ptr<float>(),
+                      idx.data_ptr<int32_t>(), (long long*)ws.data_ptr<int64_t>(),
+                      x.numel(), vals.numel(), (float)placeholder,
+                      cur_stream());
+}
+
+void bsc_unpack(torch::Tensor vals, torch::Tensor idx, torch::Tensor out,
+                bool accumulate) {
+  check_f32(vals, "vals"); check_f32(out, "out");
+  TORCH_CHECK(idx.scalar_type() == torch::kInt32 && idx.is_contiguous());
+  if (!accumulate) {
+    hipMemsetAsync(out.data_ptr<float>(), 0, out.numel() * sizeof(float),
+                   cur_stream());
+  }
+  geops_bsc_unpack(vals.data_ptr<float>(), idx.data_ptr<int32_t>(),
+                   out.data_ptr<float>(), vals.numel(), accumulate,
+                   cur_stream());
+}
+
+void dgt_contribution(torch::Tensor g, torch::Tensor out, int64_t chunk) {
+  check_f32(g, "g"); check_f32(out, "out");
+  const long long n = g.numel();
+  const int nchunks = (int)((n + chunk - 1) / chunk);
+  TORCH_CHECK(out.numel() == nchunks);
+  geops_dgt_contribution(g.data_ptr<float>(), out.data_ptr<float>(), n,
+                         (int)chunk, nchunks, cur_stream());
+}
+
+void quantize_4bit(torch::Tensor x, torch::Tensor residual,
+                   torch::Tensor packed, torch::Tensor minmax,
+                   int64_t chunk) {
+  check_f32(x, "x");
+  TORCH_CHECK(packed.scalar_type() == torch::kUInt8 && packed.is_contiguous());
+  check_f32(minmax, "minmax");
+  const long long n = x.numel();
+  const int nchunks = (int)((n + chunk - 1) / chunk);
+  TORCH_CHECK(packed.numel() == (n + 1) / 2);
+  TORCH_CHECK(minmax.numel() == 2 * nchunks);
+  const bool has_res = residual.defined() && residual.numel() > 0;
+  if (has_res) check_f32(residual, "residual");
+  geops_quantize_4bit(x.data_ptr<float>(),
+                      has_res ? residual.data_ptr<float>() : nullptr,
+                      packed.data_ptr<uint8_t>(), minmax.data_ptr<float>(),
+                      n, (int)chunk, nchunks, has_res, cur_stream());
+}
+
+void dequantize_4bit(torch::Tensor packed, torch::Tensor minmax,
+                     torch::Tensor out, int64_t chunk) {
+  check_f32(out, "out"); check_f32(minmax, "minmax");
+  TORCH_CHECK(packed.scalar_type() == torch::kUInt8 && packed.is_contiguous());
+  geops_dequantize_4bit(packed.data_ptr<uint8_t>(), minmax.data_ptr<float>(),
+                        out.data_ptr<float>(), out.numel(), (int)chunk,
+                        cur_stream());
+}
+
+void sgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
+                double rescale) {
+  check_f32(w, "w"); check_f32(g, "g");
+  TORCH_CHECK(w.numel() == g.numel());
+  geops_sgd_update(w.data_ptr<float>(), g.data_ptr<float>(), (float)lr,
+                   (float)wd, (float)rescale, w.numel(), cur_stream());
+}
+
+void sgd_mom_update(torch::Tensor w, torch::Tensor g, torch::Tensor mom,
+                    double lr, double momentum, double wd, double rescale) {
+  check_f32(w, "w"); check_f32(g, "g"); check_f32(mom, "mom");
+  geops_sgd_mom_update(w.data_ptr<float>(), g.data_ptr<float>(),
+                       mom.data_ptr<float>(), (float)lr, (float)momentum,
+                       (float)wd, (float)rescale, w.numel(), cur_stream());
+}
+
+void adam_update(torch::Tensor w, torch::Tensor g, torch::Tensor m,
+                 torch::Tensor v, int64_t t, double lr, double beta1,
+                 double beta2, double eps, double wd, double rescale) {
+  check_f32(w, "w"); check_f32(g, "g"); check_f32(m, "m"); check_f32(v, "v");
+  const double lr_t =
+      lr * std::sqrt(1.0 - std::pow(beta2, (double)t)) /
+      (1.0 - std::pow(beta1, (double)t));
+  geops_adam_update(w.data_ptr<float>(), g.data_ptr<float>(),
+                    m.data_ptr<float>(), v.data_ptr<float>(), (float)lr_t,
+                    (float)beta1, (float)beta2, (float)eps, (float)wd,
+                    (float)rescale, w.numel(), cur_stream());
+}
+
+void dcasgd_update(torch::Tensor w, torch::Tensor g, torch::Tensor prev_w,
+                   torch::Tensor mom, double lr, double lamda,
+                   double momentum, double wd, double rescale) {
+  check_f32(w, "w"); check_f32(g, "g"); check_f32(prev_w, "prev_w");
+  const bool has_mom = mom.defined() && mom.numel() > 0;
+  if (has_mom) check_f32(mom, "mom");
+  geops_dcasgd_update(w.data_ptr<float>(), g.data_ptr<float>(),
+                      prev_w.data_ptr<float>(),
+                      has_mom ? mom.data_ptr<float>() : nullptr, (float)lr,
+                      (float)lamda, (float)momentum, (float)wd,
+                      (float)rescale, w.numel(), has_mom, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "geomx_amd gfx950 kernel library";
+  m.def("quantize_2bit", &quantize_2bit);
+  m.def("dequantize_2bit", &dequantize_2bit);
+  m.def("bsc_momentum", &bsc_momentum);
+  m.def("bsc_pack", &bsc_pack);
+  m.def("bsc_pull_pack", &bsc_pull_pack);
+  m.def("bsc_unpack", &bsc_unpack);
+  m.def("dgt_contribution", &dgt_contribution);
+  m.def("quantize_4bit", &quantize_4bit);
+  m.def("dequantize_4bit", &dequantize_4bit);
+  m.def("sgd_update", &sgd_update);
+  m.def("sgd_mom_update", &sgd_mom_update);
+  m.def("adam_update", &adam_update);
+  m.def("dcasgd_update", &dcasgd_update);
+}

@@ -1,0 +1,634 @@
+// gfx950 (MI355X / CDNA4) kernels for the geomx_amd kvstore hot path.
+//
+// Every kernel here is memory-bound (elementwise / select / pack); the
+// design rules applied (from the CDNA4 programming guide):
+//   - wave64: ballots are 64-bit, lane masks use (1ull<<lane)-1
+//   - vectorized float4 (16B/lane) loads/stores on the streaming paths
+//   - grid-stride loops, grid capped at ~2048 blocks of 256 threads
+//   - wave-level prefix sums via __ballot/__popcll for the pack paths
+//     (replaces the reference's serial CPU scans,
+//      gradient_compression.cc:191-336)
+//
+// Semantics match geomx_amd/ops/reference.py (the golden model used by
+// tests/test_kernels_gpu.py).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+#define GEOPS_THREADS 256
+#define GEOPS_MAX_BLOCKS 2048
+
+static inline int geops_blocks(long long work, int per_thread = 1) {
+  long long b = (work + (long long)GEOPS_THREADS * per_thread - 1) /
+                ((long long)GEOPS_THREADS * per_thread);
+  if (b < 1) b = 1;
+  if (b > GEOPS_MAX_BLOCKS) b = GEOPS_MAX_BLOCKS;
+  return (int)b;
+}
+
+// ---------------------------------------------------------------------------
+// 2bit quantization (gradient_compression-inl.h:40-139 semantics)
+// code: 0b11 -> +thr, 0b10 -> -thr, 0b00 -> 0; 16 codes per uint32 word,
+// slot s occupies bits [2s, 2s+1].
+// ---------------------------------------------------------------------------
+
+// spread 16 bits of x to even bit positions of a 32-bit word
+__device__ __forceinline__ uint32_t spread16(uint32_t x) {
+  x &= 0x0000FFFFu;
+  x = (x | (x << 8)) & 0x00FF00FFu;
+  x = (x | (x << 4)) & 0x0F0F0F0Fu;
+  x = (x | (x << 2)) & 0x33333333u;
+  x = (x | (x << 1)) & 0x55555555u;
+  return x;
+}
+
+extern "C" __global__ void k_quantize_2bit(const float* __restrict__ grad,
+                                           float* __restrict__ residual,
+                                           uint32_t* __restrict__ out,
+                                           long long n, float thr) {
+  const int lane = threadIdx.x & 63;
+  const long long wave_id = ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const long long n_waves = ((long long)gridDim.x * blockDim.x) >> 6;
+  for (long long base = wave_id * 64; base < n; base += n_waves * 64) {
+    const long long i = base + lane;
+    const bool active = i < n;
+    uint32_t b0 = 0, b1 = 0;  // bit0 / bit1 of this lane's code
+    if (active) {
+      float r = residual[i] + grad[i];
+      if (r >= thr) {
+        b0 = 1; b1 = 1;       // 0b11
+        r -= thr;
+      } else if (r <= -thr) {
+        b1 = 1;               // 0b10
+        r += thr;
+      }
+      residual[i] = r;
+    }
+    const uint64_t m0 = __ballot(b0);
+    const uint64_t m1 = __ballot(b1);
+    if ((lane & 15) == 0) {
+      const int j = lane >> 4;  // which 16-lane group
+      const uint32_t s0 = (uint32_t)((m0 >> (j * 16)) & 0xFFFFu);
+      const uint32_t s1 = (uint32_t)((m1 >> (j * 16)) & 0xFFFFu);
+      const long long w = (base >> 4) + j;
+      if (w * 16 < n) out[w] = spread16(s0) | (spread16(s1) << 1);
+    }
+  }
+}
+
+extern "C" __global__ void k_dequantize_2bit(const uint32_t* __restrict__ in,
+                                             float* __restrict__ out,
+                                             long long n, float thr) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < n; i += stride) {
+    const uint32_t word = in[i >> 4];
+    const uint32_t code = (word >> ((i & 15) * 2)) & 3u;
+    out[i] = (code == 3u) ? thr : ((code == 2u) ? -thr : 0.0f);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Bi-Sparse: fused momentum correction (u = mu*u + g ; v += u)
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void k_bsc_momentum(const float* __restrict__ g,
+                                          float* __restrict__ u,
+                                          float* __restrict__ v,
+                                          float mu, long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long n4 = n >> 2;
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  float4* u4 = reinterpret_cast<float4*>(u);
+  float4* v4 = reinterpret_cast<float4*>(v);
+  for (long long i = tid; i < n4; i += stride) {
+    float4 gg = g4[i], uu = u4[i], vv = v4[i];
+    uu.x = uu.x * mu + gg.x; vv.x += uu.x;
+    uu.y = uu.y * mu + gg.y; vv.y += uu.y;
+    uu.z = uu.z * mu + gg.z; vv.z += uu.z;
+    uu.w = uu.w * mu + gg.w; vv.w += uu.w;
+    u4[i] = uu; v4[i] = vv;
+  }
+  for (long long i = (n4 << 2) + tid; i < n; i += stride) {
+    float uu = u[i] * mu + g[i];
+    u[i] = uu;
+    v[i] += uu;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Bi-Sparse pack: 3-phase parallel replacement of the reference's serial
+// capacity-bounded index-order scan (gradient_compression.cc:245-267).
+// Blocks own contiguous chunks so the packed output preserves index order.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ long long chunk_begin(long long n, int nb, int b) {
+  const long long chunk = (n + nb - 1) / nb;
+  long long s = (long long)b * chunk;
+  return s < n ? s : n;
+}
+
+// phase 1: per-block predicate counts
+template <bool NONZERO>
+__global__ void k_bsc_count_t(const float* __restrict__ v, float boundary,
+                              long long n, int nb,
+                              long long* __restrict__ counts) {
+  __shared__ long long lds[GEOPS_THREADS / 64];
+  const int b = blockIdx.x;
+  const long long lo = chunk_begin(n, nb, b), hi = chunk_begin(n, nb, b + 1);
+  long long cnt = 0;
+  for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    const float x = v[i];
+    const bool pred = NONZERO ? (x != 0.0f) : (fabsf(x) >= boundary);
+    cnt += pred ? 1 : 0;
+  }
+  // wave reduce then LDS reduce
+  for (int off = 32; off > 0; off >>= 1)
+    cnt += __shfl_down(cnt, off, 64);
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds[wid] = cnt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    long long total = 0;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) total += lds[w];
+    counts[b] = total;
+  }
+}
+
+// phase 2: single-block exclusive scan of per-block counts (+ total at [nb])
+extern "C" __global__ void k_bsc_scan(long long* __restrict__ counts, int nb) {
+  // nb <= 2048; serial scan by one thread is fine (tiny)
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    long long acc = 0;
+    for (int i = 0; i < nb; ++i) {
+      const long long c = counts[i];
+      counts[i] = acc;
+      acc += c;
+    }
+    counts[nb] = acc;
+  }
+}
+
+// phase 3: ordered pack with capacity bound; optionally zero u,v at the
+// positions actually sent (error feedback, reference :258-260)
+template <bool ZERO_UV>
+__global__ void k_bsc_pack_t(const float* __restrict__ v_in,
+                             float* __restrict__ v_mut,
+                             float* __restrict__ u_mut,
+                             float* __restrict__ vals,
+                             int* __restrict__ idx,
+                             const long long* __restrict__ offsets,
+                             float boundary, long long n, int nb,
+                             long long capacity) {
+  __shared__ long long wave_base[GEOPS_THREADS / 64];
+  __shared__ long long carry;
+  const int b = blockIdx.x;
+  const long long lo = chunk_begin(n, nb, b), hi = chunk_begin(n, nb, b + 1);
+  if (threadIdx.x == 0) carry = offsets[b];
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+  for (long long base = lo; base < hi; base += blockDim.x) {
+    const long long i = base + threadIdx.x;
+    const bool in_range = i < hi;
+    float x = 0.0f;
+    bool pred = false;
+    if (in_range) {
+      x = v_in[i];
+      pred = fabsf(x) >= boundary;
+    }
+    const uint64_t m = __ballot(pred);
+    const long long wave_cnt = __popcll(m);
+    const long long lane_pre = __popcll(m & ((1ull << lane) - 1ull));
+    if (lane == 0) wave_base[wid] = wave_cnt;
+    __syncthreads();
+    // exclusive scan of wave counts + add carry (one thread)
+    if (threadIdx.x == 0) {
+      long long acc = carry;
+      for (int w = 0; w < nwaves; ++w) {
+        const long long c = wave_base[w];
+        wave_base[w] = acc;
+        acc += c;
+      }
+      carry = acc;
+    }
+    __syncthreads();
+    if (pred) {
+      const long long pos = wave_base[wid] + lane_pre;
+      if (pos < capacity) {
+        vals[pos] = x;
+        idx[pos] = (int)i;
+        if (ZERO_UV) {
+          v_mut[i] = 0.0f;
+          u_mut[i] = 0.0f;
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// placeholder fill for unused capacity (reference :262-266)
+extern "C" __global__ void k_bsc_fill_tail(float* __restrict__ vals,
+                                           int* __restrict__ idx,
+                                           const long long* __restrict__ counts,
+                                           int nb, long long capacity,
+                                           float placeholder) {
+  long long total = counts[nb];
+  if (total > capacity) total = capacity;
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = total + tid; i < capacity; i += stride) {
+    vals[i] = placeholder;
+    idx[i] = -1;
+  }
+}
+
+extern "C" __global__ void k_bsc_unpack(const float* __restrict__ vals,
+                                        const int* __restrict__ idx,
+                                        float* __restrict__ out,
+                                        long long k, bool accumulate) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < k; i += stride) {
+    const int j = idx[i];
+    if (j >= 0) {
+      if (accumulate)
+        atomicAdd(out + j, vals[i]);
+      else
+        out[j] = vals[i];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// DGT: per-chunk mean(|g|) contribution (kv_app.h:853-876)
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void k_dgt_contribution(const float* __restrict__ g,
+                                              float* __restrict__ out,
+                                              long long n, int chunk,
+                                              int nchunks) {
+  __shared__ float lds[GEOPS_THREADS / 64];
+  for (int c = blockIdx.x; c < nchunks; c += gridDim.x) {
+    const long long lo = (long long)c * chunk;
+    const long long hi = min(lo + chunk, n);
+    float s = 0.0f;
+    for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x)
+      s += fabsf(g[i]);
+    for (int off = 32; off > 0; off >>= 1)
+      s += __shfl_down(s, off, 64);
+    const int wid = threadIdx.x >> 6;
+    if ((threadIdx.x & 63) == 0) lds[wid] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float total = 0.0f;
+      for (int w = 0; w < (int)(blockDim.x >> 6); ++w) total += lds[w];
+      out[c] = total / (float)(hi - lo);
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 4-bit linear quantization per chunk with residual feedback
+// (van.cc:750-824 semantics; codes are bin midpoints on [min,max])
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void k_minmax_chunk(const float* __restrict__ x,
+                                          const float* __restrict__ residual,
+                                          float* __restrict__ minmax,
+                                          long long n, int chunk, int nchunks,
+                                          bool has_res) {
+  __shared__ float lmin[GEOPS_THREADS / 64];
+  __shared__ float lmax[GEOPS_THREADS / 64];
+  for (int c = blockIdx.x; c < nchunks; c += gridDim.x) {
+    const long long lo = (long long)c * chunk;
+    const long long hi = min(lo + chunk, n);
+    float mn = 3.4e38f, mx = -3.4e38f;
+    for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+      const float f = x[i] + (has_res ? residual[i] : 0.0f);
+      mn = fminf(mn, f);
+      mx = fmaxf(mx, f);
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      mn = fminf(mn, __shfl_down(mn, off, 64));
+      mx = fmaxf(mx, __shfl_down(mx, off, 64));
+    }
+    const int wid = threadIdx.x >> 6;
+    if ((threadIdx.x & 63) == 0) { lmin[wid] = mn; lmax[wid] = mx; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      for (int w = 1; w < (int)(blockDim.x >> 6); ++w) {
+        lmin[0] = fminf(lmin[0], lmin[w]);
+        lmax[0] = fmaxf(lmax[0], lmax[w]);
+      }
+      minmax[2 * c] = lmin[0];
+      minmax[2 * c + 1] = lmax[0];
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" __global__ void k_quantize_4bit(const float* __restrict__ x,
+                                           float* __restrict__ residual,
+                                           uint8_t* __restrict__ out,
+                                           const float* __restrict__ minmax,
+                                           long long n, int chunk,
+                                           bool has_res) {
+  // each thread packs one output byte (2 elements)
+  const long long nbytes = (n + 1) >> 1;
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long b = tid; b < nbytes; b += stride) {
+    uint8_t byte = 0;
+    for (int h = 0; h < 2; ++h) {
+      const long long i = 2 * b + h;
+      if (i >= n) break;
+      const int c = (int)(i / chunk);
+      const float lo = minmax[2 * c];
+      const float span = fmaxf(minmax[2 * c + 1] - lo, 1e-30f);
+      const float step = span / 16.0f;
+      const float f = x[i] + (has_res ? residual[i] : 0.0f);
+      int code = (int)floorf((f - lo) / step);
+      code = code < 0 ? 0 : (code > 15 ? 15 : code);
+      if (has_res) residual[i] = f - (lo + ((float)code + 0.5f) * step);
+      byte |= (uint8_t)(code << (4 * h));
+    }
+    out[b] = byte;
+  }
+}
+
+extern "C" __global__ void k_dequantize_4bit(const uint8_t* __restrict__ in,
+                                             const float* __restrict__ minmax,
+                                             float* __restrict__ out,
+                                             long long n, int chunk) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < n; i += stride) {
+    const int c = (int)(i / chunk);
+    const float lo = minmax[2 * c];
+    const float span = fmaxf(minmax[2 * c + 1] - lo, 1e-30f);
+    const float step = span / 16.0f;
+    const uint8_t byte = in[i >> 1];
+    const int code = (i & 1) ? (byte >> 4) : (byte & 0x0F);
+    out[i] = lo + ((float)code + 0.5f) * step;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused optimizer updates (the server ApplyUpdates hot loop,
+// optimizer_op.cc:43-651 semantics). fp32, float4-vectorized.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void k_sgd_update(float* __restrict__ w,
+                                        const float* __restrict__ g,
+                                        float lr, float wd, float rescale,
+                                        long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long n4 = n >> 2;
+  float4* w4 = reinterpret_cast<float4*>(w);
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  for (long long i = tid; i < n4; i += stride) {
+    float4 ww = w4[i];
+    const float4 gg = g4[i];
+    ww.x -= lr * (gg.x * rescale + wd * ww.x);
+    ww.y -= lr * (gg.y * rescale + wd * ww.y);
+    ww.z -= lr * (gg.z * rescale + wd * ww.z);
+    ww.w -= lr * (gg.w * rescale + wd * ww.w);
+    w4[i] = ww;
+  }
+  for (long long i = (n4 << 2) + tid; i < n; i += stride)
+    w[i] -= lr * (g[i] * rescale + wd * w[i]);
+}
+
+extern "C" __global__ void k_sgd_mom_update(float* __restrict__ w,
+                                            const float* __restrict__ g,
+                                            float* __restrict__ mom,
+                                            float lr, float momentum,
+                                            float wd, float rescale,
+                                            long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long n4 = n >> 2;
+  float4* w4 = reinterpret_cast<float4*>(w);
+  float4* m4 = reinterpret_cast<float4*>(mom);
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  for (long long i = tid; i < n4; i += stride) {
+    float4 ww = w4[i], mm = m4[i];
+    const float4 gg = g4[i];
+    mm.x = mm.x * momentum - lr * (gg.x * rescale + wd * ww.x); ww.x += mm.x;
+    mm.y = mm.y * momentum - lr * (gg.y * rescale + wd * ww.y); ww.y += mm.y;
+    mm.z = mm.z * momentum - lr * (gg.z * rescale + wd * ww.z); ww.z += mm.z;
+    mm.w = mm.w * momentum - lr * (gg.w * rescale + wd * ww.w); ww.w += mm.w;
+    w4[i] = ww; m4[i] = mm;
+  }
+  for (long long i = (n4 << 2) + tid; i < n; i += stride) {
+    const float mm = mom[i] * momentum - lr * (g[i] * rescale + wd * w[i]);
+    mom[i] = mm;
+    w[i] += mm;
+  }
+}
+
+extern "C" __global__ void k_adam_update(float* __restrict__ w,
+                                         const float* __restrict__ g,
+                                         float* __restrict__ m,
+                                         float* __restrict__ v,
+                                         float lr_t, float beta1, float beta2,
+                                         float eps, float wd, float rescale,
+                                         long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long n4 = n >> 2;
+  float4* w4 = reinterpret_cast<float4*>(w);
+  float4* m4 = reinterpret_cast<float4*>(m);
+  float4* v4 = reinterpret_cast<float4*>(v);
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+#define ADAM1(c)                                              \
+  {                                                           \
+    const float gg = gv.c * rescale + wd * wv.c;              \
+    mv.c = beta1 * mv.c + (1.0f - beta1) * gg;                \
+    vv.c = beta2 * vv.c + (1.0f - beta2) * gg * gg;           \
+    wv.c -= lr_t * mv.c / (sqrtf(vv.c) + eps);                \
+  }
+  for (long long i = tid; i < n4; i += stride) {
+    float4 wv = w4[i], mv = m4[i], vv = v4[i];
+    const float4 gv = g4[i];
+    ADAM1(x) ADAM1(y) ADAM1(z) ADAM1(w)
+    w4[i] = wv; m4[i] = mv; v4[i] = vv;
+  }
+#undef ADAM1
+  for (long long i = (n4 << 2) + tid; i < n; i += stride) {
+    const float gg = g[i] * rescale + wd * w[i];
+    const float mm = beta1 * m[i] + (1.0f - beta1) * gg;
+    const float vv = beta2 * v[i] + (1.0f - beta2) * gg * gg;
+    m[i] = mm;
+    v[i] = vv;
+    w[i] -= lr_t * mm / (sqrtf(vv) + eps);
+  }
+}
+
+extern "C" __global__ void k_dcasgd_update(float* __restrict__ w,
+                                           const float* __restrict__ g,
+                                           float* __restrict__ prev_w,
+                                           float* __restrict__ mom,
+                                           float lr, float lamda,
+                                           float momentum, float wd,
+                                           float rescale, long long n,
+                                           bool has_mom) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < n; i += stride) {
+    const float ww = w[i];
+    const float gg = g[i] * rescale;
+    float upd = -lr * (gg + wd * ww + lamda * gg * gg * (ww - prev_w[i]));
+    if (has_mom) {
+      upd = mom[i] * momentum + upd;
+      mom[i] = upd;
+    }
+    prev_w[i] = ww;
+    w[i] = ww + upd;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// C-ABI launchers
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+void geops_quantize_2bit(const float* grad, float* residual, uint32_t* out,
+                         long long n, float thr, hipStream_t s) {
+  hipLaunchKernelGGL(k_quantize_2bit, dim3(geops_blocks(n)),
+                     dim3(GEOPS_THREADS), 0, s, grad, residual, out, n, thr);
+}
+
+void geops_dequantize_2bit(const uint32_t* in, float* out, long long n,
+                           float thr, hipStream_t s) {
+  hipLaunchKernelGGL(k_dequantize_2bit, dim3(geops_blocks(n)),
+                     dim3(GEOPS_THREADS), 0, s, in, out, n, thr);
+}
+
+void geops_bsc_momentum(const float* g, float* u, float* v, float mu,
+                        long long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_bsc_momentum, dim3(geops_blocks(n, 4)),
+                     dim3(GEOPS_THREADS), 0, s, g, u, v, mu, n);
+}
+
+static int bsc_nblocks(long long n) {
+  int nb = (int)((n + 16383) / 16384);
+  if (nb < 1) nb = 1;
+  if (nb > 1024) nb = 1024;
+  return nb;
+}
+
+// workspace: (nb+1) int64 entries
+void geops_bsc_pack(const float* v_in, float* v_mut, float* u_mut,
+                    float* vals, int* idx, long long* workspace,
+                    float boundary, long long n, long long capacity,
+                    float placeholder, bool zero_uv, hipStream_t s) {
+  const int nb = bsc_nblocks(n);
+  hipLaunchKernelGGL((k_bsc_count_t<false>), dim3(nb), dim3(GEOPS_THREADS),
+                     0, s, v_in, boundary, n, nb, workspace);
+  hipLaunchKernelGGL(k_bsc_scan, dim3(1), dim3(64), 0, s, workspace, nb);
+  if (zero_uv)
+    hipLaunchKernelGGL((k_bsc_pack_t<true>), dim3(nb), dim3(GEOPS_THREADS),
+                       0, s, v_in, v_mut, u_mut, vals, idx, workspace,
+                       boundary, n, nb, capacity);
+  else
+    hipLaunchKernelGGL((k_bsc_pack_t<false>), dim3(nb), dim3(GEOPS_THREADS),
+                       0, s, v_in, v_mut, u_mut, vals, idx, workspace,
+                       boundary, n, nb, capacity);
+  hipLaunchKernelGGL(k_bsc_fill_tail, dim3(geops_blocks(capacity)),
+                     dim3(GEOPS_THREADS), 0, s, vals, idx, workspace, nb,
+                     capacity, placeholder);
+}
+
+// pull-side pack of nonzeros: reuse the pack machinery with boundary
+// semantics pred = (x != 0): count<true> + pack with boundary=0 on |x|>0.
+// A dedicated tiny epsilon boundary on |x| >= FLT_MIN would miss true
+// zeros only; we use the NONZERO counter and a pack over |x| >= min_sub.
+void geops_bsc_pull_pack(const float* x, float* vals, int* idx,
+                         long long* workspace, long long n,
+                         long long capacity, float placeholder,
+                         hipStream_t s) {
+  const int nb = bsc_nblocks(n);
+  hipLaunchKernelGGL((k_bsc_count_t<true>), dim3(nb), dim3(GEOPS_THREADS),
+                     0, s, x, 0.0f, n, nb, workspace);
+  hipLaunchKernelGGL(k_bsc_scan, dim3(1), dim3(64), 0, s, workspace, nb);
+  // pred |x| >= FLT_TRUE_MIN  <=>  x != 0 for floats (incl. subnormals)
+  hipLaunchKernelGGL((k_bsc_pack_t<false>), dim3(nb), dim3(GEOPS_THREADS),
+                     0, s, x, nullptr, nullptr, vals, idx, workspace,
+                     1.4e-45f, n, nb, capacity);
+  hipLaunchKernelGGL(k_bsc_fill_tail, dim3(geops_blocks(capacity)),
+                     dim3(GEOPS_THREADS), 0, s, vals, idx, workspace, nb,
+                     capacity, placeholder);
+}
+
+void geops_bsc_unpack(const float* vals, const int* idx, float* out,
+                      long long k, bool accumulate, hipStream_t s) {
+  hipLaunchKernelGGL(k_bsc_unpack, dim3(geops_blocks(k)),
+                     dim3(GEOPS_THREADS), 0, s, vals, idx, out, k,
+                     accumulate);
+}
+
+void geops_dgt_contribution(const float* g, float* out, long long n,
+                            int chunk, int nchunks, hipStream_t s) {
+  int nb = nchunks < GEOPS_MAX_BLOCKS ? nchunks : GEOPS_MAX_BLOCKS;
+  hipLaunchKernelGGL(k_dgt_contribution, dim3(nb), dim3(GEOPS_THREADS),
+                     0, s, g, out, n, chunk, nchunks);
+}
+
+void geops_quantize_4bit(const float* x, float* residual, uint8_t* out,
+                         float* minmax, long long n, int chunk, int nchunks,
+                         bool has_res, hipStream_t s) {
+  int nb = nchunks < GEOPS_MAX_BLOCKS ? nchunks : GEOPS_MAX_BLOCKS;
+  hipLaunchKernelGGL(k_minmax_chunk, dim3(nb), dim3(GEOPS_THREADS), 0, s,
+                     x, residual, minmax, n, chunk, nchunks, has_res);
+  hipLaunchKernelGGL(k_quantize_4bit, dim3(geops_blocks((n + 1) / 2)),
+                     dim3(GEOPS_THREADS), 0, s, x, residual, out, minmax, n,
+                     chunk, has_res);
+}
+
+void geops_dequantize_4bit(const uint8_t* in, const float* minmax, float* out,
+                           long long n, int chunk, hipStream_t s) {
+  hipLaunchKernelGGL(k_dequantize_4bit, dim3(geops_blocks(n)),
+                     dim3(GEOPS_THREADS), 0, s, in, minmax, out, n, chunk);
+}
+
+void geops_sgd_update(float* w, const float* g, float lr, float wd,
+                      float rescale, long long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_sgd_update, dim3(geops_blocks(n, 4)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, lr, wd, rescale, n);
+}
+
+void geops_sgd_mom_update(float* w, const float* g, float* mom, float lr,
+                          float momentum, float wd, float rescale,
+                          long long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_sgd_mom_update, dim3(geops_blocks(n, 4)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, mom, lr, momentum, wd,
+                     rescale, n);
+}
+
+void geops_adam_update(float* w, const float* g, float* m, float* v,
+                       float lr_t, float beta1, float beta2, float eps,
+                       float wd, float rescale, long long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_adam_update, dim3(geops_blocks(n, 4)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, m, v, lr_t, beta1,
+                     beta2, eps, wd, rescale, n);
+}
+
+void geops_dcasgd_update(float* w, const float* g, float* prev_w, float* mom,
+                         float lr, float lamda, float momentum, float wd,
+                         float rescale, long long n, bool has_mom,
+                         hipStream_t s) {
+  hipLaunchKernelGGL(k_dcasgd_update, dim3(geops_blocks(n)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, prev_w, mom, lr, lamda,
+                     momentum, wd, rescale, n, has_mom);
+}
+
+}  // extern "C"

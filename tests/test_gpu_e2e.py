@@ -1,0 +1,85 @@
+"""GPU end-to-end: flagship model steps through the trainer and kvstore
+on one MI355X, exercising the native kernel path."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(autouse=True)
+def _require_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import geomx_amd.ops as ops
+    assert ops.native_available(), ops.native_error()
+
+
+def test_trainer_cnn_step():
+    from geomx_amd import Config
+    from geomx_amd.kvstore.optimizer import OptimizerSpec
+    from geomx_amd.models import create_model
+    from geomx_amd.parallel import GeoTrainer
+    from geomx_amd.topology import init_topology
+
+    dev = torch.device("cuda:0")
+    cfg = Config.from_env()
+    topo = init_topology(1, None, None, "cuda:0")
+    model = create_model("geomx_cnn", image_size=64).to(dev)
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec("adam", lr=1e-3))
+    x = torch.randn(16, 3, 64, 64, device=dev)
+    y = torch.randint(0, 10, (16,), device=dev)
+    losses = []
+    for _ in range(8):
+        with torch.autocast("cuda", torch.bfloat16):
+            loss = torch.nn.functional.cross_entropy(model(x), y)
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+        losses.append(loss.item())
+    # training on a fixed batch must reduce the loss
+    assert losses[-1] < losses[0], losses
+
+
+def test_kvstore_gpu_single():
+    from geomx_amd import Config
+    from geomx_amd.kvstore import create
+    from geomx_amd.kvstore.optimizer import OptimizerSpec
+
+    cfg = Config.from_env(device="cuda:0")
+    kv = create("dist_sync", cfg=cfg)
+    kv.set_optimizer(OptimizerSpec("sgd", lr=0.1))
+    kv.init("w", torch.ones(1024, device="cuda:0"))
+    kv.push("w", torch.full((1024,), 2.0, device="cuda:0"))
+    out = torch.empty(1024, device="cuda:0")
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((1024,), 0.8, device="cuda:0"))
+
+
+def test_kvstore_gpu_bsc_roundtrip():
+    from geomx_amd import Config
+    from geomx_amd.kvstore import create
+
+    cfg = Config.from_env(device="cuda:0")
+    kv = create("dist_sync", cfg=cfg)
+    kv.set_gradient_compression({"type": "bsc", "threshold": 0.01})
+    n = 1 << 20
+    kv.init("w", torch.zeros(n, device="cuda:0"))
+    g = torch.randn(n, device="cuda:0")
+    kv.push("w", g)
+    out = torch.empty(n, device="cuda:0")
+    kv.pull("w", out)
+    # single party: push is the identity (no WAN compression applied)
+    assert torch.allclose(out, g)
+
+
+def test_resnet50_smoke():
+    from geomx_amd.models import create_model
+    dev = torch.device("cuda:0")
+    model = create_model("resnet50").to(dev)
+    x = torch.randn(4, 3, 224, 224, device=dev)
+    with torch.autocast("cuda", torch.bfloat16):
+        y = model(x)
+        loss = y.float().square().mean()
+    loss.backward()
+    assert torch.isfinite(loss).item()

@@ -1,0 +1,245 @@
+"""GPU numerics tests: gfx950 HIP kernels vs the pure-torch fp32
+reference (geomx_amd.ops.reference). Every test is @pytest.mark.gpu."""
+
+import pytest
+import torch
+
+import geomx_amd.ops as ops
+from geomx_amd.ops import reference as ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def gpu_ok():
+    return torch.cuda.is_available() and ops.native_available()
+
+
+@pytest.fixture(autouse=True)
+def _require_native():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    assert ops.native_available(), \
+        f"native extension missing on GPU box: {ops.native_error()}"
+
+
+# ---------------------------------------------------------------------------
+# 2bit
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("n", [16, 1000, 4096, 1 << 20, (1 << 20) + 37])
+def test_quantize_2bit_matches_reference(n):
+    torch.manual_seed(0)
+    g = (torch.randn(n) * 2).to(DEV)
+    r_gpu = torch.zeros(n, device=DEV)
+    r_cpu = torch.zeros(n)
+    thr = 0.5
+    packed_gpu = ops.quantize_2bit(g, r_gpu, thr)
+    packed_cpu = ref.quantize_2bit(g.cpu(), r_cpu, thr)
+    assert torch.equal(packed_gpu.cpu(), packed_cpu)
+    assert torch.allclose(r_gpu.cpu(), r_cpu, atol=1e-6)
+    # dequantize round-trip
+    deq_gpu = ops.dequantize_2bit(packed_gpu, n, thr)
+    deq_cpu = ref.dequantize_2bit(packed_cpu, n, thr)
+    assert torch.equal(deq_gpu.cpu(), deq_cpu)
+
+
+def test_quantize_2bit_residual_chain():
+    n = 100000
+    g = torch.full((n,), 0.4, device=DEV)
+    r = torch.zeros(n, device=DEV)
+    for expect in [0.0, 0.0, 1.0]:
+        packed = ops.quantize_2bit(g, r, 1.0)
+        d = ops.dequantize_2bit(packed, n, 1.0)
+        assert torch.all(d == expect), expect
+    assert torch.allclose(r, torch.full((n,), 0.2, device=DEV), atol=1e-5)
+
+
+# ---------------------------------------------------------------------------
+# Bi-Sparse
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("n", [10000, 1 << 20, (1 << 22) + 13])
+def test_bsc_pack_matches_reference_fixed_boundary(n):
+    torch.manual_seed(1)
+    g = torch.randn(n)
+    ratio = 0.01
+    boundary = 2.0
+    # GPU: momentum + pack with the same explicit boundary
+    u_g = torch.zeros(n, device=DEV)
+    v_g = torch.zeros(n, device=DEV)
+    gg = g.to(DEV)
+    from geomx_amd.ops import _geops as geops
+    geops.bsc_momentum(gg, u_g, v_g, ref.BSC_MOMENTUM)
+    k = ref.bsc_capacity(n, ratio)
+    vals_g = torch.empty(k, device=DEV)
+    idx_g = torch.empty(k, dtype=torch.int32, device=DEV)
+    geops.bsc_pack(v_g, u_g, vals_g, idx_g, boundary, ref.BSC_PLACEHOLDER)
+    # CPU reference with same boundary
+    u_c = torch.zeros(n)
+    v_c = torch.zeros(n)
+    vals_c, idx_c = ref.bsc_compress(g, u_c, v_c, ratio, boundary=boundary)
+    assert torch.equal(idx_g.cpu(), idx_c)
+    assert torch.allclose(vals_g.cpu(), vals_c, atol=1e-6)
+    assert torch.allclose(u_g.cpu(), u_c, atol=1e-6)
+    assert torch.allclose(v_g.cpu(), v_c, atol=1e-6)
+
+
+def test_bsc_full_pipeline_roundtrip():
+    torch.manual_seed(2)
+    n = 1 << 20
+    ratio = 0.01
+    g = torch.randn(n, device=DEV)
+    u = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    vals, idx = ops.bsc_compress(g, u, v, ratio)
+    k = ref.bsc_capacity(n, ratio)
+    assert vals.numel() == k
+    sent = idx >= 0
+    assert sent.sum() > 0
+    out = ops.bsc_decompress(vals, idx, n)
+    ii = idx[sent].long()
+    assert torch.allclose(out[ii], g[ii], atol=1e-5)
+    assert torch.all(v[ii] == 0)
+    # conservation: out + v == g everywhere (momentum first step: v=g)
+    assert torch.allclose(out + v, g, atol=1e-5)
+
+
+def test_bsc_pull_pack_matches_reference():
+    torch.manual_seed(3)
+    n = 100000
+    x = torch.zeros(n)
+    nz = torch.randperm(n)[:500]
+    x[nz] = torch.randn(500)
+    cap = 1000
+    vals_g, idx_g = ops.bsc_pull_compress(x.to(DEV), cap)
+    vals_c, idx_c = ref.bsc_pull_compress(x, cap)
+    assert torch.equal(idx_g.cpu(), idx_c)
+    assert torch.allclose(vals_g.cpu(), vals_c, atol=1e-6)
+
+
+def test_bsc_unpack_accumulate():
+    vals = torch.tensor([1.0, 2.5], device=DEV)
+    idx = torch.tensor([5, 5], dtype=torch.int32, device=DEV)
+    out = torch.zeros(10, device=DEV)
+    ops.bsc_decompress(vals, idx, 10, out=out, accumulate=True)
+    assert out[5].item() == pytest.approx(3.5)
+    # non-accumulate zeroes the buffer first
+    ops.bsc_decompress(vals[:1], idx[:1], 10, out=out, accumulate=False)
+    assert out[5].item() == pytest.approx(1.0)
+    assert out.sum().item() == pytest.approx(1.0)
+
+
+def test_bsc_capacity_overflow():
+    n = 1000
+    ratio = 0.01  # capacity 10
+    g = torch.ones(n, device=DEV) * 5
+    u = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    vals, idx = ops.bsc_compress(g, u, v, ratio)
+    assert (idx >= 0).sum() == 10
+    assert torch.equal(idx.cpu().long(), torch.arange(10))
+    # beyond-capacity positions keep their v
+    assert torch.all(v[10:] == 5)
+
+
+# ---------------------------------------------------------------------------
+# DGT
+# ---------------------------------------------------------------------------
+
+def test_dgt_contribution_matches_reference():
+    torch.manual_seed(4)
+    g = torch.randn(100000)
+    c_gpu = ops.dgt_contribution(g.to(DEV), 1024)
+    c_cpu = ref.dgt_contribution(g, 1024)
+    assert torch.allclose(c_gpu.cpu(), c_cpu, atol=1e-4)
+
+
+def test_quantize_4bit_chunked_matches_reference():
+    torch.manual_seed(5)
+    n = 10240
+    chunk = 1024
+    x = torch.randn(n)
+    p_gpu, mm_gpu = ops.quantize_4bit_chunked(x.to(DEV), chunk)
+    p_cpu, mm_cpu = ops.quantize_4bit_chunked(x, chunk)
+    assert torch.allclose(mm_gpu.cpu(), mm_cpu, atol=1e-6)
+    assert torch.equal(p_gpu.cpu(), p_cpu)
+    y_gpu = ops.dequantize_4bit_chunked(p_gpu, mm_gpu, n, chunk)
+    y_cpu = ops.dequantize_4bit_chunked(p_cpu, mm_cpu, n, chunk)
+    assert torch.allclose(y_gpu.cpu(), y_cpu, atol=1e-6)
+    # quantization error bounded by half a step per chunk
+    err = (y_gpu.cpu() - x).abs().max()
+    steps = (mm_cpu[:, 1] - mm_cpu[:, 0]) / 16
+    assert err <= steps.max() * 0.5 + 1e-5
+
+
+def test_quantize_4bit_residual_gpu():
+    n = 4096
+    x = torch.randn(n, device=DEV)
+    res = torch.zeros(n, device=DEV)
+    p, mm = ops.quantize_4bit_chunked(x, 1024, residual=res)
+    y = ops.dequantize_4bit_chunked(p, mm, n, 1024)
+    assert torch.allclose(res, x - y, atol=1e-5)
+
+
+# ---------------------------------------------------------------------------
+# fused optimizers
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("n", [1000, (1 << 20) + 3])
+def test_sgd_mom_update_matches_reference(n):
+    torch.manual_seed(6)
+    w0 = torch.randn(n)
+    g = torch.randn(n)
+    w_g, m_g = w0.clone().to(DEV), torch.zeros(n, device=DEV)
+    w_c, m_c = w0.clone(), torch.zeros(n)
+    for _ in range(3):
+        ops.sgd_mom_update(w_g, g.to(DEV), m_g, 0.1, 0.9, 1e-4, 1.0)
+        ref.sgd_mom_update(w_c, g, m_c, 0.1, 0.9, 1e-4, 1.0)
+    assert torch.allclose(w_g.cpu(), w_c, atol=1e-5)
+    assert torch.allclose(m_g.cpu(), m_c, atol=1e-5)
+
+
+def test_adam_update_matches_reference():
+    torch.manual_seed(7)
+    n = 100003
+    w0 = torch.randn(n)
+    w_g = w0.clone().to(DEV)
+    m_g = torch.zeros(n, device=DEV)
+    v_g = torch.zeros(n, device=DEV)
+    w_c = w0.clone()
+    m_c = torch.zeros(n)
+    v_c = torch.zeros(n)
+    for t in range(1, 5):
+        g = torch.randn(n)
+        ops.adam_update(w_g, g.to(DEV), m_g, v_g, t, 0.01)
+        ref.adam_update(w_c, g, m_c, v_c, t, 0.01)
+    assert torch.allclose(w_g.cpu(), w_c, atol=1e-5)
+
+
+def test_dcasgd_update_matches_reference():
+    torch.manual_seed(8)
+    n = 50000
+    w0 = torch.randn(n)
+    w_g = w0.clone().to(DEV)
+    p_g = w0.clone().to(DEV)
+    w_c = w0.clone()
+    p_c = w0.clone()
+    for _ in range(3):
+        g = torch.randn(n)
+        ops.dcasgd_update(w_g, g.to(DEV), p_g, None, 0.01, 0.04)
+        ref.dcasgd_update(w_c, g, p_c, None, 0.01, 0.04)
+    assert torch.allclose(w_g.cpu(), w_c, atol=1e-5)
+    assert torch.allclose(p_g.cpu(), p_c, atol=1e-5)
+
+
+def test_sgd_update_matches_reference():
+    n = 12345
+    w0 = torch.randn(n)
+    g = torch.randn(n)
+    w_g = w0.clone().to(DEV)
+    w_c = w0.clone()
+    ops.sgd_update(w_g, g.to(DEV), 0.1, 1e-4, 2.0)
+    ref.sgd_update(w_c, g, 0.1, 1e-4, 2.0)
+    assert torch.allclose(w_g.cpu(), w_c, atol=1e-6)
