@@ -56,6 +56,9 @@ def parse_args():
     p.add_argument("--optimizer", type=str, default="sgd_mom")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--json-out", type=str, default=None)
+    p.add_argument("--no-channels-last", action="store_true",
+                   help="disable NHWC layout (NHWC avoids MIOpen's "
+                        "batched_transpose + slow NCHW pooling kernels)")
     return p.parse_args()
 
 
@@ -90,13 +93,20 @@ def main():
                                + str(ops.native_error()))
 
     torch.manual_seed(1234)  # same random init on all ranks
+    if use_cuda:
+        torch.backends.cudnn.benchmark = True  # MIOpen find mode
     model = create_model(args.model, image_size=args.image_size,
                          num_classes=args.num_classes).to(device)
+    channels_last = use_cuda and not args.no_channels_last
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
     spec = OptimizerSpec(name=args.optimizer, lr=0.01, momentum=0.9)
     trainer = GeoTrainer(model, cfg, topo, spec, mode=args.mode)
 
     bs = args.batch_size
     x = torch.randn(bs, 3, args.image_size, args.image_size, device=device)
+    if channels_last:
+        x = x.to(memory_format=torch.channels_last)
     y = torch.randint(0, args.num_classes, (bs,), device=device)
 
     amp_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32

@@ -63,6 +63,19 @@ class _Bucket:
         self.bsc_v: Optional[torch.Tensor] = None
 
 
+def _alias_view(flat: torch.Tensor, off: int, p: torch.Tensor) -> torch.Tensor:
+    """A view of flat[off:off+numel] shaped like p, preserving p's memory
+    format (channels_last conv weights keep NHWC strides so MIOpen never
+    re-transposes them)."""
+    n = p.numel()
+    if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last) \
+            and not p.is_contiguous():
+        N, C, H, W = p.shape
+        return flat[off:off + n].as_strided((N, C, H, W),
+                                            (C * H * W, 1, W * C, C))
+    return flat[off:off + n].view(p.shape)
+
+
 class GeoTrainer:
     def __init__(self, model: torch.nn.Module, cfg: Config, topo: Topology,
                  optimizer: Optional[OptimizerSpec] = None,
@@ -113,11 +126,11 @@ class GeoTrainer:
             off = 0
             for p in b.params:
                 n = p.numel()
-                view = b.flat[off:off + n].view(p.shape)
+                view = _alias_view(b.flat, off, p)
                 b.views.append(view)
                 # grads write straight into the bucket
                 p.grad = view
-                pview = b.param_flat[off:off + n].view(p.shape)
+                pview = _alias_view(b.param_flat, off, p)
                 with torch.no_grad():
                     pview.copy_(p.data.float())
                 p.data = pview
