@@ -40,7 +40,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch-size", type=int, default=256,
+    p.add_argument("--batch-size", type=int, default=512,
                    help="per-GPU batch (weak scaling)")
     p.add_argument("--model", type=str, default="geomx_cnn")
     p.add_argument("--image-size", type=int, default=224)
