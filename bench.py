@@ -71,18 +71,19 @@ def main():
     device_str = "cuda" if use_cuda else "cpu"
     backend = "nccl" if use_cuda else "gloo"
 
-    if args.mode == "hips" and world > 1:
-        parties = args.parties or max(2, world // 4)
-        if world % parties:
-            parties = 1
-    else:
+    parties = args.parties or (max(2, world // 4) if world > 1 else 1)
+    if world % max(1, parties):
         parties = 1
+    # flat mode: topology has ONE party (no hierarchy) but the WAN
+    # accounting still uses `parties` so flat-vs-hips is charged the
+    # same emulated inter-DC link
+    topo_parties = parties if args.mode == "hips" and world > 1 else 1
 
     cfg = Config.from_env(
         num_parties=parties, backend=backend,
         compression=args.compress, bsc_ratio=args.bsc_ratio,
         wan_gbps=args.wan_gbps, bucket_mb=args.bucket_mb)
-    topo = init_topology(parties, None, backend)
+    topo = init_topology(topo_parties, None, backend)
     device = topo.device
 
     # fail loudly if the native extension is missing on a GPU machine

@@ -127,6 +127,18 @@ class KVStoreDist(KVStoreBase):
 
     _barrier = barrier
 
+    def get_num_dead_node(self, node_id: int = -1, timeout_s: float = 60.0) -> int:
+        """Heartbeat-based dead-peer count (kvstore_dist.h:225-234 parity;
+        node_id is accepted for API compatibility, the count covers the
+        whole world)."""
+        if self.topo.world_size <= 1:
+            return 0
+        if not hasattr(self, "_hb"):
+            from ..utils.health import HeartbeatMonitor
+            self._hb = HeartbeatMonitor()
+            self._hb.start()
+        return self._hb.get_num_dead_node(timeout_s)
+
     # ------------------------------------------------------------------
     # configuration commands (the reference sends these as in-band server
     # commands: CommandType, kvstore_dist_server.h:49-52)

@@ -1,0 +1,156 @@
+"""Tests for checkpoint / metrics / data / profiler utilities."""
+
+import torch
+
+from geomx_amd import Config
+from geomx_amd.kvstore.optimizer import OptimizerSpec
+from geomx_amd.models import geo_cnn
+from geomx_amd.parallel import GeoTrainer
+from geomx_amd.topology import init_topology
+from geomx_amd.utils import checkpoint as ckpt
+from geomx_amd.utils.data import (ClassSplitSampler, SplitSampler,
+                                  SyntheticImageDataset, worker_loader)
+from geomx_amd.utils.metrics import Accuracy, Measure
+
+
+def test_save_load_parameters(tmp_path):
+    m1 = torch.nn.Linear(4, 3)
+    f = str(tmp_path / "m.params")
+    ckpt.save_parameters(m1, f)
+    m2 = torch.nn.Linear(4, 3)
+    ckpt.load_parameters(m2, f)
+    for a, b in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(a, b)
+
+
+def test_checkpoint_roundtrip_with_trainer(tmp_path):
+    torch.manual_seed(0)
+    cfg = Config.from_env(device="cpu")
+    topo = init_topology(1, None, "gloo", "cpu")
+    model = torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 2))
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec("adam", lr=0.01))
+    x = torch.randn(4, 8)
+    for _ in range(2):
+        loss = model(x).square().mean()
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+    files = ckpt.save_checkpoint(model, tr, str(tmp_path / "ck"), 1)
+    w_before = [p.detach().clone() for p in model.parameters()]
+
+    # new model+trainer, resumed
+    torch.manual_seed(123)
+    model2 = torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 2))
+    tr2 = GeoTrainer(model2, cfg, topo, OptimizerSpec("adam", lr=0.01))
+    ckpt.load_checkpoint(model2, tr2, str(tmp_path / "ck"), 1)
+    for a, b in zip(w_before, model2.parameters()):
+        assert torch.allclose(a, b, atol=1e-7)
+    # optimizer state came back
+    assert tr2.server_opt.step_count == tr.server_opt.step_count
+    # training continues identically
+    for t in (tr, tr2):
+        pass
+    loss1 = model(x).square().mean()
+    tr.zero_grad(); loss1.backward(); tr.step()
+    loss2 = model2(x).square().mean()
+    tr2.zero_grad(); loss2.backward(); tr2.step()
+    for a, b in zip(model.parameters(), model2.parameters()):
+        assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_split_sampler_partitions():
+    s0 = SplitSampler(100, 4, 0, shuffle=False)
+    s3 = SplitSampler(100, 4, 3, shuffle=False)
+    assert list(s0) == list(range(25))
+    assert list(s3) == list(range(75, 100))
+    all_idx = []
+    for i in range(4):
+        all_idx += list(SplitSampler(100, 4, i, shuffle=False))
+    assert sorted(all_idx) == list(range(100))
+
+
+def test_class_split_sampler_disjoint():
+    ds = SyntheticImageDataset(200, shape=(1, 8, 8), num_classes=10)
+    seen = [set(ds.y[list(ClassSplitSampler(ds.y, 2, i))].tolist())
+            for i in range(2)]
+    assert seen[0].isdisjoint(seen[1])
+    assert seen[0] | seen[1] == set(range(10))
+
+
+def test_worker_loader_batches():
+    ds = SyntheticImageDataset(128, shape=(3, 8, 8))
+    dl = worker_loader(ds, 16, 2, 0)
+    xb, yb = next(iter(dl))
+    assert xb.shape == (16, 3, 8, 8)
+
+
+def test_measure_and_accuracy(tmp_path):
+    m = Measure(sync_cuda=False)
+    m.start("fwd")
+    m.stop("fwd")
+    row = m.next_iteration(iter=1)
+    assert "fwd" in row and row["fwd"] >= 0
+    f = str(tmp_path / "meas.jsonl")
+    m.dump(f)
+    assert open(f).read().strip()
+
+    acc = Accuracy()
+    acc.update(torch.tensor([1, 2]), torch.tensor([[0, 1.0, 0], [0, 0, 1.0]]))
+    assert acc.get() == 1.0
+
+
+def test_example_convergence_cpu():
+    """Convergence smoke (the reference's de-facto system test is
+    per-iteration accuracy climbing, cnn.py:129-131): single-process
+    vanilla kvstore training on learnable synthetic data."""
+    from geomx_amd.kvstore import create
+    from geomx_amd.utils.metrics import eval_acc
+
+    cfg = Config.from_env(device="cpu")
+    kv = create("dist_sync", cfg=cfg)
+    kv.set_optimizer(OptimizerSpec("adam", lr=0.003))
+    torch.manual_seed(0)
+    net = geo_cnn(in_channels=3, image_size=16)
+    ds = SyntheticImageDataset(512, shape=(3, 16, 16), seed=1)
+    dl = torch.utils.data.DataLoader(ds, batch_size=64, shuffle=True)
+    test = torch.utils.data.DataLoader(
+        SyntheticImageDataset(128, shape=(3, 16, 16), seed=1), batch_size=64)
+    params = [p for p in net.parameters()]
+    for i, p in enumerate(params):
+        kv.init(i, p.data)
+    acc0 = eval_acc(net, test, "cpu")
+    for epoch in range(3):
+        for x, y in dl:
+            loss = torch.nn.functional.cross_entropy(net(x), y)
+            net.zero_grad()
+            loss.backward()
+            for i, p in enumerate(params):
+                kv.push(i, p.grad / x.shape[0])
+                kv.pull(i, p.data)
+    acc1 = eval_acc(net, test, "cpu")
+    assert acc1 > max(0.3, acc0 + 0.15), (acc0, acc1)
+
+
+def test_heartbeat_monitor_single():
+    import time
+
+    from geomx_amd.utils.health import HeartbeatMonitor
+
+    class FakeStore:
+        def __init__(self):
+            self.d = {}
+
+        def set(self, k, v):
+            self.d[k] = v
+
+        def get(self, k):
+            return self.d[k]
+
+    st = FakeStore()
+    hb = HeartbeatMonitor(interval_s=0.05, store=st, rank=0, world_size=2)
+    hb.start()
+    time.sleep(0.2)
+    # rank 0 alive, rank 1 never beat
+    assert hb.dead_nodes(timeout_s=10.0) == [1]
+    assert hb.get_num_dead_node(10.0) == 1
+    hb.stop()
