@@ -95,6 +95,10 @@ class Config:
     wan_gbps: float = 0.0
 
     # --- runtime -------------------------------------------------------
+    # P3/MultiGPS big-tensor slicing: keys with numel >= bigarray_bound
+    # are sliced across party leaders (EncodeP3Key / bigarray_bound_,
+    # kvstore_dist.h:69,763-799; MXNET_KVSTORE_BIGARRAY_BOUND)
+    bigarray_bound: int = 1_000_000
     bucket_mb: int = 25          # gradient bucket size for fused collectives
     overlap: bool = True         # overlap comm with backward
     backend: Optional[str] = None  # override; default nccl on GPU, gloo on CPU
@@ -120,6 +124,9 @@ class Config:
             dgt_alpha=_env_float(["DGT_CONTRIBUTION_ALPHA", "GEOMX_DGT_ALPHA"], 0.3),
             wan_gbps=_env_float(["GEOMX_WAN_GBPS"], 0.0),
             bucket_mb=_env_int(["GEOMX_BUCKET_MB"], 25),
+            bigarray_bound=_env_int(
+                ["MXNET_KVSTORE_BIGARRAY_BOUND", "GEOMX_BIGARRAY_BOUND"],
+                1_000_000),
         )
         sizes = _env_str(["GEOMX_PARTY_SIZES"], None)
         if sizes:
@@ -134,7 +141,7 @@ class Config:
     def validate(self):
         if self.mode not in ("dist_sync", "dist_async", "local"):
             raise ValueError(f"mode must be dist_sync|dist_async|local, got {self.mode}")
-        if self.compression not in (None, "2bit", "bsc", "fp16", "mpq"):
+        if self.compression not in (None, "2bit", "bsc", "fp16", "mpq", "dgt"):
             raise ValueError(f"unknown compression {self.compression!r}")
         if not (0 < self.bsc_ratio < 1):
             raise ValueError("bsc_ratio must be in (0,1)")

@@ -1,0 +1,65 @@
+"""DGT (Differential Gradient Transmission) for the WAN tier.
+
+Reference behavior (3rdparty/ps-lite: kv_app.h:842-995, van.cc:707-824):
+each push tensor is split into fixed-size chunks; a per-chunk EWMA
+"contribution" (mean |g|) ranks them; the top DMLC_K fraction travels
+reliably at full precision, the rest over lossy low-priority channels,
+optionally 4-bit linearly quantized with residual feedback
+(ENABLE_DGT=3).
+
+MI355X-native mapping: xGMI/RCCL has no lossy transport, so
+"unimportant" chunks become 4-bit-quantized chunks (same traffic
+reduction, deterministic instead of lossy) and the priority ordering
+becomes the charge model: exact bytes for important chunks + half-byte
+per element for the rest.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import torch
+
+from .. import ops
+
+
+class DGTState:
+    def __init__(self, numel: int, device, chunk_elems: int = 1024,
+                 k: float = 0.5, alpha: float = 0.3):
+        self.numel = numel
+        self.chunk = chunk_elems
+        self.k = k
+        self.alpha = alpha
+        self.nchunks = (numel + chunk_elems - 1) // chunk_elems
+        self.contrib = None
+        self.residual = torch.zeros(numel, device=device)
+
+    def transform(self, x: torch.Tensor) -> Tuple[torch.Tensor, int]:
+        """Return (lossy reconstruction, wire bytes). Important chunks
+        pass through exactly; unimportant chunks are 4-bit quantized
+        with residual feedback."""
+        flat = x.reshape(-1)
+        contrib = ops.dgt_contribution(flat, self.chunk)
+        if self.contrib is None or self.contrib.numel() != contrib.numel():
+            self.contrib = contrib
+        else:
+            self.contrib = self.alpha * contrib + (1 - self.alpha) * self.contrib
+        n_keep = max(1, int(math.ceil(self.k * self.nchunks)))
+        keep = torch.topk(self.contrib, n_keep).indices
+        keep_mask = torch.zeros(self.nchunks, dtype=torch.bool,
+                                device=flat.device)
+        keep_mask[keep] = True
+        packed, minmax = ops.quantize_4bit_chunked(flat, self.chunk,
+                                                   self.residual)
+        deq = ops.dequantize_4bit_chunked(packed, minmax.to(flat.device),
+                                          self.numel, self.chunk)
+        elem_mask = keep_mask.repeat_interleave(self.chunk)[:self.numel]
+        out = torch.where(elem_mask, flat, deq.to(flat.device))
+        # residual only meaningful for quantized chunks; zero it for
+        # exact chunks (they carried no error)
+        self.residual[elem_mask] = 0.0
+        exact_bytes = n_keep * self.chunk * 4
+        lossy_bytes = (self.nchunks - n_keep) * (self.chunk // 2 + 8)
+        wire = min(self.numel * 4, exact_bytes) + max(0, lossy_bytes)
+        return out, wire

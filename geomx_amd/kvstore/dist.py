@@ -59,7 +59,8 @@ from .wan import TokenBucket, cross_party_bytes
 class _KeyState:
     __slots__ = ("shape", "numel", "dtype", "stored", "update_buf",
                  "bsc_u", "bsc_v", "residual_2bit", "milestone",
-                 "push_count", "owner_party", "dgt_contrib", "dgt_residual")
+                 "push_count", "owner_party", "dgt_contrib", "dgt_residual",
+                 "sliced", "padded")
 
     def __init__(self, shape, numel, dtype):
         self.shape = shape
@@ -75,6 +76,8 @@ class _KeyState:
         self.dgt_residual: Optional[torch.Tensor] = None
         self.push_count = 0
         self.owner_party = 0
+        self.sliced = False   # P3/MultiGPS big-tensor slicing across leaders
+        self.padded = 0       # padded numel (multiple of P) when sliced
 
 
 class KVStoreDist(KVStoreBase):
@@ -173,6 +176,14 @@ class KVStoreDist(KVStoreBase):
             raise ValueError(f"key {key!r} already initialised")
         st = _KeyState(tuple(value.shape), value.numel(), value.dtype)
         st.owner_party = len(self._key_order) % self.topo.num_parties
+        P = self.topo.num_parties
+        if P > 1 and self.global_mode == "sharded" \
+                and st.numel >= self.cfg.bigarray_bound:
+            # P3/MultiGPS: slice the key uniformly across ALL leaders
+            # (EncodeP3Key kvstore_dist.h:763-799; server-side sharding
+            # kvstore_dist_server.h:1770-1810)
+            st.sliced = True
+            st.padded = ((st.numel + P - 1) // P) * P
         self.keys[key] = st
         self._key_order.append(key)
         flat = value.detach().reshape(-1).float().to(self._device)
@@ -318,6 +329,31 @@ class KVStoreDist(KVStoreBase):
             self.wan.charge(cross_party_bytes("all_gather", nbytes, P))
             return list(hlist)
 
+        if st.sliced and self.cfg.mode == "dist_sync":
+            # P3/MultiGPS sliced dense path: each leader owns 1/P of the
+            # key; reduce each slice to its owner (a reduce_scatter),
+            # update locally, all_gather on pull.
+            chunk = st.padded // P
+            for c in range(P):
+                lo, hi = c * chunk, min(st.numel, (c + 1) * chunk)
+                if lo >= hi:
+                    continue
+                sl = party_sum[lo:hi].clone()
+                dist.reduce(sl, dst=topo.leader_ranks[c],
+                            op=dist.ReduceOp.SUM, group=group)
+                if topo.party_id == c:
+                    my_slice = sl
+            self.wan.charge(cross_party_bytes("reduce", st.numel * 4, P))
+            lo = topo.party_id * chunk
+            hi = min(st.numel, lo + chunk)
+            if lo < hi:
+                w_slice = st.stored[lo:hi]
+                if self.optimizer is not None:
+                    self.optimizer.update((key, "slice"), w_slice, my_slice)
+                else:
+                    w_slice.copy_(my_slice)
+            return None  # pull all_gathers the slices
+
         # dense fp32
         if self.global_mode == "replicated" or self.cfg.mode == "dist_async":
             flist = [torch.empty_like(party_sum) for _ in range(P)]
@@ -406,6 +442,21 @@ class KVStoreDist(KVStoreBase):
             return
         group = topo.leader_group
         owner_leader = topo.leader_ranks[st.owner_party]
+        if st.sliced:
+            # all_gather the per-leader slices (MultiGPS reassembly,
+            # kvstore_dist_server.h:1039-1094 — here order is by
+            # construction, no sort needed)
+            chunk = st.padded // P
+            lo = topo.party_id * chunk
+            hi = min(st.numel, lo + chunk)
+            mine = torch.zeros(chunk, device=self._device)
+            if lo < hi:
+                mine[:hi - lo] = st.stored[lo:hi]
+            parts = [torch.empty_like(mine) for _ in range(P)]
+            dist.all_gather(parts, mine, group=group)
+            self.wan.charge(cross_party_bytes("all_gather", chunk * 4, P))
+            st.stored = torch.cat(parts)[:st.numel]
+            return
         if ctype == "bsc" and self.optimizer is None:
             # pull-side re-sparsification (BSCPullCompress,
             # gradient_compression.cc:271-308) — only meaningful for the
@@ -453,37 +504,22 @@ class KVStoreDist(KVStoreBase):
         self.optimizer.load_state_dict(blob, device=self._device)
 
     # ------------------------------------------------------------------
-    # DGT transform (priority chunks; lossy low-priority chunks)
+    # DGT transform (shared DGTState, kvstore/dgt.py)
     # ------------------------------------------------------------------
     def _dgt_transform(self, st: _KeyState, x: torch.Tensor) -> torch.Tensor:
-        """Split into chunks, EWMA contribution, keep top DMLC_K fraction
-        exact; 4-bit-quantize the rest (enable_dgt==3 semantics,
-        van.cc:750-824 + kv_app.h:853-894). Returns the reconstructed
-        (lossy) tensor; _dgt_wire_bytes gives its wire cost."""
-        chunk = max(64, self.cfg.dgt_block_size // 4)
-        contrib = ops.dgt_contribution(x, chunk)
-        if st.dgt_contrib is None or st.dgt_contrib.numel() != contrib.numel():
-            st.dgt_contrib = contrib
-            st.dgt_residual = torch.zeros(st.numel, device=x.device)
-        else:
-            a = self.cfg.dgt_alpha
-            st.dgt_contrib = a * contrib + (1 - a) * st.dgt_contrib
-        nchunks = contrib.numel()
-        n_keep = max(1, int(math.ceil(self.cfg.dgt_k * nchunks)))
-        keep = torch.topk(st.dgt_contrib, n_keep).indices
-        keep_mask = torch.zeros(nchunks, dtype=torch.bool, device=x.device)
-        keep_mask[keep] = True
-        self._dgt_last = (nchunks, n_keep, chunk)
-        out = x.clone()
-        # quantize unimportant chunks (with residual feedback)
-        packed, minmax = ops.quantize_4bit_chunked(x, chunk, st.dgt_residual)
-        deq = ops.dequantize_4bit_chunked(packed, minmax, st.numel, chunk)
-        elem_mask = keep_mask.repeat_interleave(chunk)[:st.numel]
-        out[~elem_mask] = deq.to(out.device)[~elem_mask]
+        if not hasattr(self, "_dgt_states"):
+            self._dgt_states = {}
+        key = id(st)
+        dg = self._dgt_states.get(key)
+        if dg is None or dg.numel != st.numel:
+            from .dgt import DGTState
+            dg = DGTState(st.numel, x.device,
+                          chunk_elems=max(64, self.cfg.dgt_block_size // 4),
+                          k=self.cfg.dgt_k, alpha=self.cfg.dgt_alpha)
+            self._dgt_states[key] = dg
+        out, wire = dg.transform(x)
+        self._dgt_wire = wire
         return out
 
     def _dgt_wire_bytes(self, st: _KeyState) -> int:
-        nchunks, n_keep, chunk = self._dgt_last
-        exact = n_keep * chunk * 4
-        lossy = (nchunks - n_keep) * (chunk // 2 + 8)
-        return min(st.numel * 4, exact) + max(0, lossy)
+        return self._dgt_wire

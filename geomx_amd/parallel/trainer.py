@@ -48,7 +48,7 @@ from ..topology import Topology
 
 class _Bucket:
     __slots__ = ("params", "flat", "param_flat", "views", "ready", "expected",
-                 "work", "index", "bsc_u", "bsc_v")
+                 "work", "index", "bsc_u", "bsc_v", "dgt")
 
     def __init__(self, index: int):
         self.index = index
@@ -61,6 +61,7 @@ class _Bucket:
         self.work = None
         self.bsc_u: Optional[torch.Tensor] = None
         self.bsc_v: Optional[torch.Tensor] = None
+        self.dgt = None
 
 
 def _alias_view(flat: torch.Tensor, off: int, p: torch.Tensor) -> torch.Tensor:
@@ -218,6 +219,18 @@ class GeoTrainer:
                         ops.bsc_decompress(v_, i_, b.flat.numel(),
                                            out=acc, accumulate=True)
                     b.flat.copy_(acc)
+                elif ctype == "dgt":
+                    if b.dgt is None:
+                        from ..kvstore.dgt import DGTState
+                        b.dgt = DGTState(b.flat.numel(), b.flat.device,
+                                         chunk_elems=max(
+                                             64, self.cfg.dgt_block_size // 4),
+                                         k=self.cfg.dgt_k,
+                                         alpha=self.cfg.dgt_alpha)
+                    lossy, wire = b.dgt.transform(b.flat)
+                    b.flat.copy_(lossy)
+                    dist.all_reduce(b.flat, group=topo.leader_group)
+                    self.wan.charge(cross_party_bytes("all_reduce", wire, P))
                 elif ctype in ("fp16", "mpq"):
                     h = b.flat.to(torch.float16)
                     dist.all_reduce(h, group=topo.leader_group)

@@ -283,3 +283,29 @@ def _2bit(rank, world):
 
 def test_2bit_ws2():
     run_dist(2, _2bit)
+
+
+# ---------------------------------------------------------------------------
+# P3/MultiGPS big-tensor slicing across leaders
+# ---------------------------------------------------------------------------
+
+def _sliced_big_tensor(rank, world):
+    kv = _mk(num_parties=2, bigarray_bound=100)
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    n = 1001  # odd size: exercises padding
+    kv.init("big", torch.ones(n))
+    assert kv.keys["big"].sliced
+    kv.push("big", torch.full((n,), 1.0))  # sum over 4 workers = 4
+    out = torch.empty(n)
+    kv.pull("big", out)
+    assert torch.allclose(out, torch.full((n,), 0.6), atol=1e-6), (rank, out)
+    # small key stays unsliced
+    kv.init("small", torch.zeros(10))
+    assert not kv.keys["small"].sliced
+    kv.push("small", torch.ones(10))
+    kv.pull("small", out[:10])
+    assert torch.allclose(out[:10], torch.full((10,), -0.4), atol=1e-6)
+
+
+def test_sliced_big_tensor_ws4():
+    run_dist(4, _sliced_big_tensor)

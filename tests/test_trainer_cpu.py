@@ -114,3 +114,30 @@ def _trainer_hips_bsc(rank, world):
 
 def test_hips_bsc_consistent_ws4():
     run_dist(4, _trainer_hips_bsc)
+
+
+def _trainer_hips_dgt(rank, world):
+    cfg = Config.from_env(num_parties=2, backend="gloo", device="cpu",
+                          bucket_mb=1, compression="dgt", dgt_k=0.5,
+                          dgt_block_size=256)
+    topo = init_topology(2, None, "gloo", "cpu")
+    model = _tiny_model()
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec(name="sgd", lr=0.05),
+                    mode="hips")
+    for s in range(3):
+        x, y = _make_data(seed=11 + s)
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+    import torch.distributed as dist
+    for p in model.parameters():
+        assert torch.isfinite(p.data).all()
+        ref = p.data.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(p.data, ref, atol=1e-6)
+    assert tr.wan.total_bytes == 0  # no cap set -> no charging
+
+
+def test_hips_dgt_consistent_ws4():
+    run_dist(4, _trainer_hips_dgt)
