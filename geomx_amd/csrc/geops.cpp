@@ -149,6 +149,8 @@ void quantize_4bit(torch::Tensor x, torch::Tensor residual,
   const int nchunks = (int)((n + chunk - 1) / chunk);
   TORCH_CHECK(packed.numel() == (n + 1) / 2);
   TORCH_CHECK(minmax.numel() == 2 * nchunks);
+  TORCH_CHECK((chunk & (chunk - 1)) == 0 && chunk >= 4,
+              "GPU quantize_4bit needs power-of-two chunk >= 4");
   const bool has_res = residual.defined() && residual.numel() > 0;
   if (has_res) check_f32(residual, "residual");
   geops_quantize_4bit(x.data_ptr<float>(),
@@ -161,6 +163,8 @@ void dequantize_4bit(torch::Tensor packed, torch::Tensor minmax,
                      torch::Tensor out, int64_t chunk) {
   check_f32(out, "out"); check_f32(minmax, "minmax");
   TORCH_CHECK(packed.scalar_type() == torch::kUInt8 && packed.is_contiguous());
+  TORCH_CHECK((chunk & (chunk - 1)) == 0 && chunk >= 4,
+              "GPU dequantize_4bit needs power-of-two chunk >= 4");
   geops_dequantize_4bit(packed.data_ptr<uint8_t>(), minmax.data_ptr<float>(),
                         out.data_ptr<float>(), out.numel(), (int)chunk,
                         cur_stream());

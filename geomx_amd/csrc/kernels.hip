@@ -77,15 +77,33 @@ extern "C" __global__ void k_quantize_2bit(const float* __restrict__ grad,
   }
 }
 
+__device__ __forceinline__ float decode2(uint32_t c, float thr) {
+  return (c == 3u) ? thr : ((c == 2u) ? -thr : 0.0f);
+}
+
+// vectorized main body: each thread decodes 4 elements (one byte of a
+// word) and stores one float4 — lane-contiguous 16B stores; scalar tail.
 extern "C" __global__ void k_dequantize_2bit(const uint32_t* __restrict__ in,
                                              float* __restrict__ out,
                                              long long n, float thr) {
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
-  for (long long i = tid; i < n; i += stride) {
+  const long long n4 = n >> 2;  // float4 groups fully inside n
+  float4* out4 = reinterpret_cast<float4*>(out);
+  for (long long q = tid; q < n4; q += stride) {
+    // elements 4q..4q+3 live in byte (q&3) of word (q>>2)
+    const uint32_t word = in[q >> 2];
+    const uint32_t byte = (word >> ((q & 3) * 8)) & 0xFFu;
+    float4 o;
+    o.x = decode2(byte & 3u, thr);
+    o.y = decode2((byte >> 2) & 3u, thr);
+    o.z = decode2((byte >> 4) & 3u, thr);
+    o.w = decode2((byte >> 6) & 3u, thr);
+    out4[q] = o;
+  }
+  for (long long i = (n4 << 2) + tid; i < n; i += stride) {
     const uint32_t word = in[i >> 4];
-    const uint32_t code = (word >> ((i & 15) * 2)) & 3u;
-    out[i] = (code == 3u) ? thr : ((code == 2u) ? -thr : 0.0f);
+    out[i] = decode2((word >> ((i & 15) * 2)) & 3u, thr);
   }
 }
 
@@ -125,12 +143,15 @@ extern "C" __global__ void k_bsc_momentum(const float* __restrict__ g,
 // ---------------------------------------------------------------------------
 
 __device__ __forceinline__ long long chunk_begin(long long n, int nb, int b) {
-  const long long chunk = (n + nb - 1) / nb;
+  // chunk rounded to a multiple of 4 so every block's slice is
+  // 16-byte aligned for float4 loads
+  const long long chunk = (((n + nb - 1) / nb) + 3) & ~3LL;
   long long s = (long long)b * chunk;
   return s < n ? s : n;
 }
 
-// phase 1: per-block predicate counts
+// phase 1: per-block predicate counts (float4 main body + scalar tail;
+// chunk boundaries are 4-aligned by construction in chunk_begin)
 template <bool NONZERO>
 __global__ void k_bsc_count_t(const float* __restrict__ v, float boundary,
                               long long n, int nb,
@@ -139,7 +160,18 @@ __global__ void k_bsc_count_t(const float* __restrict__ v, float boundary,
   const int b = blockIdx.x;
   const long long lo = chunk_begin(n, nb, b), hi = chunk_begin(n, nb, b + 1);
   long long cnt = 0;
-  for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+  const long long nvec = (hi - lo) >> 2;
+  const float4* v4 = reinterpret_cast<const float4*>(v + lo);
+  for (long long q = threadIdx.x; q < nvec; q += blockDim.x) {
+    const float4 x = v4[q];
+    if (NONZERO) {
+      cnt += (x.x != 0.0f) + (x.y != 0.0f) + (x.z != 0.0f) + (x.w != 0.0f);
+    } else {
+      cnt += (fabsf(x.x) >= boundary) + (fabsf(x.y) >= boundary) +
+             (fabsf(x.z) >= boundary) + (fabsf(x.w) >= boundary);
+    }
+  }
+  for (long long i = lo + (nvec << 2) + threadIdx.x; i < hi; i += blockDim.x) {
     const float x = v[i];
     const bool pred = NONZERO ? (x != 0.0f) : (fabsf(x) >= boundary);
     cnt += pred ? 1 : 0;
@@ -173,6 +205,10 @@ extern "C" __global__ void k_bsc_scan(long long* __restrict__ counts, int nb) {
 
 // phase 3: ordered pack with capacity bound; optionally zero u,v at the
 // positions actually sent (error feedback, reference :258-260)
+// ordered capacity-bounded pack: each thread takes 4 consecutive
+// elements per iteration (block covers 1024), per-thread counts are
+// scanned wave-wide with __shfl_up, wave totals with one LDS pass —
+// index order is preserved across threads, waves, iterations, blocks.
 template <bool ZERO_UV>
 __global__ void k_bsc_pack_t(const float* __restrict__ v_in,
                              float* __restrict__ v_mut,
@@ -183,48 +219,74 @@ __global__ void k_bsc_pack_t(const float* __restrict__ v_in,
                              float boundary, long long n, int nb,
                              long long capacity) {
   __shared__ long long wave_base[GEOPS_THREADS / 64];
-  __shared__ long long carry;
+  __shared__ long long carry_s;
   const int b = blockIdx.x;
   const long long lo = chunk_begin(n, nb, b), hi = chunk_begin(n, nb, b + 1);
-  if (threadIdx.x == 0) carry = offsets[b];
+  if (threadIdx.x == 0) carry_s = offsets[b];
   __syncthreads();
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int nwaves = blockDim.x >> 6;
-  for (long long base = lo; base < hi; base += blockDim.x) {
-    const long long i = base + threadIdx.x;
-    const bool in_range = i < hi;
-    float x = 0.0f;
-    bool pred = false;
-    if (in_range) {
-      x = v_in[i];
-      pred = fabsf(x) >= boundary;
+  const long long span = (long long)blockDim.x * 4;
+  for (long long base = lo; base < hi; base += span) {
+    const long long i0 = base + (long long)threadIdx.x * 4;
+    float x[4];
+    bool p[4];
+    int cnt = 0;
+    if (i0 + 4 <= hi) {  // aligned float4 fast path (lo is 4-aligned)
+      const float4 xx = *reinterpret_cast<const float4*>(v_in + i0);
+      x[0] = xx.x; x[1] = xx.y; x[2] = xx.z; x[3] = xx.w;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        p[j] = fabsf(x[j]) >= boundary;
+        cnt += p[j] ? 1 : 0;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long long i = i0 + j;
+        const bool in_range = i < hi;
+        x[j] = in_range ? v_in[i] : 0.0f;
+        p[j] = in_range && (fabsf(x[j]) >= boundary);
+        cnt += p[j] ? 1 : 0;
+      }
     }
-    const uint64_t m = __ballot(pred);
-    const long long wave_cnt = __popcll(m);
-    const long long lane_pre = __popcll(m & ((1ull << lane) - 1ull));
-    if (lane == 0) wave_base[wid] = wave_cnt;
+    // wave-inclusive scan of per-thread counts
+    int incl = cnt;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      const int y = __shfl_up(incl, off, 64);
+      if (lane >= off) incl += y;
+    }
+    const int excl = incl - cnt;
+    const int wave_total = __shfl(incl, 63, 64);
+    if (lane == 63) wave_base[wid] = incl;  // wave total
     __syncthreads();
-    // exclusive scan of wave counts + add carry (one thread)
     if (threadIdx.x == 0) {
-      long long acc = carry;
+      long long acc = carry_s;
       for (int w = 0; w < nwaves; ++w) {
         const long long c = wave_base[w];
         wave_base[w] = acc;
         acc += c;
       }
-      carry = acc;
+      carry_s = acc;
     }
     __syncthreads();
-    if (pred) {
-      const long long pos = wave_base[wid] + lane_pre;
-      if (pos < capacity) {
-        vals[pos] = x;
-        idx[pos] = (int)i;
-        if (ZERO_UV) {
-          v_mut[i] = 0.0f;
-          u_mut[i] = 0.0f;
+    long long pos = wave_base[wid] + excl;
+    (void)wave_total;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (p[j]) {
+        if (pos < capacity) {
+          const long long i = i0 + j;
+          vals[pos] = x[j];
+          idx[pos] = (int)i;
+          if (ZERO_UV) {
+            v_mut[i] = 0.0f;
+            u_mut[i] = 0.0f;
+          }
         }
+        ++pos;
       }
     }
     __syncthreads();
@@ -277,7 +339,14 @@ extern "C" __global__ void k_dgt_contribution(const float* __restrict__ g,
     const long long lo = (long long)c * chunk;
     const long long hi = min(lo + chunk, n);
     float s = 0.0f;
-    for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x)
+    const long long nvec = (hi - lo) >> 2;
+    const float4* g4 = reinterpret_cast<const float4*>(g + lo);
+    for (long long q = threadIdx.x; q < nvec; q += blockDim.x) {
+      const float4 v = g4[q];
+      s += fabsf(v.x) + fabsf(v.y) + fabsf(v.z) + fabsf(v.w);
+    }
+    for (long long i = lo + (nvec << 2) + threadIdx.x; i < hi;
+         i += blockDim.x)
       s += fabsf(g[i]);
     for (int off = 32; off > 0; off >>= 1)
       s += __shfl_down(s, off, 64);
@@ -309,7 +378,20 @@ extern "C" __global__ void k_minmax_chunk(const float* __restrict__ x,
     const long long lo = (long long)c * chunk;
     const long long hi = min(lo + chunk, n);
     float mn = 3.4e38f, mx = -3.4e38f;
-    for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    const long long nvec = (hi - lo) >> 2;
+    const float4* x4 = reinterpret_cast<const float4*>(x + lo);
+    const float4* r4 = reinterpret_cast<const float4*>(residual + lo);
+    for (long long q = threadIdx.x; q < nvec; q += blockDim.x) {
+      float4 f = x4[q];
+      if (has_res) {
+        const float4 r = r4[q];
+        f.x += r.x; f.y += r.y; f.z += r.z; f.w += r.w;
+      }
+      mn = fminf(fminf(fminf(mn, f.x), f.y), fminf(f.z, f.w));
+      mx = fmaxf(fmaxf(fmaxf(mx, f.x), f.y), fmaxf(f.z, f.w));
+    }
+    for (long long i = lo + (nvec << 2) + threadIdx.x; i < hi;
+         i += blockDim.x) {
       const float f = x[i] + (has_res ? residual[i] : 0.0f);
       mn = fminf(mn, f);
       mx = fmaxf(mx, f);
@@ -333,46 +415,88 @@ extern "C" __global__ void k_minmax_chunk(const float* __restrict__ x,
   }
 }
 
+// chunk must be a power of two on the GPU path (log2c = log2(chunk));
+// each thread packs 4 elements (reads float4, writes one u16).
 extern "C" __global__ void k_quantize_4bit(const float* __restrict__ x,
                                            float* __restrict__ residual,
                                            uint8_t* __restrict__ out,
                                            const float* __restrict__ minmax,
-                                           long long n, int chunk,
+                                           long long n, int log2c,
                                            bool has_res) {
-  // each thread packs one output byte (2 elements)
-  const long long nbytes = (n + 1) >> 1;
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
-  for (long long b = tid; b < nbytes; b += stride) {
-    uint8_t byte = 0;
-    for (int h = 0; h < 2; ++h) {
-      const long long i = 2 * b + h;
-      if (i >= n) break;
-      const int c = (int)(i / chunk);
-      const float lo = minmax[2 * c];
-      const float span = fmaxf(minmax[2 * c + 1] - lo, 1e-30f);
-      const float step = span / 16.0f;
-      const float f = x[i] + (has_res ? residual[i] : 0.0f);
-      int code = (int)floorf((f - lo) / step);
-      code = code < 0 ? 0 : (code > 15 ? 15 : code);
-      if (has_res) residual[i] = f - (lo + ((float)code + 0.5f) * step);
-      byte |= (uint8_t)(code << (4 * h));
+  const long long nq = n >> 2;
+  uint16_t* out16 = reinterpret_cast<uint16_t*>(out);
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  float4* r4 = reinterpret_cast<float4*>(residual);
+  for (long long q = tid; q < nq; q += stride) {
+    const int c = (int)((q << 2) >> log2c);  // chunk >= 64: one chunk per q
+    const float lo = minmax[2 * c];
+    const float step = fmaxf(minmax[2 * c + 1] - lo, 1e-30f) / 16.0f;
+    const float inv = 1.0f / step;
+    const float4 xx = x4[q];
+    float f[4] = {xx.x, xx.y, xx.z, xx.w};
+    if (has_res) {
+      const float4 rr = r4[q];
+      f[0] += rr.x; f[1] += rr.y; f[2] += rr.z; f[3] += rr.w;
     }
-    out[b] = byte;
+    uint16_t word = 0;
+    float rout[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int code = (int)floorf((f[j] - lo) * inv);
+      code = code < 0 ? 0 : (code > 15 ? 15 : code);
+      rout[j] = f[j] - (lo + ((float)code + 0.5f) * step);
+      word |= (uint16_t)(code << (4 * j));
+    }
+    if (has_res) r4[q] = make_float4(rout[0], rout[1], rout[2], rout[3]);
+    out16[q] = word;
+  }
+  // scalar tail (< 4 elements): one thread packs the remaining bytes
+  if (tid == 0) {
+    for (long long i = (nq << 2); i < n; i += 2) {
+      uint8_t byte = 0;
+      for (int h = 0; h < 2 && i + h < n; ++h) {
+        const long long ii = i + h;
+        const int c = (int)(ii >> log2c);
+        const float lo = minmax[2 * c];
+        const float step = fmaxf(minmax[2 * c + 1] - lo, 1e-30f) / 16.0f;
+        const float f = x[ii] + (has_res ? residual[ii] : 0.0f);
+        int code = (int)floorf((f - lo) / step);
+        code = code < 0 ? 0 : (code > 15 ? 15 : code);
+        if (has_res) residual[ii] = f - (lo + ((float)code + 0.5f) * step);
+        byte |= (uint8_t)(code << (4 * h));
+      }
+      out[i >> 1] = byte;
+    }
   }
 }
 
 extern "C" __global__ void k_dequantize_4bit(const uint8_t* __restrict__ in,
                                              const float* __restrict__ minmax,
                                              float* __restrict__ out,
-                                             long long n, int chunk) {
+                                             long long n, int log2c) {
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
-  for (long long i = tid; i < n; i += stride) {
-    const int c = (int)(i / chunk);
+  const long long nq = n >> 2;
+  const uint16_t* in16 = reinterpret_cast<const uint16_t*>(in);
+  float4* out4 = reinterpret_cast<float4*>(out);
+  for (long long q = tid; q < nq; q += stride) {
+    const int c = (int)((q << 2) >> log2c);
     const float lo = minmax[2 * c];
-    const float span = fmaxf(minmax[2 * c + 1] - lo, 1e-30f);
-    const float step = span / 16.0f;
+    const float step = fmaxf(minmax[2 * c + 1] - lo, 1e-30f) / 16.0f;
+    const uint16_t w = in16[q];
+    float4 o;
+    o.x = lo + ((float)(w & 0xF) + 0.5f) * step;
+    o.y = lo + ((float)((w >> 4) & 0xF) + 0.5f) * step;
+    o.z = lo + ((float)((w >> 8) & 0xF) + 0.5f) * step;
+    o.w = lo + ((float)((w >> 12) & 0xF) + 0.5f) * step;
+    out4[q] = o;
+  }
+  for (long long i = (nq << 2) + tid; i < n; i += stride) {
+    const int c = (int)(i >> log2c);
+    const float lo = minmax[2 * c];
+    const float step = fmaxf(minmax[2 * c + 1] - lo, 1e-30f) / 16.0f;
     const uint8_t byte = in[i >> 1];
     const int code = (i & 1) ? (byte >> 4) : (byte & 0x0F);
     out[i] = lo + ((float)code + 0.5f) * step;
@@ -583,21 +707,29 @@ void geops_dgt_contribution(const float* g, float* out, long long n,
                      0, s, g, out, n, chunk, nchunks);
 }
 
+static int ilog2(int x) {
+  int l = 0;
+  while ((1 << l) < x) ++l;
+  return l;
+}
+
+// chunk must be a power of two (enforced by the binding)
 void geops_quantize_4bit(const float* x, float* residual, uint8_t* out,
                          float* minmax, long long n, int chunk, int nchunks,
                          bool has_res, hipStream_t s) {
   int nb = nchunks < GEOPS_MAX_BLOCKS ? nchunks : GEOPS_MAX_BLOCKS;
   hipLaunchKernelGGL(k_minmax_chunk, dim3(nb), dim3(GEOPS_THREADS), 0, s,
                      x, residual, minmax, n, chunk, nchunks, has_res);
-  hipLaunchKernelGGL(k_quantize_4bit, dim3(geops_blocks((n + 1) / 2)),
+  hipLaunchKernelGGL(k_quantize_4bit, dim3(geops_blocks(n, 4)),
                      dim3(GEOPS_THREADS), 0, s, x, residual, out, minmax, n,
-                     chunk, has_res);
+                     ilog2(chunk), has_res);
 }
 
 void geops_dequantize_4bit(const uint8_t* in, const float* minmax, float* out,
                            long long n, int chunk, hipStream_t s) {
-  hipLaunchKernelGGL(k_dequantize_4bit, dim3(geops_blocks(n)),
-                     dim3(GEOPS_THREADS), 0, s, in, minmax, out, n, chunk);
+  hipLaunchKernelGGL(k_dequantize_4bit, dim3(geops_blocks(n, 4)),
+                     dim3(GEOPS_THREADS), 0, s, in, minmax, out, n,
+                     ilog2(chunk));
 }
 
 void geops_sgd_update(float* w, const float* g, float lr, float wd,

@@ -243,3 +243,36 @@ def test_sgd_update_matches_reference():
     ops.sgd_update(w_g, g.to(DEV), 0.1, 1e-4, 2.0)
     ref.sgd_update(w_c, g, 0.1, 1e-4, 2.0)
     assert torch.allclose(w_g.cpu(), w_c, atol=1e-6)
+
+
+def test_quantize_4bit_ragged_n():
+    torch.manual_seed(9)
+    n = 4099  # non-multiple of 4: exercises the serial tail
+    x = torch.randn(n)
+    p_gpu, mm_gpu = ops.quantize_4bit_chunked(x.to(DEV), 1024)
+    p_cpu, mm_cpu = ops.quantize_4bit_chunked(x, 1024)
+    assert torch.allclose(mm_gpu.cpu(), mm_cpu, atol=1e-6)
+    assert torch.equal(p_gpu.cpu(), p_cpu)
+    y_gpu = ops.dequantize_4bit_chunked(p_gpu, mm_gpu, n, 1024)
+    y_cpu = ops.dequantize_4bit_chunked(p_cpu, mm_cpu, n, 1024)
+    assert torch.allclose(y_gpu.cpu(), y_cpu, atol=1e-6)
+
+
+def test_bsc_pack_ragged_sizes():
+    for n in [16387, 65536 + 5]:
+        g = torch.randn(n)
+        boundary = 1.0
+        from geomx_amd.ops import _geops as geops
+        u_g = torch.zeros(n, device=DEV)
+        v_g = g.clone().to(DEV)
+        k = max(16, int(n * 0.1))
+        vals_g = torch.empty(k, device=DEV)
+        idx_g = torch.empty(k, dtype=torch.int32, device=DEV)
+        geops.bsc_pack(v_g, u_g, vals_g, idx_g, boundary, ref.BSC_PLACEHOLDER)
+        u_c = torch.zeros(n)
+        v_c = g.clone()
+        mask = v_c.abs() >= boundary
+        sel = mask.nonzero().flatten()[:k]
+        assert torch.equal(idx_g.cpu()[:sel.numel()].long(), sel)
+        assert torch.allclose(vals_g.cpu()[:sel.numel()], g[sel], atol=1e-6)
+        assert (idx_g.cpu()[sel.numel():] == -1).all()
