@@ -40,6 +40,11 @@ void geops_adam_update(float*, const float*, float*, float*, float, float,
                        float, float, float, float, long long, hipStream_t);
 void geops_dcasgd_update(float*, const float*, float*, float*, float, float,
                          float, float, float, long long, bool, hipStream_t);
+void geops_relu_maxpool2_fwd(const unsigned short*, unsigned short*, uint8_t*,
+                             long long, int, int, int, hipStream_t);
+void geops_relu_maxpool2_bwd(const unsigned short*, const uint8_t*,
+                             unsigned short*, long long, int, int, int,
+                             hipStream_t);
 }
 
 namespace {
@@ -170,6 +175,35 @@ void dequantize_4bit(torch::Tensor packed, torch::Tensor minmax,
                         cur_stream());
 }
 
+void relu_maxpool2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor idx,
+                       int64_t N, int64_t C, int64_t Hi, int64_t Wi) {
+  TORCH_CHECK(in.is_cuda() && out.is_cuda() && idx.is_cuda());
+  TORCH_CHECK(in.scalar_type() == torch::kBFloat16 &&
+              out.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(idx.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(C % 8 == 0 && Hi % 2 == 0 && Wi % 2 == 0,
+              "relu_maxpool2 needs C%8==0 and even H,W");
+  const int Ho = (int)(Hi / 2), Wo = (int)(Wi / 2);
+  const long long n_vec = (long long)N * Ho * Wo * (C / 8);
+  geops_relu_maxpool2_fwd((const unsigned short*)in.data_ptr(),
+                          (unsigned short*)out.data_ptr(),
+                          idx.data_ptr<uint8_t>(), n_vec, Ho, Wo, (int)C,
+                          cur_stream());
+}
+
+void relu_maxpool2_bwd(torch::Tensor grad_out, torch::Tensor idx,
+                       torch::Tensor grad_in, int64_t N, int64_t C,
+                       int64_t Hi, int64_t Wi) {
+  TORCH_CHECK(grad_out.is_cuda() && idx.is_cuda() && grad_in.is_cuda());
+  TORCH_CHECK(grad_out.scalar_type() == torch::kBFloat16 &&
+              grad_in.scalar_type() == torch::kBFloat16);
+  const long long n_vec_in = (long long)N * Hi * Wi * (C / 8);
+  geops_relu_maxpool2_bwd((const unsigned short*)grad_out.data_ptr(),
+                          idx.data_ptr<uint8_t>(),
+                          (unsigned short*)grad_in.data_ptr(), n_vec_in,
+                          (int)Hi, (int)Wi, (int)C, cur_stream());
+}
+
 void sgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
                 double rescale) {
   check_f32(w, "w"); check_f32(g, "g");
@@ -225,6 +259,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dgt_contribution", &dgt_contribution);
   m.def("quantize_4bit", &quantize_4bit);
   m.def("dequantize_4bit", &dequantize_4bit);
+  m.def("relu_maxpool2_fwd", &relu_maxpool2_fwd);
+  m.def("relu_maxpool2_bwd", &relu_maxpool2_bwd);
   m.def("sgd_update", &sgd_update);
   m.def("sgd_mom_update", &sgd_mom_update);
   m.def("adam_update", &adam_update);

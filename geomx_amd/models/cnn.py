@@ -19,13 +19,12 @@ class GeoCNN(nn.Module):
     def __init__(self, in_channels: int = 3, image_size: int = 224,
                  num_classes: int = 10):
         super().__init__()
+        from ..ops.fused import FusedReLUPool2
         self.features = nn.Sequential(
             nn.Conv2d(in_channels, 16, kernel_size=5),
-            nn.ReLU(inplace=True),
-            nn.MaxPool2d(2, 2),
+            FusedReLUPool2(),   # relu+maxpool in one gfx950 kernel on GPU
             nn.Conv2d(16, 32, kernel_size=5),
-            nn.ReLU(inplace=True),
-            nn.MaxPool2d(2, 2),
+            FusedReLUPool2(),
         )
         with torch.no_grad():
             probe = torch.zeros(1, in_channels, image_size, image_size)

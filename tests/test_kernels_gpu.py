@@ -276,3 +276,32 @@ def test_bsc_pack_ragged_sizes():
         assert torch.equal(idx_g.cpu()[:sel.numel()].long(), sel)
         assert torch.allclose(vals_g.cpu()[:sel.numel()], g[sel], atol=1e-6)
         assert (idx_g.cpu()[sel.numel():] == -1).all()
+
+
+def test_fused_relu_pool_matches_aten():
+    from geomx_amd.ops.fused import FusedReLUPool2, _ReLUPool2Fn
+    torch.manual_seed(11)
+    x = torch.randn(4, 16, 20, 24, device=DEV, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = _ReLUPool2Fn.apply(x)
+    ref_x = x.detach().clone().requires_grad_(True)
+    ref_y = torch.nn.functional.max_pool2d(
+        torch.nn.functional.relu(ref_x), 2, 2)
+    assert torch.equal(y, ref_y)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref_y.backward(g)
+    # backward can differ on exact ties; random bf16 ties are measure-zero
+    assert torch.equal(x.grad, ref_x.grad)
+
+
+def test_fused_relu_pool_in_model():
+    from geomx_amd.models import create_model
+    m = create_model("geomx_cnn", image_size=64).to(DEV) \
+        .to(memory_format=torch.channels_last)
+    x = torch.randn(8, 3, 64, 64, device=DEV) \
+        .to(memory_format=torch.channels_last)
+    with torch.autocast("cuda", torch.bfloat16):
+        loss = m(x).float().square().mean()
+    loss.backward()
+    assert torch.isfinite(loss).item()
