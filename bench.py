@@ -55,6 +55,7 @@ def parse_args():
     p.add_argument("--bucket-mb", type=int, default=25)
     p.add_argument("--optimizer", type=str, default="sgd_mom")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--comm-dtype", type=str, default="fp32", choices=["fp32", "bf16"])
     p.add_argument("--json-out", type=str, default=None)
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout (NHWC avoids MIOpen's "
@@ -82,7 +83,8 @@ def main():
     cfg = Config.from_env(
         num_parties=parties, backend=backend,
         compression=args.compress, bsc_ratio=args.bsc_ratio,
-        wan_gbps=args.wan_gbps, bucket_mb=args.bucket_mb)
+        wan_gbps=args.wan_gbps, bucket_mb=args.bucket_mb,
+        comm_dtype=args.comm_dtype)
     topo = init_topology(topo_parties, None, backend)
     device = topo.device
 

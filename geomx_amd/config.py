@@ -100,6 +100,10 @@ class Config:
     # kvstore_dist.h:69,763-799; MXNET_KVSTORE_BIGARRAY_BOUND)
     bigarray_bound: int = 1_000_000
     bucket_mb: int = 25          # gradient bucket size for fused collectives
+    # wire dtype for the flat-mode bucket all_reduce: "fp32" (exact) or
+    # "bf16" (half traffic — the FP16-transmission feature applied to the
+    # fast path; server/optimizer math stays fp32)
+    comm_dtype: str = "fp32"
     overlap: bool = True         # overlap comm with backward
     backend: Optional[str] = None  # override; default nccl on GPU, gloo on CPU
     device: Optional[str] = None

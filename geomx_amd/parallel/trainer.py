@@ -64,6 +64,20 @@ class _Bucket:
         self.dgt = None
 
 
+class _Bf16Work:
+    """Wraps an in-flight bf16 all_reduce; wait() casts back into the
+    fp32 bucket."""
+
+    def __init__(self, work, h, flat):
+        self.work = work
+        self.h = h
+        self.flat = flat
+
+    def wait(self):
+        self.work.wait()
+        self.flat.copy_(self.h.float())
+
+
 def _alias_view(flat: torch.Tensor, off: int, p: torch.Tensor) -> torch.Tensor:
     """A view of flat[off:off+numel] shaped like p, preserving p's memory
     format (channels_last conv weights keep NHWC strides so MIOpen never
@@ -159,7 +173,14 @@ class GeoTrainer:
             return
         if self.mode == "flat":
             b.flat.div_(self.topo.world_size)
-            b.work = dist.all_reduce(b.flat, async_op=True)
+            if self.cfg.comm_dtype == "bf16":
+                # half-traffic wire format (FP16-transmission analog);
+                # RCCL sums in bf16, result cast back for the fp32 update
+                h = b.flat.to(torch.bfloat16)
+                work = dist.all_reduce(h, async_op=True)
+                b.work = _Bf16Work(work, h, b.flat)
+            else:
+                b.work = dist.all_reduce(b.flat, async_op=True)
         else:
             b.flat.div_(self.topo.num_all_workers)
             b.work = dist.all_reduce(b.flat, group=self.topo.party_group,

@@ -141,3 +141,28 @@ def _trainer_hips_dgt(rank, world):
 
 def test_hips_dgt_consistent_ws4():
     run_dist(4, _trainer_hips_dgt)
+
+
+def _trainer_bf16_comm(rank, world):
+    cfg = Config.from_env(backend="gloo", device="cpu", bucket_mb=1,
+                          comm_dtype="bf16")
+    topo = init_topology(1, None, "gloo", "cpu")
+    model = _tiny_model()
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec(name="sgd", lr=0.05),
+                    mode="flat")
+    for s in range(2):
+        x, y = _make_data(seed=50 + s)
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+    import torch.distributed as dist
+    for p in model.parameters():
+        assert torch.isfinite(p.data).all()
+        ref = p.data.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(p.data, ref)
+
+
+def test_bf16_comm_ws2():
+    run_dist(2, _trainer_bf16_comm)
