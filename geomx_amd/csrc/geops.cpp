@@ -45,6 +45,9 @@ void geops_relu_maxpool2_fwd(const unsigned short*, unsigned short*, uint8_t*,
 void geops_relu_maxpool2_bwd(const unsigned short*, const uint8_t*,
                              unsigned short*, long long, int, int, int,
                              hipStream_t);
+int geops_conv5_nhwc(const unsigned short*, const unsigned short*,
+                     const float*, unsigned short*, int, int, int, int, int,
+                     int, int, int, hipStream_t);
 }
 
 namespace {
@@ -204,6 +207,27 @@ void relu_maxpool2_bwd(torch::Tensor grad_out, torch::Tensor idx,
                           (int)Hi, (int)Wi, (int)C, cur_stream());
 }
 
+void conv5_nhwc(torch::Tensor in, torch::Tensor w_frags, torch::Tensor bias,
+                torch::Tensor out, int64_t N, int64_t Hi, int64_t Wi,
+                int64_t Ho, int64_t Wo, int64_t CI, int64_t CO,
+                int64_t pad) {
+  TORCH_CHECK(in.is_cuda() && w_frags.is_cuda() && out.is_cuda());
+  TORCH_CHECK(in.scalar_type() == torch::kBFloat16 &&
+              w_frags.scalar_type() == torch::kBFloat16 &&
+              out.scalar_type() == torch::kBFloat16);
+  const bool has_bias = bias.defined() && bias.numel() > 0;
+  if (has_bias)
+    TORCH_CHECK(bias.scalar_type() == torch::kFloat32 && bias.is_cuda());
+  const int rc = geops_conv5_nhwc(
+      (const unsigned short*)in.data_ptr(),
+      (const unsigned short*)w_frags.data_ptr(),
+      has_bias ? bias.data_ptr<float>() : nullptr,
+      (unsigned short*)out.data_ptr(), (int)N, (int)Hi, (int)Wi, (int)Ho,
+      (int)Wo, (int)CI, (int)CO, (int)pad, cur_stream());
+  TORCH_CHECK(rc == 0, "conv5_nhwc: unsupported geometry CI=", CI,
+              " CO=", CO, " pad=", pad);
+}
+
 void sgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
                 double rescale) {
   check_f32(w, "w"); check_f32(g, "g");
@@ -259,6 +283,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dgt_contribution", &dgt_contribution);
   m.def("quantize_4bit", &quantize_4bit);
   m.def("dequantize_4bit", &dequantize_4bit);
+  m.def("conv5_nhwc", &conv5_nhwc);
   m.def("relu_maxpool2_fwd", &relu_maxpool2_fwd);
   m.def("relu_maxpool2_bwd", &relu_maxpool2_bwd);
   m.def("sgd_update", &sgd_update);

@@ -19,11 +19,12 @@ class GeoCNN(nn.Module):
     def __init__(self, in_channels: int = 3, image_size: int = 224,
                  num_classes: int = 10):
         super().__init__()
+        from ..ops.conv import GeoConv5
         from ..ops.fused import FusedReLUPool2
         self.features = nn.Sequential(
-            nn.Conv2d(in_channels, 16, kernel_size=5),
+            GeoConv5(in_channels, 16),   # MFMA direct conv on GPU
             FusedReLUPool2(),   # relu+maxpool in one gfx950 kernel on GPU
-            nn.Conv2d(16, 32, kernel_size=5),
+            GeoConv5(16, 32),
             FusedReLUPool2(),
         )
         with torch.no_grad():

@@ -1,0 +1,172 @@
+"""GeoConv5: 5x5 stride-1 convolution on the hand-written gfx950 MFMA
+kernel (csrc/conv.hip) for the flagship CNN's shapes.
+
+Forward and the data gradient run on the custom implicit-GEMM kernel
+(NHWC bf16, weights pre-packed into per-lane MFMA B-fragment order);
+the weight/bias gradients go through ATen's convolution_backward.
+Anything outside the supported geometry falls back to F.conv2d.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import native_available
+
+_SUPPORTED_FWD = {(4, 16), (16, 32), (16, 16), (32, 32)}
+_SUPPORTED_DGRAD = {(32, 16), (16, 16)}
+
+
+def _frag_index(CO: int, CI: int, numel: int, entry) -> torch.Tensor:
+    """Build the gather index mapping weight.flatten() -> w_frags layout
+    [CO/16][nK][64][8]; index `numel` means 'zero' (the caller appends a
+    zero element to the flat weight). `entry(o, k)` returns the flat
+    weight index for output channel o, im2col position k, or None."""
+    S = 5 * CI
+    Sp = (S + 7) & ~7
+    K = 5 * Sp
+    nK = (K + 31) // 32
+    cot = CO // 16
+    idx = torch.full((cot, nK, 64, 8), numel, dtype=torch.int64)
+    for ct in range(cot):
+        for km in range(nK):
+            for lane in range(64):
+                q, n = lane >> 4, lane & 15
+                o = ct * 16 + n
+                for j in range(8):
+                    k = km * 32 + q * 8 + j
+                    if k >= K:
+                        continue
+                    kh, jj = divmod(k, Sp)
+                    if jj >= S or kh >= 5:
+                        continue
+                    kw, ci = divmod(jj, CI)
+                    e = entry(o, kh, kw, ci)
+                    if e is not None:
+                        idx[ct, km, lane, j] = e
+    return idx.reshape(-1)
+
+
+def build_fwd_index(weight_shape) -> torch.Tensor:
+    """weight [CO][CIr][5][5] (torch OIHW), CI zero-padded to mult of 4."""
+    CO, CIr, KH, KW = weight_shape
+    CI = (CIr + 3) & ~3
+    numel = CO * CIr * KH * KW
+
+    def entry(o, kh, kw, ci):
+        if ci >= CIr:
+            return None
+        return ((o * CIr + ci) * KH + kh) * KW + kw
+
+    return _frag_index(CO, CI, numel, entry)
+
+
+def build_dgrad_index(weight_shape) -> torch.Tensor:
+    """Data-grad pass: roles swap (CI'=CO, CO'=CIr padded to 16) and the
+    kernel is spatially flipped: W'[o'=ci][kh][kw][ci'=o] =
+    W[o][ci][4-kh][4-kw]."""
+    CO, CIr, KH, KW = weight_shape
+    COp = (CIr + 15) & ~15
+    numel = CO * CIr * KH * KW
+
+    def entry(op, kh, kw, cip):
+        # op: output channel of the dgrad conv = original input channel
+        if op >= CIr:
+            return None
+        o, ci = cip, op
+        return ((o * CIr + ci) * KH + (4 - kh)) * KW + (4 - kw)
+
+    return _frag_index(COp, CO, numel, entry)
+
+
+class _Conv5Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, fwd_idx, dgrad_idx):
+        from geomx_amd import _geops
+        N, CIr, Hi, Wi = x.shape
+        CO = weight.shape[0]
+        CI = (CIr + 3) & ~3
+        Ho, Wo = Hi - 4, Wi - 4
+        xb = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+        xs = xb  # saved for the weight-grad pass (pre-padding)
+        if CI != CIr:  # zero-pad channels (conv1: 3 -> 4)
+            pad = torch.zeros(N, CI - CIr, Hi, Wi, dtype=torch.bfloat16,
+                              device=x.device)
+            xb = torch.cat([xb, pad], dim=1) \
+                .contiguous(memory_format=torch.channels_last)
+        wb = weight.detach().to(torch.bfloat16).reshape(-1)
+        wz = torch.cat([wb, wb.new_zeros(1)])
+        w_frags = wz[fwd_idx].contiguous()
+        out = torch.empty(N, CO, Ho, Wo, dtype=torch.bfloat16,
+                          device=x.device,
+                          memory_format=torch.channels_last)
+        b = bias.detach().float() if bias is not None else torch.Tensor()
+        _geops.conv5_nhwc(xb, w_frags, b, out, N, Hi, Wi, Ho, Wo, CI, CO, 0)
+        ctx.save_for_backward(xs, weight, dgrad_idx)
+        ctx.has_bias = bias is not None
+        ctx.dims = (N, CIr, CI, CO, Hi, Wi, Ho, Wo)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        from geomx_amd import _geops
+        x, weight, dgrad_idx = ctx.saved_tensors
+        N, CIr, CI, CO, Hi, Wi, Ho, Wo = ctx.dims
+        go = grad_out.to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        grad_x = None
+        if ctx.needs_input_grad[0]:
+            COp = (CIr + 15) & ~15
+            wb = weight.detach().to(torch.bfloat16).reshape(-1)
+            wz = torch.cat([wb, wb.new_zeros(1)])
+            w_frags = wz[dgrad_idx].contiguous()
+            gx = torch.empty(N, COp, Hi, Wi, dtype=torch.bfloat16,
+                             device=x.device,
+                             memory_format=torch.channels_last)
+            _geops.conv5_nhwc(go, w_frags, torch.Tensor(), gx, N, Ho, Wo,
+                              Hi, Wi, CO, COp, 4)
+            grad_x = gx[:, :CIr] if COp != CIr else gx
+        grad_w = grad_b = None
+        if ctx.needs_input_grad[1] or (ctx.has_bias and
+                                       ctx.needs_input_grad[2]):
+            gi, gw, gb = torch.ops.aten.convolution_backward(
+                go, x, weight.to(torch.bfloat16),
+                [CO] if ctx.has_bias else None,
+                [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+                [False, True, ctx.has_bias])
+            grad_w = gw.to(weight.dtype)
+            grad_b = gb.to(weight.dtype) if ctx.has_bias else None
+        return grad_x, grad_w, grad_b, None, None
+
+
+class GeoConv5(torch.nn.Conv2d):
+    """nn.Conv2d drop-in (kernel 5, stride 1, pad 0) running on the
+    gfx950 MFMA direct-conv kernel when eligible."""
+
+    def __init__(self, in_channels, out_channels, **kw):
+        super().__init__(in_channels, out_channels, kernel_size=5, **kw)
+        self._fwd_idx = None
+        self._dgrad_idx = None
+
+    def _eligible(self, x) -> bool:
+        CI = (self.in_channels + 3) & ~3
+        dgrad_ok = (not x.requires_grad) or \
+            ((self.out_channels, (self.in_channels + 15) & ~15)
+             in _SUPPORTED_DGRAD)
+        return (x.is_cuda and native_available()
+                and (CI, self.out_channels) in _SUPPORTED_FWD
+                and dgrad_ok
+                and self.stride == (1, 1) and self.padding == (0, 0)
+                and self.kernel_size == (5, 5) and self.groups == 1)
+
+    def forward(self, x):
+        if not self._eligible(x):
+            return super().forward(x)
+        if self._fwd_idx is None or self._fwd_idx.device != x.device:
+            self._fwd_idx = build_fwd_index(self.weight.shape).to(x.device)
+            self._dgrad_idx = build_dgrad_index(self.weight.shape).to(x.device)
+        return _Conv5Fn.apply(x, self.weight, self.bias, self._fwd_idx,
+                              self._dgrad_idx)

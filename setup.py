@@ -21,6 +21,7 @@ ext = CUDAExtension(
     sources=[
         "geomx_amd/csrc/geops.cpp",
         "geomx_amd/csrc/kernels.hip",
+        "geomx_amd/csrc/conv.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
