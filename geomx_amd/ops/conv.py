@@ -123,11 +123,16 @@ class _Conv5Fn(torch.autograd.Function):
             wb = weight.detach().to(torch.bfloat16).reshape(-1)
             wz = torch.cat([wb, wb.new_zeros(1)])
             w_frags = wz[dgrad_idx].contiguous()
+            # physical zero-pad of grad_out: lets the dgrad run on the
+            # interior-only kernel (the in-kernel masked path cost 256
+            # VGPRs -> 1 wave/SIMD)
+            gop = F.pad(go, (4, 4, 4, 4)) \
+                .contiguous(memory_format=torch.channels_last)
             gx = torch.empty(N, COp, Hi, Wi, dtype=torch.bfloat16,
                              device=x.device,
                              memory_format=torch.channels_last)
-            _geops.conv5_nhwc(go, w_frags, torch.Tensor(), gx, N, Ho, Wo,
-                              Hi, Wi, CO, COp, 4)
+            _geops.conv5_nhwc(gop, w_frags, torch.Tensor(), gx, N, Ho + 8,
+                              Wo + 8, Hi, Wi, CO, COp, 0)
             grad_x = gx[:, :CIr] if COp != CIr else gx
         grad_w = grad_b = None
         if ctx.needs_input_grad[1] or (ctx.has_bias and
