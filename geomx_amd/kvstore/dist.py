@@ -482,6 +482,23 @@ class KVStoreDist(KVStoreBase):
         # placeholder for party-internal consistency; pull() broadcasts.
         pass
 
+    def row_sparse_pull(self, key, out: torch.Tensor, row_ids: torch.Tensor,
+                        priority: int = 0) -> None:
+        """Pull only the rows named by row_ids (python/mxnet/kvstore.py:316;
+        GPU row-id dedup was cub Unique in the reference,
+        kvstore_utils.cu:44-111 — torch.unique is the rocPRIM-backed
+        equivalent). `out` must be [len(row_ids), row_width]."""
+        st = self._state(key)
+        if len(st.shape) < 2:
+            raise ValueError("row_sparse_pull needs a >=2d key")
+        rows = st.shape[0]
+        width = st.numel // rows
+        full = torch.empty(st.shape, dtype=out.dtype, device=self._device)
+        self.pull(key, full, priority)
+        sel = full.reshape(rows, width)[row_ids.long()]
+        with torch.no_grad():
+            out.reshape(len(row_ids), width).copy_(sel)
+
     # ------------------------------------------------------------------
     # checkpointing (layout parity: named-param dict + separate optimizer
     # state blob — gluon/block.py:315,356 + kvstore.py:566-592)

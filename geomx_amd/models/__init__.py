@@ -7,5 +7,6 @@ def create_model(name: str, **kwargs):
     if name in ("geomx_cnn", "geomx-cnn", "cnn"):
         return geo_cnn(**kwargs)
     if name in ("resnet50", "resnet-50"):
+        kwargs.pop("image_size", None)  # ResNet is size-agnostic
         return resnet50(**kwargs)
     raise ValueError(f"unknown model {name!r}")

@@ -110,3 +110,23 @@ def test_set_gradient_compression_validation():
 def test_mx_kv_alias():
     kv = geomx_amd.kv.create("dist_sync")
     assert kv.type == "dist_sync"
+
+
+def test_row_sparse_pull():
+    kv = make_kv()
+    w = torch.arange(20, dtype=torch.float32).reshape(5, 4)
+    kv.init("emb", w)
+    kv.push("emb", torch.zeros(5, 4))
+    out = torch.empty(3, 4)
+    kv.row_sparse_pull("emb", out, torch.tensor([0, 2, 2]))
+    # no optimizer: pull returns aggregated grad (zeros here)
+    assert torch.all(out == 0)
+    kv2 = make_kv()
+    from geomx_amd.kvstore.optimizer import OptimizerSpec
+    kv2.set_optimizer(OptimizerSpec("sgd", lr=0.0))  # identity update
+    kv2.init("emb", w)
+    kv2.push("emb", torch.zeros(5, 4))
+    kv2.row_sparse_pull("emb", out, torch.tensor([1, 4, 4]))
+    assert torch.allclose(out[0], w[1])
+    assert torch.allclose(out[1], w[4])
+    assert torch.allclose(out[2], w[4])
