@@ -82,77 +82,81 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_nhwc(
       ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const long long n_waves = ((long long)gridDim.x * blockDim.x) >> 6;
 
+  // per-lane k-run constants (kh, j0 depend only on q and km)
   for (long long row = wave_id; row < n_rows; row += n_waves) {
     const int ho = (int)(row % Ho);
     const long long n = row / Ho;
     const long long in_n = (long long)n * Hi * Wi * CI;
     const long long out_row = ((long long)n * Ho + ho) * (long long)Wo * CO;
 
-    for (int tw = 0; tw < tiles_w; ++tw) {
-      const int wo0 = tw << 4;
-      const int wo_raw = wo0 + p;
-      // out-of-tile lanes clamp to a valid pixel; their D rows are
-      // masked at the store
-      const bool full = (wo0 + 16 <= Wo);
+    // chunked fragment loader (PAD==0 kernels: loads are always
+    // in-bounds once the lane's pixel is clamped to Wo-1, because
+    // (Wo-1) + 5 taps == Wi exactly). CH frags per chunk keeps the
+    // double-buffer register cost at 2*CH*4 VGPRs regardless of nK.
+    constexpr int CH = (nK >= 8) ? 4 : nK;
+    constexpr int nCH = (nK + CH - 1) / CH;
+    auto load_chunk = [&](int tw, int c, bf16x8(&dst)[CH]) {
+      const int wo_raw = (tw << 4) + p;
       const int wo = wo_raw < Wo ? wo_raw : (Wo - 1);
-      const int woM = full ? (wo0 + 15) : (Wo - 1);
-      // PAD == 0: the clamped wo makes every vector load in-bounds
-      // ((Wo-1)+5 == Wi), so the masked path compiles out entirely
-      const bool interior =
-          (PAD == 0) || ((ho >= PAD) && (ho - PAD + 4 < Hi) &&
-                         (wo0 >= PAD) && (woM - PAD + 5 <= Wi));
+      const long long e_base = in_n + (long long)(wo - PAD) * CI +
+                               (long long)(ho - PAD) * (Wi * CI);
+#pragma unroll
+      for (int j = 0; j < CH; ++j) {
+        const int km = c * CH + j;
+        const int k0 = km * 32 + q * 8;
+        const int kh = k0 / Sp;
+        const int j0 = k0 % Sp;
+        dst[j] = (bf16x8)0;
+        if (km < nK && k0 < K && j0 < S) {
+          const long long e = e_base + (long long)kh * (Wi * CI) + j0;
+          if (S - j0 >= 8) {
+            dst[j] = *reinterpret_cast<const bf16x8*>(in + e);
+          } else {  // 4 real + 4 zero-pad (conv1's Sp > S)
+            const uint2 v = *reinterpret_cast<const uint2*>(in + e);
+            union { uint4 u; bf16x8 h; } cv;
+            cv.u = make_uint4(v.x, v.y, 0u, 0u);
+            dst[j] = cv.h;
+          }
+        }
+      }
+    };
 
+    // software pipeline: the NEXT chunk's loads are issued before this
+    // chunk's MFMAs, so global-memory latency hides under matrix work
+    // (without this the per-tile load->mfma chain parks waves 89% of
+    // the time: SQ_WAIT_ANY/SQ_WAVE_CYCLES = 0.89 measured)
+    bf16x8 af[CH];
+    load_chunk(0, 0, af);
+
+    for (int tw = 0; tw < tiles_w; ++tw) {
       f32x4 acc[COT];
 #pragma unroll
       for (int ct = 0; ct < COT; ++ct) acc[ct] = (f32x4)0.0f;
 
-      const int wbase = wo - PAD;
-      const long long e_base = in_n + (long long)wbase * CI;
-
-      // unroll capped at 4: a full unroll hoists every A-load and burns
-      // 100+ VGPRs on in-flight fragments (256 VGPR -> 1 wave/SIMD)
-#pragma unroll 4
-      for (int km = 0; km < nK; ++km) {
-        const int k0 = km * 32 + q * 8;
-        const int kh = k0 / Sp;
-        const int j0 = k0 % Sp;
-        bf16x8 afrag = (bf16x8)0;
-        const int hi = ho - PAD + kh;
-        const bool k_valid = (k0 < K) && (j0 < S);
-        const bool h_ok = (PAD == 0) || (hi >= 0 && hi < Hi);
-        if (k_valid && h_ok) {
-          if (PAD == 0 || interior) {
-            const long long e = e_base + (long long)hi * (Wi * CI) + j0;
-            if (S - j0 >= 8) {
-              afrag = *reinterpret_cast<const bf16x8*>(in + e);
-            } else {  // 4 real + 4 zero-pad (conv1's Sp > S)
-              const uint2 v = *reinterpret_cast<const uint2*>(in + e);
-              union { uint4 u; bf16x8 h; } cv;
-              cv.u = make_uint4(v.x, v.y, 0u, 0u);
-              afrag = cv.h;
-            }
-          } else {
-            bf16_t tmp[8];
+#pragma unroll 1
+      for (int c = 0; c < nCH; ++c) {
+        bf16x8 afn[CH];
+        if (c + 1 < nCH)
+          load_chunk(tw, c + 1, afn);
+        else if (tw + 1 < tiles_w)
+          load_chunk(tw + 1, 0, afn);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int jj = j0 + j;
-              const int kw = jj / CI;
-              const int ci = jj - kw * CI;
-              const int wi = wbase + kw;
-              tmp[j] = (jj < S && wi >= 0 && wi < Wi)
-                           ? in[in_n + ((long long)hi * Wi + wi) * CI + ci]
-                           : (bf16_t)0;
-            }
-            afrag = *reinterpret_cast<bf16x8*>(tmp);
+        for (int j = 0; j < CH; ++j) {
+          const int km = c * CH + j;
+          if (km < nK) {
+#pragma unroll
+            for (int ct = 0; ct < COT; ++ct)
+              acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[j], lds_bv[(ct * nK + km) * 64], acc[ct], 0, 0, 0);
           }
         }
 #pragma unroll
-        for (int ct = 0; ct < COT; ++ct)
-          acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, lds_bv[(ct * nK + km) * 64], acc[ct], 0, 0, 0);
+        for (int j = 0; j < CH; ++j) af[j] = afn[j];
       }
 
       // store: lane l covers pixels (l>>4)*4+i at channel l&15
+      const int wo0 = tw << 4;
+      const bool full = (wo0 + 16 <= Wo);
       const int o = lane & 15;
       const long long s_base = out_row + (long long)wo0 * CO + o;
       if (full) {
