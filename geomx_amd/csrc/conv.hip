@@ -56,19 +56,32 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_nhwc(
   constexpr int nK = (K + 31) / 32;
   constexpr int CO = COT * 16;
 
-  // stage B fragments into LDS (shared by the block's 4 waves)
-  __shared__ __attribute__((aligned(16))) short lds_b[COT * nK * 64 * 8];
-  for (int i = threadIdx.x; i < COT * nK * 64; i += blockDim.x) {
-    reinterpret_cast<uint4*>(lds_b)[i] =
-        reinterpret_cast<const uint4*>(w_frags)[i];
+  // small variants (nK < 8, i.e. conv1) keep B fragments in REGISTERS
+  // (<= 32 VGPRs) so the inner loop has no per-MFMA ds_read dependency;
+  // large variants stage B into LDS shared by the block's 4 waves
+  constexpr bool BREG = (nK < 8);
+  __shared__ __attribute__((aligned(16))) short
+      lds_b[BREG ? 8 : COT * nK * 64 * 8];
+  if (!BREG) {
+    for (int i = threadIdx.x; i < COT * nK * 64; i += blockDim.x) {
+      reinterpret_cast<uint4*>(lds_b)[i] =
+          reinterpret_cast<const uint4*>(w_frags)[i];
+    }
+    __syncthreads();
   }
-  __syncthreads();
 
   const int lane = threadIdx.x & 63;
   const int p = lane & 15;        // pixel slot within a tile
   const int q = lane >> 4;        // k-run selector
   // this lane's B read base: ds_read_b128 at [(ct*nK+km)*64+lane]*16B
   const bf16x8* lds_bv = reinterpret_cast<const bf16x8*>(lds_b) + lane;
+
+  bf16x8 breg[BREG ? COT * nK : 1];
+  if (BREG) {
+#pragma unroll
+    for (int i = 0; i < COT * nK; ++i)
+      breg[i] = *reinterpret_cast<const bf16x8*>(w_frags + (i * 64 + lane) * 8);
+  }
 
   float bias_v[COT];
 #pragma unroll
@@ -145,9 +158,14 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_nhwc(
           const int km = c * CH + j;
           if (km < nK) {
 #pragma unroll
-            for (int ct = 0; ct < COT; ++ct)
-              acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  af[j], lds_bv[(ct * nK + km) * 64], acc[ct], 0, 0, 0);
+            for (int ct = 0; ct < COT; ++ct) {
+              if constexpr (BREG)  // nCH==1 here, so km == j statically
+                acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[j], breg[ct * nK + j], acc[ct], 0, 0, 0);
+              else
+                acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[j], lds_bv[(ct * nK + km) * 64], acc[ct], 0, 0, 0);
+            }
           }
         }
 #pragma unroll
@@ -185,10 +203,35 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_nhwc(
   }
 }
 
+// NHWC channel pad: [N,H,W,3] (fp32 or bf16) -> [N,H,W,4] bf16 with a
+// zero 4th channel. One thread per OUTPUT pixel: reads 3 elems, writes
+// one 8-byte bf16x4.
+template <typename TIN>
+__global__ void k_pad_ch3to4_nhwc(const TIN* __restrict__ in,
+                                  bf16_t* __restrict__ out,
+                                  long long npix) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < npix; i += stride) {
+    bf16_t v[4];
+#pragma unroll
+    for (int c = 0; c < 3; ++c) {
+      const TIN x = in[i * 3 + c];
+      if constexpr (sizeof(TIN) == 4) {
+        v[c] = cf2bf((float)x);
+      } else {
+        v[c] = (bf16_t)x;
+      }
+    }
+    v[3] = 0;
+    *reinterpret_cast<uint2*>(out + i * 4) = *reinterpret_cast<uint2*>(v);
+  }
+}
+
 static inline int conv_blocks(long long rows) {
   long long blocks = (rows + 3) / 4;  // 4 waves per block, 1 row per wave
   if (blocks < 1) blocks = 1;
-  if (blocks > 2048) blocks = 2048;
+  if (blocks > 32768) blocks = 32768;  // block turnover = extra latency TLP
   return (int)blocks;
 }
 
@@ -216,6 +259,18 @@ int geops_conv5_nhwc(const bf16_t* in, const bf16_t* w_frags,
   }
 #undef LAUNCH
   return -1;
+}
+
+void geops_pad_ch3to4_nhwc(const void* in, bf16_t* out, long long npix,
+                           int in_is_fp32, hipStream_t s) {
+  long long blocks = (npix + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (in_is_fp32)
+    hipLaunchKernelGGL((k_pad_ch3to4_nhwc<float>), dim3((int)blocks),
+                       dim3(256), 0, s, (const float*)in, out, npix);
+  else
+    hipLaunchKernelGGL((k_pad_ch3to4_nhwc<bf16_t>), dim3((int)blocks),
+                       dim3(256), 0, s, (const bf16_t*)in, out, npix);
 }
 
 }  // extern "C"

@@ -48,6 +48,8 @@ void geops_relu_maxpool2_bwd(const unsigned short*, const uint8_t*,
 int geops_conv5_nhwc(const unsigned short*, const unsigned short*,
                      const float*, unsigned short*, int, int, int, int, int,
                      int, int, int, hipStream_t);
+void geops_pad_ch3to4_nhwc(const void*, unsigned short*, long long, int,
+                           hipStream_t);
 }
 
 namespace {
@@ -228,6 +230,15 @@ void conv5_nhwc(torch::Tensor in, torch::Tensor w_frags, torch::Tensor bias,
               " CO=", CO, " pad=", pad);
 }
 
+void pad_ch3to4_nhwc(torch::Tensor in, torch::Tensor out, int64_t npix) {
+  TORCH_CHECK(in.is_cuda() && out.is_cuda());
+  TORCH_CHECK(out.scalar_type() == torch::kBFloat16);
+  const bool fp32 = in.scalar_type() == torch::kFloat32;
+  TORCH_CHECK(fp32 || in.scalar_type() == torch::kBFloat16);
+  geops_pad_ch3to4_nhwc(in.data_ptr(), (unsigned short*)out.data_ptr(),
+                        npix, fp32 ? 1 : 0, cur_stream());
+}
+
 void sgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
                 double rescale) {
   check_f32(w, "w"); check_f32(g, "g");
@@ -284,6 +295,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_4bit", &quantize_4bit);
   m.def("dequantize_4bit", &dequantize_4bit);
   m.def("conv5_nhwc", &conv5_nhwc);
+  m.def("pad_ch3to4_nhwc", &pad_ch3to4_nhwc);
   m.def("relu_maxpool2_fwd", &relu_maxpool2_fwd);
   m.def("relu_maxpool2_bwd", &relu_maxpool2_bwd);
   m.def("sgd_update", &sgd_update);

@@ -90,13 +90,17 @@ class _Conv5Fn(torch.autograd.Function):
         CO = weight.shape[0]
         CI = (CIr + 3) & ~3
         Ho, Wo = Hi - 4, Wi - 4
-        xb = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
-        xs = xb  # saved for the weight-grad pass (pre-padding)
-        if CI != CIr:  # zero-pad channels (conv1: 3 -> 4)
-            pad = torch.zeros(N, CI - CIr, Hi, Wi, dtype=torch.bfloat16,
-                              device=x.device)
-            xb = torch.cat([xb, pad], dim=1) \
+        if CI != CIr:  # zero-pad channels (conv1: 3 -> 4), fused kernel
+            xc = x.contiguous(memory_format=torch.channels_last)
+            xb = torch.empty(N, CI, Hi, Wi, dtype=torch.bfloat16,
+                             device=x.device,
+                             memory_format=torch.channels_last)
+            _geops.pad_ch3to4_nhwc(xc, xb, N * Hi * Wi)
+            xs = xb[:, :CIr]  # saved for the weight-grad pass
+        else:
+            xb = x.to(torch.bfloat16) \
                 .contiguous(memory_format=torch.channels_last)
+            xs = xb
         wb = weight.detach().to(torch.bfloat16).reshape(-1)
         wz = torch.cat([wb, wb.new_zeros(1)])
         w_frags = wz[fwd_idx].contiguous()
