@@ -19,6 +19,12 @@ from . import native_available
 _SUPPORTED_FWD = {(4, 16), (16, 32), (16, 16), (32, 32)}
 _SUPPORTED_DGRAD = {(32, 16), (16, 16)}
 
+# shapes where the custom kernel MEASURES faster than MIOpen on MI355X
+# (conv1 3->16 @224px: 0.73 vs 1.63 ms). conv2-class shapes currently tie
+# or trail MIOpen's igemm (0.63 vs 0.54 fwd) and stay on the ATen path;
+# flip them on here as the kernel improves.
+DEFAULT_ENABLED = {(4, 16)}
+
 
 def _frag_index(CO: int, CI: int, numel: int, entry) -> torch.Tensor:
     """Build the gather index mapping weight.flatten() -> w_frags layout
@@ -155,10 +161,12 @@ class GeoConv5(torch.nn.Conv2d):
     """nn.Conv2d drop-in (kernel 5, stride 1, pad 0) running on the
     gfx950 MFMA direct-conv kernel when eligible."""
 
-    def __init__(self, in_channels, out_channels, **kw):
+    def __init__(self, in_channels, out_channels, enabled_shapes=None, **kw):
         super().__init__(in_channels, out_channels, kernel_size=5, **kw)
         self._fwd_idx = None
         self._dgrad_idx = None
+        self.enabled_shapes = (DEFAULT_ENABLED if enabled_shapes is None
+                               else enabled_shapes)
 
     def _eligible(self, x) -> bool:
         CI = (self.in_channels + 3) & ~3
@@ -167,6 +175,7 @@ class GeoConv5(torch.nn.Conv2d):
              in _SUPPORTED_DGRAD)
         return (x.is_cuda and native_available()
                 and (CI, self.out_channels) in _SUPPORTED_FWD
+                and (CI, self.out_channels) in self.enabled_shapes
                 and dgrad_ok
                 and self.stride == (1, 1) and self.padding == (0, 0)
                 and self.kernel_size == (5, 5) and self.groups == 1)

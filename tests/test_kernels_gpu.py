@@ -313,9 +313,9 @@ def test_fused_relu_pool_in_model():
 
 @pytest.mark.parametrize("ci,co,hw", [(3, 16, 36), (16, 32, 30), (3, 16, 224)])
 def test_conv5_fwd_matches_aten(ci, co, hw):
-    from geomx_amd.ops.conv import GeoConv5
+    from geomx_amd.ops.conv import GeoConv5, _SUPPORTED_FWD
     torch.manual_seed(20)
-    m = GeoConv5(ci, co).to(DEV)
+    m = GeoConv5(ci, co, enabled_shapes=_SUPPORTED_FWD).to(DEV)
     x = torch.randn(4, ci, hw, hw, device=DEV) \
         .to(memory_format=torch.channels_last)
     y = m(x)  # custom kernel (x requires no grad -> fwd only)
@@ -329,9 +329,9 @@ def test_conv5_fwd_matches_aten(ci, co, hw):
 
 
 def test_conv5_backward_matches_aten():
-    from geomx_amd.ops.conv import GeoConv5
+    from geomx_amd.ops.conv import GeoConv5, _SUPPORTED_FWD
     torch.manual_seed(21)
-    m = GeoConv5(16, 32).to(DEV)
+    m = GeoConv5(16, 32, enabled_shapes=_SUPPORTED_FWD).to(DEV)
     x = torch.randn(2, 16, 40, 40, device=DEV, dtype=torch.bfloat16) \
         .to(memory_format=torch.channels_last).requires_grad_(True)
     y = m(x)
