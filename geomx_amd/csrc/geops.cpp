@@ -40,6 +40,14 @@ void geops_adam_update(float*, const float*, float*, float*, float, float,
                        float, float, float, float, long long, hipStream_t);
 void geops_dcasgd_update(float*, const float*, float*, float*, float, float,
                          float, float, float, long long, bool, hipStream_t);
+void geops_rmsprop_update(float*, const float*, float*, float, float, float,
+                          float, float, long long, hipStream_t);
+void geops_adagrad_update(float*, const float*, float*, float, float, float,
+                          float, long long, hipStream_t);
+void geops_signsgd_update(float*, const float*, float, float, float,
+                          long long, hipStream_t);
+void geops_signum_update(float*, const float*, float*, float, float, float,
+                         float, long long, hipStream_t);
 void geops_relu_maxpool2_fwd(const unsigned short*, unsigned short*, uint8_t*,
                              long long, int, int, int, hipStream_t);
 void geops_relu_maxpool2_bwd(const unsigned short*, const uint8_t*,
@@ -281,6 +289,39 @@ void dcasgd_update(torch::Tensor w, torch::Tensor g, torch::Tensor prev_w,
                       (float)rescale, w.numel(), has_mom, cur_stream());
 }
 
+void rmsprop_update(torch::Tensor w, torch::Tensor g, torch::Tensor n,
+                    double lr, double rho, double eps, double wd,
+                    double rescale) {
+  check_f32(w, "w"); check_f32(g, "g"); check_f32(n, "n");
+  geops_rmsprop_update(w.data_ptr<float>(), g.data_ptr<float>(),
+                       n.data_ptr<float>(), (float)lr, (float)rho,
+                       (float)eps, (float)wd, (float)rescale, w.numel(),
+                       cur_stream());
+}
+
+void adagrad_update(torch::Tensor w, torch::Tensor g, torch::Tensor h,
+                    double lr, double eps, double wd, double rescale) {
+  check_f32(w, "w"); check_f32(g, "g"); check_f32(h, "h");
+  geops_adagrad_update(w.data_ptr<float>(), g.data_ptr<float>(),
+                       h.data_ptr<float>(), (float)lr, (float)eps,
+                       (float)wd, (float)rescale, w.numel(), cur_stream());
+}
+
+void signsgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
+                    double rescale) {
+  check_f32(w, "w"); check_f32(g, "g");
+  geops_signsgd_update(w.data_ptr<float>(), g.data_ptr<float>(), (float)lr,
+                       (float)wd, (float)rescale, w.numel(), cur_stream());
+}
+
+void signum_update(torch::Tensor w, torch::Tensor g, torch::Tensor mom,
+                   double lr, double momentum, double wd, double rescale) {
+  check_f32(w, "w"); check_f32(g, "g"); check_f32(mom, "mom");
+  geops_signum_update(w.data_ptr<float>(), g.data_ptr<float>(),
+                      mom.data_ptr<float>(), (float)lr, (float)momentum,
+                      (float)wd, (float)rescale, w.numel(), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -302,4 +343,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_mom_update", &sgd_mom_update);
   m.def("adam_update", &adam_update);
   m.def("dcasgd_update", &dcasgd_update);
+  m.def("rmsprop_update", &rmsprop_update);
+  m.def("adagrad_update", &adagrad_update);
+  m.def("signsgd_update", &signsgd_update);
+  m.def("signum_update", &signum_update);
 }

@@ -596,6 +596,82 @@ extern "C" __global__ void k_adam_update(float* __restrict__ w,
   }
 }
 
+extern "C" __global__ void k_rmsprop_update(float* __restrict__ w,
+                                            const float* __restrict__ g,
+                                            float* __restrict__ n_st,
+                                            float lr, float rho, float eps,
+                                            float wd, float rescale,
+                                            long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long n4 = n >> 2;
+  float4* w4 = reinterpret_cast<float4*>(w);
+  float4* s4 = reinterpret_cast<float4*>(n_st);
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+#define RMS1(c)                                            \
+  {                                                        \
+    const float gg = gv.c * rescale + wd * wv.c;           \
+    sv.c = rho * sv.c + (1.0f - rho) * gg * gg;            \
+    wv.c -= lr * gg / (sqrtf(sv.c) + eps);                 \
+  }
+  for (long long i = tid; i < n4; i += stride) {
+    float4 wv = w4[i], sv = s4[i];
+    const float4 gv = g4[i];
+    RMS1(x) RMS1(y) RMS1(z) RMS1(w)
+    w4[i] = wv; s4[i] = sv;
+  }
+#undef RMS1
+  for (long long i = (n4 << 2) + tid; i < n; i += stride) {
+    const float gg = g[i] * rescale + wd * w[i];
+    const float sv = rho * n_st[i] + (1.0f - rho) * gg * gg;
+    n_st[i] = sv;
+    w[i] -= lr * gg / (sqrtf(sv) + eps);
+  }
+}
+
+extern "C" __global__ void k_adagrad_update(float* __restrict__ w,
+                                            const float* __restrict__ g,
+                                            float* __restrict__ h,
+                                            float lr, float eps, float wd,
+                                            float rescale, long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < n; i += stride) {
+    const float gg = g[i] * rescale + wd * w[i];
+    const float hv = h[i] + gg * gg;
+    h[i] = hv;
+    w[i] -= lr * gg / (sqrtf(hv) + eps);
+  }
+}
+
+extern "C" __global__ void k_signsgd_update(float* __restrict__ w,
+                                            const float* __restrict__ g,
+                                            float lr, float wd,
+                                            float rescale, long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < n; i += stride) {
+    const float gg = g[i] * rescale + wd * w[i];
+    w[i] -= lr * ((gg > 0.0f) ? 1.0f : ((gg < 0.0f) ? -1.0f : 0.0f));
+  }
+}
+
+extern "C" __global__ void k_signum_update(float* __restrict__ w,
+                                           const float* __restrict__ g,
+                                           float* __restrict__ mom,
+                                           float lr, float momentum,
+                                           float wd, float rescale,
+                                           long long n) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = tid; i < n; i += stride) {
+    const float gg = g[i] * rescale + wd * w[i];
+    const float mv = momentum * mom[i] + (1.0f - momentum) * gg;
+    mom[i] = mv;
+    w[i] -= lr * ((mv > 0.0f) ? 1.0f : ((mv < 0.0f) ? -1.0f : 0.0f));
+  }
+}
+
 extern "C" __global__ void k_dcasgd_update(float* __restrict__ w,
                                            const float* __restrict__ g,
                                            float* __restrict__ prev_w,
@@ -872,6 +948,36 @@ void geops_adam_update(float* w, const float* g, float* m, float* v,
   hipLaunchKernelGGL(k_adam_update, dim3(geops_blocks(n, 4)),
                      dim3(GEOPS_THREADS), 0, s, w, g, m, v, lr_t, beta1,
                      beta2, eps, wd, rescale, n);
+}
+
+void geops_rmsprop_update(float* w, const float* g, float* n_st, float lr,
+                          float rho, float eps, float wd, float rescale,
+                          long long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_rmsprop_update, dim3(geops_blocks(n, 4)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, n_st, lr, rho, eps,
+                     wd, rescale, n);
+}
+
+void geops_adagrad_update(float* w, const float* g, float* h, float lr,
+                          float eps, float wd, float rescale, long long n,
+                          hipStream_t s) {
+  hipLaunchKernelGGL(k_adagrad_update, dim3(geops_blocks(n)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, h, lr, eps, wd,
+                     rescale, n);
+}
+
+void geops_signsgd_update(float* w, const float* g, float lr, float wd,
+                          float rescale, long long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_signsgd_update, dim3(geops_blocks(n)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, lr, wd, rescale, n);
+}
+
+void geops_signum_update(float* w, const float* g, float* mom, float lr,
+                         float momentum, float wd, float rescale,
+                         long long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_signum_update, dim3(geops_blocks(n)),
+                     dim3(GEOPS_THREADS), 0, s, w, g, mom, lr, momentum, wd,
+                     rescale, n);
 }
 
 void geops_dcasgd_update(float* w, const float* g, float* prev_w, float* mom,

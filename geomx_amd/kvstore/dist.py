@@ -80,6 +80,20 @@ class _KeyState:
         self.padded = 0       # padded numel (multiple of P) when sliced
 
 
+class _UpdaterAdapter:
+    """Wraps a python updater fn(key, grad, stored) in the
+    ServerOptimizer interface (the reference's Updater + pickled-
+    controller path, kvstore_server.py:30-100)."""
+
+    def __init__(self, fn):
+        self.fn = fn
+        self.spec = None
+
+    def update(self, key, w, grad, rescale=None):
+        g = grad if rescale in (None, 1.0) else grad * rescale
+        self.fn(key, g, w)
+
+
 class KVStoreDist(KVStoreBase):
     def __init__(self, cfg: Config, topo: Optional[Topology] = None,
                  global_mode: str = "sharded"):
@@ -155,6 +169,29 @@ class KVStoreDist(KVStoreBase):
             raise TypeError("set_optimizer expects OptimizerSpec or dict")
         # every rank creates it; only leaders/owners apply it (state is lazy)
         self.optimizer = ServerOptimizer(spec)
+
+    def set_updater(self, updater) -> None:
+        """Install a custom python updater fn(key, grad, stored) that
+        mutates `stored` in place (python/mxnet/kvstore.py:593-632
+        parity). Replaces the fused-optimizer path for every key."""
+        if not callable(updater):
+            raise TypeError("updater must be callable")
+        self.optimizer = _UpdaterAdapter(updater)
+
+    def _send_command_to_servers(self, head: int, body: str) -> None:
+        """Generic in-band server command (kvstore.py:644-661 parity).
+        SPMD transport: rank 0 broadcasts; leader ranks (the 'servers')
+        receive it via the registered handler."""
+        obj = [int(head), str(body)]
+        if dist.is_initialized() and self.topo.world_size > 1:
+            dist.broadcast_object_list(obj, src=0)
+        if self.topo.is_leader and self._server_command_handler is not None:
+            self._server_command_handler(obj[0], obj[1])
+
+    _server_command_handler = None
+
+    def set_server_command_handler(self, fn) -> None:
+        self._server_command_handler = fn
 
     def set_gradient_compression(self, compression_params: Dict) -> None:
         params = dict(compression_params)

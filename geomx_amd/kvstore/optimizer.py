@@ -22,7 +22,7 @@ from .. import ops
 
 @dataclasses.dataclass
 class OptimizerSpec:
-    name: str = "sgd"            # sgd | sgd_mom | adam | dcasgd
+    name: str = "sgd"   # sgd|sgd_mom|adam|dcasgd|rmsprop|adagrad|signsgd|signum
     lr: float = 0.01
     momentum: float = 0.9
     beta1: float = 0.9
@@ -30,10 +30,12 @@ class OptimizerSpec:
     eps: float = 1e-8
     wd: float = 0.0
     lamda: float = 0.04          # DCASGD delay-compensation scale
+    rho: float = 0.9             # RMSProp decay
     rescale_grad: float = 1.0
 
     def validate(self):
-        if self.name not in ("sgd", "sgd_mom", "adam", "dcasgd"):
+        if self.name not in ("sgd", "sgd_mom", "adam", "dcasgd", "rmsprop",
+                             "adagrad", "signsgd", "signum"):
             raise ValueError(f"unknown optimizer {self.name!r}")
         return self
 
@@ -59,6 +61,10 @@ class ServerOptimizer:
                 st["prev_w"] = w.clone()
                 if self.spec.momentum != 0.0:
                     st["mom"] = torch.zeros_like(w)
+            elif self.spec.name in ("rmsprop", "adagrad"):
+                st["n"] = torch.zeros_like(w)
+            elif self.spec.name == "signum":
+                st["mom"] = torch.zeros_like(w)
             self.state[key] = st
             self.step_count[key] = 0
         return st
@@ -81,6 +87,14 @@ class ServerOptimizer:
         elif s.name == "dcasgd":
             ops.dcasgd_update(w, grad, st["prev_w"], st.get("mom"), s.lr,
                               s.lamda, s.momentum, s.wd, rs)
+        elif s.name == "rmsprop":
+            ops.rmsprop_update(w, grad, st["n"], s.lr, s.rho, s.eps, s.wd, rs)
+        elif s.name == "adagrad":
+            ops.adagrad_update(w, grad, st["n"], s.lr, s.eps, s.wd, rs)
+        elif s.name == "signsgd":
+            ops.signsgd_update(w, grad, s.lr, s.wd, rs)
+        elif s.name == "signum":
+            ops.signum_update(w, grad, st["mom"], s.lr, s.momentum, s.wd, rs)
 
     # -- checkpointing (kvstore.save_optimizer_states layout: a separate
     #    optimizer-state blob, python/mxnet/kvstore.py:566-592) ----------

@@ -233,6 +233,40 @@ def adam_update(w, g, m, v, t, lr, beta1=0.9, beta2=0.999, eps=1e-8, wd=0.0,
     ref.adam_update(w, g, m, v, t, lr, beta1, beta2, eps, wd, rescale)
 
 
+def rmsprop_update(w, g, n, lr, rho=0.9, eps=1e-8, wd=0.0, rescale=1.0):
+    if _on_gpu(w):
+        _require_native().rmsprop_update(w.reshape(-1), g.reshape(-1),
+                                         n.reshape(-1), lr, rho, eps, wd,
+                                         rescale)
+        return
+    ref.rmsprop_update(w, g, n, lr, rho, eps, wd, rescale)
+
+
+def adagrad_update(w, g, h, lr, eps=1e-7, wd=0.0, rescale=1.0):
+    if _on_gpu(w):
+        _require_native().adagrad_update(w.reshape(-1), g.reshape(-1),
+                                         h.reshape(-1), lr, eps, wd, rescale)
+        return
+    ref.adagrad_update(w, g, h, lr, eps, wd, rescale)
+
+
+def signsgd_update(w, g, lr, wd=0.0, rescale=1.0):
+    if _on_gpu(w):
+        _require_native().signsgd_update(w.reshape(-1), g.reshape(-1), lr,
+                                         wd, rescale)
+        return
+    ref.signsgd_update(w, g, lr, wd, rescale)
+
+
+def signum_update(w, g, mom, lr, momentum=0.9, wd=0.0, rescale=1.0):
+    if _on_gpu(w):
+        _require_native().signum_update(w.reshape(-1), g.reshape(-1),
+                                        mom.reshape(-1), lr, momentum, wd,
+                                        rescale)
+        return
+    ref.signum_update(w, g, mom, lr, momentum, wd, rescale)
+
+
 def dcasgd_update(w, g, prev_w, mom, lr, lamda=0.04, momentum=0.0, wd=0.0,
                   rescale=1.0):
     if _on_gpu(w):

@@ -273,6 +273,39 @@ def adam_update(w: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
     w.addcdiv_(m, v.sqrt().add_(eps), value=-lr_t)
 
 
+def rmsprop_update(w: torch.Tensor, g: torch.Tensor, n: torch.Tensor,
+                   lr: float, rho: float = 0.9, eps: float = 1e-8,
+                   wd: float = 0.0, rescale: float = 1.0):
+    """RMSProp (optimizer_op.cc rmsprop_update semantics)."""
+    g = g * rescale + wd * w
+    n.mul_(rho).addcmul_(g, g, value=1 - rho)
+    w.addcdiv_(g, n.sqrt().add_(eps), value=-lr)
+
+
+def adagrad_update(w: torch.Tensor, g: torch.Tensor, h: torch.Tensor,
+                   lr: float, eps: float = 1e-7, wd: float = 0.0,
+                   rescale: float = 1.0):
+    g = g * rescale + wd * w
+    h.addcmul_(g, g, value=1.0)
+    w.addcdiv_(g, h.sqrt().add_(eps), value=-lr)
+
+
+def signsgd_update(w: torch.Tensor, g: torch.Tensor, lr: float,
+                   wd: float = 0.0, rescale: float = 1.0):
+    """SignSGD (optimizer_op.cc signsgd_update)."""
+    g = g * rescale + wd * w
+    w.add_(torch.sign(g), alpha=-lr)
+
+
+def signum_update(w: torch.Tensor, g: torch.Tensor, mom: torch.Tensor,
+                  lr: float, momentum: float = 0.9, wd: float = 0.0,
+                  rescale: float = 1.0):
+    """Signum: momentum then sign (optimizer_op.cc signum_update)."""
+    g = g * rescale + wd * w
+    mom.mul_(momentum).add_(g, alpha=1 - momentum)
+    w.add_(torch.sign(mom), alpha=-lr)
+
+
 def dcasgd_update(w: torch.Tensor, g: torch.Tensor, prev_w: torch.Tensor,
                   mom: Optional[torch.Tensor], lr: float, lamda: float = 0.04,
                   momentum: float = 0.0, wd: float = 0.0, rescale: float = 1.0):

@@ -361,3 +361,29 @@ def test_conv5_cpu_fallback():
     y = m(x)
     ref = torch.nn.functional.conv2d(x, m.weight, m.bias)
     assert torch.allclose(y, ref)
+
+
+@pytest.mark.parametrize("name", ["rmsprop", "adagrad", "signsgd", "signum"])
+def test_extra_optimizers_match_reference(name):
+    torch.manual_seed(30)
+    n = 100001
+    w0 = torch.randn(n)
+    w_g = w0.clone().to(DEV)
+    w_c = w0.clone()
+    st_g = torch.zeros(n, device=DEV)
+    st_c = torch.zeros(n)
+    for _ in range(3):
+        g = torch.randn(n)
+        if name == "rmsprop":
+            ops.rmsprop_update(w_g, g.to(DEV), st_g, 0.01)
+            ref.rmsprop_update(w_c, g, st_c, 0.01)
+        elif name == "adagrad":
+            ops.adagrad_update(w_g, g.to(DEV), st_g, 0.01)
+            ref.adagrad_update(w_c, g, st_c, 0.01)
+        elif name == "signsgd":
+            ops.signsgd_update(w_g, g.to(DEV), 0.01)
+            ref.signsgd_update(w_c, g, 0.01)
+        else:
+            ops.signum_update(w_g, g.to(DEV), st_g, 0.01)
+            ref.signum_update(w_c, g, st_c, 0.01)
+    assert torch.allclose(w_g.cpu(), w_c, atol=1e-5)

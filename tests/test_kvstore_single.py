@@ -130,3 +130,36 @@ def test_row_sparse_pull():
     assert torch.allclose(out[0], w[1])
     assert torch.allclose(out[1], w[4])
     assert torch.allclose(out[2], w[4])
+
+
+def test_server_optimizer_all_names():
+    from geomx_amd.kvstore.optimizer import OptimizerSpec, ServerOptimizer
+    for name in ["sgd", "sgd_mom", "adam", "dcasgd", "rmsprop", "adagrad",
+                 "signsgd", "signum"]:
+        opt = ServerOptimizer(OptimizerSpec(name, lr=0.01))
+        w = torch.randn(32)
+        for _ in range(2):
+            opt.update("k", w, torch.randn(32))
+        assert torch.isfinite(w).all(), name
+
+
+def test_set_updater_and_server_command():
+    kv = make_kv()
+    seen = {}
+
+    def upd(key, grad, stored):
+        stored.sub_(0.5 * grad)
+        seen[key] = True
+
+    kv.set_updater(upd)
+    kv.init("w", torch.ones(4))
+    kv.push("w", torch.ones(4))
+    out = torch.empty(4)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((4,), 0.5))
+    assert seen.get("w")
+
+    cmds = []
+    kv.set_server_command_handler(lambda h, b: cmds.append((h, b)))
+    kv._send_command_to_servers(7, "hello")
+    assert cmds == [(7, "hello")]

@@ -228,3 +228,34 @@ def test_dequantize_2bit_into_out():
     out = torch.empty(n)
     ref.dequantize_2bit(packed, n, 0.7, out=out)
     assert torch.allclose(out + r, g, atol=1e-6)
+
+
+def test_rmsprop_update():
+    w = torch.zeros(8)
+    n = torch.zeros(8)
+    g = torch.ones(8)
+    ref.rmsprop_update(w, g, n, lr=0.1, rho=0.9, eps=1e-8)
+    # n = 0.1, w = -0.1/ (sqrt(0.1)+1e-8)
+    assert torch.allclose(n, torch.full((8,), 0.1))
+    assert torch.allclose(w, torch.full((8,), -0.1 / (0.1 ** 0.5 + 1e-8)),
+                          atol=1e-6)
+
+
+def test_adagrad_update():
+    w = torch.zeros(4)
+    h = torch.zeros(4)
+    g = torch.full((4,), 2.0)
+    ref.adagrad_update(w, g, h, lr=0.1, eps=0.0)
+    assert torch.allclose(h, torch.full((4,), 4.0))
+    assert torch.allclose(w, torch.full((4,), -0.1))
+
+
+def test_signsgd_signum():
+    w = torch.zeros(3)
+    ref.signsgd_update(w, torch.tensor([5.0, -3.0, 0.0]), lr=0.1)
+    assert torch.allclose(w, torch.tensor([-0.1, 0.1, 0.0]))
+    w = torch.zeros(2)
+    mom = torch.zeros(2)
+    ref.signum_update(w, torch.tensor([1.0, -1.0]), mom, lr=0.1, momentum=0.9)
+    assert torch.allclose(mom, torch.tensor([0.1, -0.1]))
+    assert torch.allclose(w, torch.tensor([-0.1, 0.1]))
