@@ -58,6 +58,9 @@ int geops_conv5_nhwc(const unsigned short*, const unsigned short*,
                      int, int, int, hipStream_t);
 void geops_pad_ch3to4_nhwc(const void*, unsigned short*, long long, int,
                            hipStream_t);
+int geops_conv5_wrw_nhwc(const unsigned short*, const unsigned short*,
+                         float*, int, int, int, int, int, int, int, int,
+                         hipStream_t);
 }
 
 namespace {
@@ -238,6 +241,22 @@ void conv5_nhwc(torch::Tensor in, torch::Tensor w_frags, torch::Tensor bias,
               " CO=", CO, " pad=", pad);
 }
 
+void conv5_wrw_nhwc(torch::Tensor in, torch::Tensor gout, torch::Tensor part,
+                    int64_t N, int64_t Hi, int64_t Wi, int64_t Ho,
+                    int64_t Wo, int64_t CI, int64_t CO, int64_t n_wg) {
+  TORCH_CHECK(in.is_cuda() && gout.is_cuda() && part.is_cuda());
+  TORCH_CHECK(in.scalar_type() == torch::kBFloat16 &&
+              gout.scalar_type() == torch::kBFloat16 &&
+              part.scalar_type() == torch::kFloat32);
+  const int rc = geops_conv5_wrw_nhwc(
+      (const unsigned short*)in.data_ptr(),
+      (const unsigned short*)gout.data_ptr(), part.data_ptr<float>(),
+      (int)N, (int)Hi, (int)Wi, (int)Ho, (int)Wo, (int)CI, (int)CO,
+      (int)n_wg, cur_stream());
+  TORCH_CHECK(rc > 0, "conv5_wrw_nhwc: unsupported geometry CI=", CI,
+              " CO=", CO, " Wo=", Wo);
+}
+
 void pad_ch3to4_nhwc(torch::Tensor in, torch::Tensor out, int64_t npix) {
   TORCH_CHECK(in.is_cuda() && out.is_cuda());
   TORCH_CHECK(out.scalar_type() == torch::kBFloat16);
@@ -337,6 +356,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequantize_4bit", &dequantize_4bit);
   m.def("conv5_nhwc", &conv5_nhwc);
   m.def("pad_ch3to4_nhwc", &pad_ch3to4_nhwc);
+  m.def("conv5_wrw_nhwc", &conv5_wrw_nhwc);
   m.def("relu_maxpool2_fwd", &relu_maxpool2_fwd);
   m.def("relu_maxpool2_bwd", &relu_maxpool2_bwd);
   m.def("sgd_update", &sgd_update);

@@ -88,9 +88,47 @@ def build_dgrad_index(weight_shape) -> torch.Tensor:
     return _frag_index(COp, CO, numel, entry)
 
 
+_WRW_SUPPORTED = {(4, 16), (16, 32), (16, 16)}
+_WRW_NWG = 768  # partial slabs (3 workgroups per CU)
+
+
+def build_wrw_unpack_index(weight_shape) -> torch.Tensor:
+    """Gather index from the wrw kernel's [T16][CO] tap-major layout
+    (tap = (kh*5+kw)*CI + ci, CI padded to mult of 4) back to OIHW."""
+    CO, CIr, KH, KW = weight_shape
+    CI = (CIr + 3) & ~3
+    NT = (25 * CI + 15) // 16
+    T16 = NT * 16
+    idx = torch.empty(CO, CIr, KH, KW, dtype=torch.int64)
+    for o in range(CO):
+        for ci in range(CIr):
+            for kh in range(KH):
+                for kw in range(KW):
+                    t = (kh * 5 + kw) * CI + ci
+                    idx[o, ci, kh, kw] = t * CO + o
+    return idx.reshape(-1), T16
+
+
+def wrw_via_kernel(xb_padded: torch.Tensor, go: torch.Tensor,
+                   unpack_idx: torch.Tensor, T16: int, weight_shape):
+    """Run the custom wrw kernel; returns dW in OIHW fp32."""
+    from geomx_amd import _geops
+    N, CI, Hi, Wi = xb_padded.shape
+    CO = go.shape[1]
+    Ho, Wo = go.shape[2], go.shape[3]
+    n_blocks = N * ((Ho + 3) // 4)
+    n_wg = min(_WRW_NWG, n_blocks)
+    part = torch.empty(n_wg, T16, CO, dtype=torch.float32,
+                       device=go.device)
+    _geops.conv5_wrw_nhwc(xb_padded, go, part, N, Hi, Wi, Ho, Wo, CI, CO,
+                          n_wg)
+    full = part.sum(dim=0).reshape(-1)
+    return full[unpack_idx].reshape(weight_shape)
+
+
 class _Conv5Fn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, fwd_idx, dgrad_idx):
+    def forward(ctx, x, weight, bias, fwd_idx, dgrad_idx, wrw_idx, wrw_t16):
         from geomx_amd import _geops
         N, CIr, Hi, Wi = x.shape
         CO = weight.shape[0]
@@ -115,7 +153,12 @@ class _Conv5Fn(torch.autograd.Function):
                           memory_format=torch.channels_last)
         b = bias.detach().float() if bias is not None else torch.Tensor()
         _geops.conv5_nhwc(xb, w_frags, b, out, N, Hi, Wi, Ho, Wo, CI, CO, 0)
-        ctx.save_for_backward(xs, weight, dgrad_idx)
+        use_wrw = (wrw_idx is not None and
+                   (CI, CO) in _WRW_SUPPORTED)
+        # save the PADDED input when the custom wrw runs (it wants CI%4==0);
+        # the unpadded view is recovered as xb[:, :CIr]
+        ctx.save_for_backward(xb if use_wrw else xs, weight, dgrad_idx)
+        ctx.wrw_pack = (wrw_idx, wrw_t16) if use_wrw else None
         ctx.has_bias = bias is not None
         ctx.dims = (N, CIr, CI, CO, Hi, Wi, Ho, Wo)
         return out
@@ -147,14 +190,22 @@ class _Conv5Fn(torch.autograd.Function):
         grad_w = grad_b = None
         if ctx.needs_input_grad[1] or (ctx.has_bias and
                                        ctx.needs_input_grad[2]):
-            gi, gw, gb = torch.ops.aten.convolution_backward(
-                go, x, weight.to(torch.bfloat16),
-                [CO] if ctx.has_bias else None,
-                [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
-                [False, True, ctx.has_bias])
-            grad_w = gw.to(weight.dtype)
-            grad_b = gb.to(weight.dtype) if ctx.has_bias else None
-        return grad_x, grad_w, grad_b, None, None
+            if ctx.wrw_pack is not None:
+                xb_pad = x  # saved padded (see forward)
+                unpack_idx, T16 = ctx.wrw_pack
+                grad_w = wrw_via_kernel(xb_pad, go, unpack_idx, T16,
+                                        weight.shape).to(weight.dtype)
+                if ctx.has_bias:
+                    grad_b = go.float().sum(dim=(0, 2, 3)).to(weight.dtype)
+            else:
+                gi, gw, gb = torch.ops.aten.convolution_backward(
+                    go, x, weight.to(torch.bfloat16),
+                    [CO] if ctx.has_bias else None,
+                    [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+                    [False, True, ctx.has_bias])
+                grad_w = gw.to(weight.dtype)
+                grad_b = gb.to(weight.dtype) if ctx.has_bias else None
+        return grad_x, grad_w, grad_b, None, None, None, None
 
 
 class GeoConv5(torch.nn.Conv2d):
@@ -186,5 +237,8 @@ class GeoConv5(torch.nn.Conv2d):
         if self._fwd_idx is None or self._fwd_idx.device != x.device:
             self._fwd_idx = build_fwd_index(self.weight.shape).to(x.device)
             self._dgrad_idx = build_dgrad_index(self.weight.shape).to(x.device)
+            wi, t16 = build_wrw_unpack_index(self.weight.shape)
+            self._wrw_idx = wi.to(x.device)
+            self._wrw_t16 = t16
         return _Conv5Fn.apply(x, self.weight, self.bias, self._fwd_idx,
-                              self._dgrad_idx)
+                              self._dgrad_idx, self._wrw_idx, self._wrw_t16)

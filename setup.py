@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "geomx_amd/csrc/geops.cpp",
         "geomx_amd/csrc/kernels.hip",
         "geomx_amd/csrc/conv.hip",
+        "geomx_amd/csrc/convwrw.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
