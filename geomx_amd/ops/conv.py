@@ -90,6 +90,11 @@ def build_dgrad_index(weight_shape) -> torch.Tensor:
 
 _WRW_SUPPORTED = {(4, 16), (16, 32), (16, 16)}
 _WRW_NWG = 768  # partial slabs (3 workgroups per CU)
+# The custom wrw kernel is numerically verified (tests) but currently
+# measures SLOWER than MIOpen's igemm_wrw (conv2: 2.07 vs 0.67 ms;
+# conv1: 2.41 vs 1.64 ms) — LDS gather-bound. Off by default until the
+# A-fragment gather is restructured.
+WRW_ENABLED = False
 
 
 def build_wrw_unpack_index(weight_shape) -> torch.Tensor:
@@ -153,7 +158,7 @@ class _Conv5Fn(torch.autograd.Function):
                           memory_format=torch.channels_last)
         b = bias.detach().float() if bias is not None else torch.Tensor()
         _geops.conv5_nhwc(xb, w_frags, b, out, N, Hi, Wi, Ho, Wo, CI, CO, 0)
-        use_wrw = (wrw_idx is not None and
+        use_wrw = (WRW_ENABLED and wrw_idx is not None and
                    (CI, CO) in _WRW_SUPPORTED)
         # save the PADDED input when the custom wrw runs (it wants CI%4==0);
         # the unpadded view is recovered as xb[:, :CIr]
