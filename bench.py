@@ -57,6 +57,8 @@ def parse_args():
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--comm-dtype", type=str, default="fp32", choices=["fp32", "bf16"])
     p.add_argument("--json-out", type=str, default=None)
+    p.add_argument("--backend", type=str, default="auto",
+                   choices=["auto", "nccl", "gloo"])
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout (NHWC avoids MIOpen's "
                         "batched_transpose + slow NCHW pooling kernels)")
@@ -70,7 +72,8 @@ def main():
 
     use_cuda = torch.cuda.is_available()
     device_str = "cuda" if use_cuda else "cpu"
-    backend = "nccl" if use_cuda else "gloo"
+    backend = ("nccl" if use_cuda else "gloo") if args.backend == "auto" \
+        else args.backend
 
     parties = args.parties or (max(2, world // 4) if world > 1 else 1)
     if world % max(1, parties):

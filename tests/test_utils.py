@@ -154,3 +154,30 @@ def test_heartbeat_monitor_single():
     assert hb.dead_nodes(timeout_s=10.0) == [1]
     assert hb.get_num_dead_node(10.0) == 1
     hb.stop()
+
+
+def test_profiler_wrapper(tmp_path):
+    from geomx_amd.utils.profiler import Profiler
+    p = Profiler(filename=str(tmp_path / "trace.json"))
+    p.set_state("run")
+    x = torch.randn(64, 64)
+    (x @ x).sum()
+    p.set_state("stop")
+    out = p.dump(rank=3)
+    assert out and out.endswith("rank3_trace.json")
+    import os
+    assert os.path.exists(out)
+
+
+def test_server_profiler_command_single():
+    from geomx_amd import Config
+    from geomx_amd.kvstore import create
+    from geomx_amd.utils import profiler as prof
+
+    kv = create("dist_sync", cfg=Config.from_env(device="cpu"))
+    prof.send_server_profiler_command(kv, prof.ServerProfilerCommand.STATE,
+                                      "run")
+    prof.send_server_profiler_command(kv, prof.ServerProfilerCommand.PAUSE)
+    out = prof.send_server_profiler_command(
+        kv, prof.ServerProfilerCommand.DUMP)
+    prof._default.set_state("stop")
