@@ -39,18 +39,22 @@ class Profiler:
         if with_stack is not None:
             self.with_stack = with_stack
 
+    _running = False
+
     def set_state(self, state: str):
         if state == "run":
-            if self._prof is None:
+            if not self._running:
                 acts = [torch.profiler.ProfilerActivity.CPU]
                 if torch.cuda.is_available():
                     acts.append(torch.profiler.ProfilerActivity.CUDA)
                 self._prof = torch.profiler.profile(
                     activities=acts, with_stack=self.with_stack)
                 self._prof.__enter__()
+                self._running = True
         elif state == "stop":
-            if self._prof is not None:
+            if self._prof is not None and self._running:
                 self._prof.__exit__(None, None, None)
+                self._running = False
         else:
             raise ValueError(state)
 
@@ -63,6 +67,8 @@ class Profiler:
     def dump(self, rank: Optional[int] = None):
         if self._prof is None:
             return None
+        if self._running:  # chrome export needs a stopped profiler
+            self.set_state("stop")
         if rank is None:
             rank = dist.get_rank() if dist.is_initialized() else 0
         base = os.path.basename(self.filename)
