@@ -58,6 +58,28 @@ class TokenBucket:
             self.total_wait += wait
             _sleep_precise(wait)
 
+    def charge_async(self, nbytes: float) -> float:
+        """Reserve link time WITHOUT blocking; returns the monotonic time
+        at which the emulated transfer completes. Used by the pipelined
+        (one-step-stale) WAN tier: the link 'transfers' while compute
+        proceeds; wait_until() at the apply point sleeps only the
+        remainder."""
+        if not self.enabled or nbytes <= 0:
+            return 0.0
+        now = time.perf_counter()
+        start = max(now, self._debt_until)
+        self._debt_until = start + nbytes * 8.0 / (self.gbps * 1e9)
+        self.total_bytes += nbytes
+        return self._debt_until
+
+    def wait_until(self, ready_time: float):
+        if ready_time <= 0:
+            return
+        now = time.perf_counter()
+        if ready_time > now:
+            self.total_wait += ready_time - now
+            _sleep_precise(ready_time - now)
+
     def stats(self):
         return {"gbps": self.gbps, "total_bytes": self.total_bytes,
                 "total_wait_s": self.total_wait}

@@ -48,7 +48,8 @@ from ..topology import Topology
 
 class _Bucket:
     __slots__ = ("params", "flat", "param_flat", "views", "ready", "expected",
-                 "work", "index", "bsc_u", "bsc_v", "dgt")
+                 "work", "index", "bsc_u", "bsc_v", "dgt",
+                 "wan_buf", "wan_work", "wan_ready")
 
     def __init__(self, index: int):
         self.index = index
@@ -62,6 +63,10 @@ class _Bucket:
         self.bsc_u: Optional[torch.Tensor] = None
         self.bsc_v: Optional[torch.Tensor] = None
         self.dgt = None
+        # pipelined (dist_async) WAN tier state
+        self.wan_buf: Optional[torch.Tensor] = None
+        self.wan_work = None
+        self.wan_ready = 0.0
 
 
 class _Bf16Work:
@@ -197,6 +202,11 @@ class GeoTrainer:
             if b.work is not None:
                 b.work.wait()
                 b.work = None
+        if self.mode == "hips" and self.topo.num_parties > 1 \
+                and self.cfg.mode == "dist_async" \
+                and self.cfg.compression in (None, "fp16"):
+            self._wan_tier_async()
+            return
         if self.mode == "hips" and self.topo.num_parties > 1:
             self._wan_tier()
         elif self.mode == "flat" and self.wan.enabled \
@@ -210,6 +220,58 @@ class GeoTrainer:
                     "all_reduce", b.flat.numel() * 4, self.cfg.num_parties))
         for b in self.buckets:
             self.server_opt.update(("bucket", b.index), b.param_flat, b.flat)
+            b.ready = 0
+
+    # -- pipelined WAN tier (dist_async / MixedSync realized) -----------
+    def _wan_tier_async(self):
+        """One-step-stale global tier: this step's party-averaged
+        gradients start their WAN exchange NOW (async; the emulated link
+        'transfers' while the next step computes), and the optimizer
+        consumes the COMPLETED global gradient from the previous step.
+        This is the wall-clock form of the reference's MixedSync
+        (async global server, DataHandleAsyncDefault
+        kvstore_dist_server.h:1519-1611): WAN latency hides entirely
+        under compute instead of serializing every step."""
+        topo = self.topo
+        P = topo.num_parties
+        ctype = self.cfg.compression
+        for b in self.buckets:
+            # 1) apply the PENDING (stale) global gradient, if any
+            if b.wan_work is not None or b.wan_ready:
+                if b.wan_work is not None:
+                    b.wan_work.wait()
+                    b.wan_work = None
+                self.wan.wait_until(b.wan_ready)
+                apply_buf = b.wan_buf
+                if ctype == "fp16":
+                    apply_buf = b.wan_buf.float()
+                if topo.num_workers > 1:
+                    # LAN fan-out of the arrived global gradient
+                    dist.broadcast(apply_buf, src=topo.leader_rank,
+                                   group=topo.party_group)
+                self.server_opt.update(("bucket", b.index), b.param_flat,
+                                       apply_buf)
+            # 2) launch this step's WAN exchange from a snapshot
+            if topo.is_leader:
+                if ctype == "fp16":
+                    snap = b.flat.to(torch.float16)
+                    nbytes = snap.numel() * 2
+                else:
+                    snap = b.flat.clone()
+                    nbytes = snap.numel() * 4
+                b.wan_buf = snap
+                b.wan_work = dist.all_reduce(snap, group=topo.leader_group,
+                                             async_op=True)
+                b.wan_ready = self.wan.charge_async(
+                    cross_party_bytes("all_reduce", nbytes, P))
+            else:
+                if b.wan_buf is None or b.wan_buf.numel() != b.flat.numel():
+                    b.wan_buf = torch.empty_like(b.flat)
+                if ctype == "fp16":
+                    b.wan_buf = torch.empty(b.flat.numel(),
+                                            dtype=torch.float16,
+                                            device=b.flat.device)
+                b.wan_ready = 1e-9  # marks a pending apply next step
             b.ready = 0
 
     # -- WAN tier --------------------------------------------------------
