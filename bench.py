@@ -59,6 +59,9 @@ def parse_args():
     p.add_argument("--json-out", type=str, default=None)
     p.add_argument("--backend", type=str, default="auto",
                    choices=["auto", "nccl", "gloo"])
+    p.add_argument("--sync-mode", type=str, default="dist_sync",
+                   choices=["dist_sync", "dist_async"],
+                   help="dist_async = pipelined one-step-stale WAN tier")
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout (NHWC avoids MIOpen's "
                         "batched_transpose + slow NCHW pooling kernels)")
@@ -84,7 +87,7 @@ def main():
     topo_parties = parties if args.mode == "hips" and world > 1 else 1
 
     cfg = Config.from_env(
-        num_parties=parties, backend=backend,
+        num_parties=parties, backend=backend, mode=args.sync_mode,
         compression=args.compress, bsc_ratio=args.bsc_ratio,
         wan_gbps=args.wan_gbps, bucket_mb=args.bucket_mb,
         comm_dtype=args.comm_dtype)
@@ -177,6 +180,7 @@ def main():
                                 f"{world // max(1, topo.num_parties)}"
                                 if args.mode == "hips" else f"dp{world}"),
                 "compression": args.compress,
+                "sync_mode": args.sync_mode,
                 "wan_gbps": args.wan_gbps,
                 "optimizer": args.optimizer,
             },

@@ -63,6 +63,10 @@ def main():
                       "--bsc-ratio", str(args.bsc_ratio)]),
         ("hips mpq", ["--mode", "hips", "--compress", "mpq",
                       "--bsc-ratio", str(args.bsc_ratio)]),
+        ("hips async (stale-1)", ["--mode", "hips", "--sync-mode",
+                                  "dist_async"]),
+        ("hips async+fp16", ["--mode", "hips", "--sync-mode", "dist_async",
+                             "--compress", "fp16"]),
     ]
     rows = []
     port = args.port
