@@ -39,6 +39,10 @@ def base_parser():
     p.add_argument("--data-n", type=int, default=1024)
     p.add_argument("--max-iters", type=int, default=0)
     p.add_argument("--measure-out", type=str, default=None)
+    p.add_argument("--global-mode", type=str, default="sharded",
+                   choices=["sharded", "replicated"],
+                   help="global tier: sharded = MultiGPS key owners; "
+                        "replicated = TSEngine-style incast-free ring")
     return p
 
 
@@ -46,7 +50,8 @@ def setup(args, mode="dist_sync", **cfg_overrides):
     parties = args.parties if topo_world() > 1 else 1
     cfg = Config.from_env(num_parties=parties, **cfg_overrides)
     topo = init_topology(parties)
-    kv = create(mode, cfg=cfg, topo=topo)
+    kv = create(mode, cfg=cfg, topo=topo,
+                global_mode=getattr(args, "global_mode", "sharded"))
     device = topo.device
     torch.manual_seed(0)
     net = geo_cnn(in_channels=3, image_size=args.image_size).to(device)
