@@ -134,14 +134,22 @@ class GeoTrainer:
         if cur.params:
             self.buckets.append(cur)
         self.param_bucket: Dict[torch.nn.Parameter, tuple] = {}
+        ALIGN = 64  # elements: every param/grad view starts 256B-aligned
+                    # (unaligned weight base addresses push MIOpen onto
+                    # its unaligned igemm variants: conv2 wrw 1.62 ms vs
+                    # 0.69 ms for the same shape aligned)
+
+        def aligned(n):
+            return (n + ALIGN - 1) // ALIGN * ALIGN
+
         for b in self.buckets:
-            total = sum(p.numel() for p in b.params)
+            total = sum(aligned(p.numel()) for p in b.params)
             b.flat = torch.zeros(total, dtype=torch.float32, device=self.device)
             b.expected = len(b.params)
             # flatten parameters too: p.data becomes a view of one fp32
             # buffer per bucket, so the fused optimizer kernel updates
             # every parameter of the bucket in ONE launch, zero copies
-            b.param_flat = torch.empty(total, dtype=torch.float32,
+            b.param_flat = torch.zeros(total, dtype=torch.float32,
                                        device=self.device)
             off = 0
             for p in b.params:
@@ -155,7 +163,7 @@ class GeoTrainer:
                     pview.copy_(p.data.float())
                 p.data = pview
                 self.param_bucket[p] = (b, len(b.views) - 1)
-                off += n
+                off += aligned(n)
 
     def _register_hooks(self):
         self._hook_handles = []
@@ -336,6 +344,7 @@ class GeoTrainer:
     def refresh_params(self):
         """Re-attach parameters to the flat buffers after an external
         load (e.g. load_parameters replaced p.data)."""
+        ALIGN = 64
         for b in self.buckets:
             off = 0
             for p in b.params:
@@ -345,7 +354,7 @@ class GeoTrainer:
                     with torch.no_grad():
                         pv.copy_(p.data.float())
                     p.data = pv
-                off += n
+                off += (n + ALIGN - 1) // ALIGN * ALIGN
 
     # checkpoint parity with the kvstore API
     def save_optimizer_states(self, fname: str, dump_optimizer: bool = False):
