@@ -8,7 +8,7 @@ load_optimizer_states/_barrier) and include/mxnet/kvstore.h:59-352.
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 

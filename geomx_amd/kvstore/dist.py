@@ -41,7 +41,7 @@ Synchronization modes:
 
 from __future__ import annotations
 
-import math
+
 import pickle
 from typing import Dict, List, Optional
 
@@ -249,9 +249,7 @@ class KVStoreDist(KVStoreBase):
 
         if topo.num_parties == 1:
             if topo.is_leader:
-                agg = party_sum
-                self._apply_global(key, st, [agg])
-            self._leader_cache_sync(st)
+                self._apply_global(key, st, [party_sum])
             return
 
         if self.cfg.use_hfa:
@@ -514,10 +512,6 @@ class KVStoreDist(KVStoreBase):
             return
         dist.broadcast(st.stored, src=owner_leader, group=group)
         self.wan.charge(cross_party_bytes("broadcast", st.numel * 4, P))
-
-    def _leader_cache_sync(self, st: _KeyState):
-        # placeholder for party-internal consistency; pull() broadcasts.
-        pass
 
     def row_sparse_pull(self, key, out: torch.Tensor, row_ids: torch.Tensor,
                         priority: int = 0) -> None:

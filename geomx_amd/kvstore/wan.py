@@ -23,7 +23,7 @@ full payload — which is exactly why HiPS + compression wins under a cap.
 from __future__ import annotations
 
 import time
-from typing import Optional
+
 
 import torch
 

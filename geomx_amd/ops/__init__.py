@@ -11,7 +11,7 @@ Policy (matches the project contract):
 
 from __future__ import annotations
 
-import os
+
 from typing import Optional, Tuple
 
 import torch

@@ -2,14 +2,16 @@
 kernel (csrc/conv.hip) for the flagship CNN's shapes.
 
 Forward and the data gradient run on the custom implicit-GEMM kernel
-(NHWC bf16, weights pre-packed into per-lane MFMA B-fragment order);
-the weight/bias gradients go through ATen's convolution_backward.
-Anything outside the supported geometry falls back to F.conv2d.
+(NHWC bf16, weights pre-packed into per-lane MFMA B-fragment order).
+The weight/bias gradients go through ATen's convolution_backward by
+default; a custom wrw kernel (csrc/convwrw.hip) exists and is verified
+but disabled (see WRW_ENABLED). Anything outside the supported geometry
+falls back to F.conv2d.
 """
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+
 
 import torch
 import torch.nn.functional as F

@@ -33,7 +33,7 @@ Design notes (MI355X-first):
 
 from __future__ import annotations
 
-import math
+
 from typing import Dict, List, Optional
 
 import torch
