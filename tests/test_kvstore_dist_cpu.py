@@ -309,3 +309,23 @@ def _sliced_big_tensor(rank, world):
 
 def test_sliced_big_tensor_ws4():
     run_dist(4, _sliced_big_tensor)
+
+
+# ---------------------------------------------------------------------------
+# uneven party sizes (explicit GEOMX_PARTY_SIZES)
+# ---------------------------------------------------------------------------
+
+def _uneven_parties(rank, world):
+    kv = _mk(num_parties=2, party_sizes=[1, 3])
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    kv.init("w", torch.ones(6))
+    kv.push("w", torch.full((6,), 1.0))  # sum over all 4 workers = 4
+    out = torch.empty(6)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((6,), 0.6), atol=1e-6), (rank, out)
+    # rank 0 is a whole party of 1; ranks 1..3 are party 1
+    assert kv.num_workers == (1 if rank == 0 else 3)
+
+
+def test_uneven_parties_ws4():
+    run_dist(4, _uneven_parties)
