@@ -136,10 +136,12 @@ __global__ __launch_bounds__(WRW_THREADS) void k_conv5_wrw_nhwc(
     for (int r = 0; r < nrows; ++r) {
 
       // ---- accumulate this row into the wave-owned pairs
-#pragma unroll 2
+      // FULL unroll: acc[j] must be statically indexed or it spills to
+      // scratch (224 B/lane measured with a rolled loop — rule 20)
+#pragma unroll
       for (int j = 0; j < MAXP; ++j) {
         const int pair = wid + 4 * j;
-        if (pair >= NPAIR) break;
+        if (pair >= NPAIR) continue;
         const int tt = pair / COT;   // tap tile
         const int ot = pair - tt * COT;
         // this lane's tap (A row) within the tile
@@ -152,17 +154,55 @@ __global__ __launch_bounds__(WRW_THREADS) void k_conv5_wrw_nhwc(
         const bf16_t* arow = lds_in + ((r + kh) * CI + ci) * INPP + kw;
         const bf16_t* brow =
             lds_gout + ((r * CO) + (ot * 16) + (lane & 15)) * PIXP;
+        // A-fragment base WITHOUT the kw shift: for CI >= 16 a tap tile
+        // is a single (kh,kw) and kw is wave-uniform, so the 8-pixel
+        // window [kw, kw+8) can be cut out of 12 ALIGNED elements with
+        // v_alignbyte instead of 8 scalar ds_read_u16 gathers.
+        const bf16_t* arow0 = arow - kw;
         for (int w = 0; w < W; ++w) {
           const int pix0 = w * 32 + q * 8;
-          bf16_t av[8];
+          bf16x8 afrag = (bf16x8)0;
           if (valid) {
+            if (CI >= 16) {
+              // aligned: row base is 16B-aligned (INPP*2 % 16 == 0) and
+              // pix0*2 % 16 == 0
+              const uint4 lo =
+                  *reinterpret_cast<const uint4*>(arow0 + pix0);
+              const uint2 hi =
+                  *reinterpret_cast<const uint2*>(arow0 + pix0 + 8);
+              // constant indices only (a runtime-indexed local array
+              // would spill to scratch — guide rule 20); kw is a
+              // wave-uniform scalar so the switch is one cheap branch
+              const unsigned d0 = lo.x, d1 = lo.y, d2 = lo.z, d3 = lo.w;
+              const unsigned d4 = hi.x, d5 = hi.y;
+              unsigned o0, o1, o2, o3;
+              switch (kw) {
+                case 0: o0 = d0; o1 = d1; o2 = d2; o3 = d3; break;
+                case 1:
+                  o0 = __builtin_amdgcn_alignbyte(d1, d0, 2);
+                  o1 = __builtin_amdgcn_alignbyte(d2, d1, 2);
+                  o2 = __builtin_amdgcn_alignbyte(d3, d2, 2);
+                  o3 = __builtin_amdgcn_alignbyte(d4, d3, 2);
+                  break;
+                case 2: o0 = d1; o1 = d2; o2 = d3; o3 = d4; break;
+                case 3:
+                  o0 = __builtin_amdgcn_alignbyte(d2, d1, 2);
+                  o1 = __builtin_amdgcn_alignbyte(d3, d2, 2);
+                  o2 = __builtin_amdgcn_alignbyte(d4, d3, 2);
+                  o3 = __builtin_amdgcn_alignbyte(d5, d4, 2);
+                  break;
+                default: o0 = d2; o1 = d3; o2 = d4; o3 = d5; break;
+              }
+              union { uint4 u; bf16x8 h; } cv;
+              cv.u = make_uint4(o0, o1, o2, o3);
+              afrag = cv.h;
+            } else {
+              bf16_t av[8];
 #pragma unroll
-            for (int jj = 0; jj < 8; ++jj) av[jj] = arow[pix0 + jj];
-          } else {
-#pragma unroll
-            for (int jj = 0; jj < 8; ++jj) av[jj] = (bf16_t)0;
+              for (int jj = 0; jj < 8; ++jj) av[jj] = arow[pix0 + jj];
+              afrag = *reinterpret_cast<bf16x8*>(av);
+            }
           }
-          const bf16x8 afrag = *reinterpret_cast<bf16x8*>(av);
           // B: lane l -> col o = l&15, pixels pix0..pix0+7
           const bf16x8 bfrag =
               *reinterpret_cast<const bf16x8*>(brow + pix0);
@@ -176,10 +216,10 @@ __global__ __launch_bounds__(WRW_THREADS) void k_conv5_wrw_nhwc(
 
   // ---- flush: part[wg][tap][o] (unique per wg -> plain stores)
   float* base = part + (long long)wg * T16 * CO;
-#pragma unroll 1
+#pragma unroll
   for (int j = 0; j < MAXP; ++j) {
     const int pair = wid + 4 * j;
-    if (pair >= NPAIR) break;
+    if (pair >= NPAIR) continue;
     const int tt = pair / COT;
     const int ot = pair - tt * COT;
     const int o = ot * 16 + (lane & 15);
