@@ -215,11 +215,24 @@ class _Conv5Fn(torch.autograd.Function):
         return grad_x, grad_w, grad_b, None, None, None, None
 
 
+_wrw_idx_cache = {}
+
+
+def _wrw_cached_index(weight_shape, device):
+    key = (tuple(weight_shape), str(device))
+    ent = _wrw_idx_cache.get(key)
+    if ent is None:
+        idx, t16 = build_wrw_unpack_index(weight_shape)
+        ent = (idx.to(device), t16)
+        _wrw_idx_cache[key] = ent
+    return ent
+
+
 class _AtenSplitConvFn(torch.autograd.Function):
-    """ATen forward + SPLIT backward: dgrad and wrw+bias as separate
-    convolution_backward calls. Probing showed MIOpen picks a 2.3x
-    slower wrw igemm when the masks are combined in the training
-    context; splitting lets each pass get its own best algo."""
+    """ATen forward + SPLIT backward: dgrad via ATen, weight grad via
+    the custom gfx950 wrw kernel where supported (1.10 vs the 1.62 ms
+    igemm MIOpen picks in the training context), bias via a channel
+    sum. Falls back to a separate ATen wrw call otherwise."""
 
     @staticmethod
     def forward(ctx, x, weight, bias):
@@ -244,11 +257,22 @@ class _AtenSplitConvFn(torch.autograd.Function):
                 [True, False, False])
             grad_x = gx
         if ctx.needs_input_grad[1]:
-            _, gw, gb = torch.ops.aten.convolution_backward(
-                go, x, w, [CO] if ctx.has_bias else None, [1, 1], [0, 0],
-                [1, 1], False, [0, 0], 1, [False, True, ctx.has_bias])
-            grad_w = gw.float()
-            grad_b = gb.float() if ctx.has_bias else None
+            CIr = w.shape[1]
+            if native_available() and (CIr, CO) in _WRW_SUPPORTED \
+                    and CIr % 4 == 0:
+                idx, t16 = _wrw_cached_index(w.shape, go.device)
+                grad_w = wrw_via_kernel(
+                    x.contiguous(memory_format=torch.channels_last), go,
+                    idx, t16, w.shape)
+                grad_b = go.float().sum(dim=(0, 2, 3)) if ctx.has_bias \
+                    else None
+            else:
+                _, gw, gb = torch.ops.aten.convolution_backward(
+                    go, x, w, [CO] if ctx.has_bias else None, [1, 1],
+                    [0, 0], [1, 1], False, [0, 0], 1,
+                    [False, True, ctx.has_bias])
+                grad_w = gw.float()
+                grad_b = gb.float() if ctx.has_bias else None
         return grad_x, grad_w, grad_b
 
 
@@ -275,8 +299,10 @@ class GeoConv5(torch.nn.Conv2d):
                 and self.stride == (1, 1) and self.padding == (0, 0)
                 and self.kernel_size == (5, 5) and self.groups == 1)
 
-    # flip to try the split-backward ATen path on non-custom shapes
-    SPLIT_BACKWARD = False
+    # split-backward ATen path on non-custom shapes: dgrad by ATen,
+    # weight grad by the custom wrw kernel (conv2: 1.10 ms vs the
+    # 1.62 ms igemm MIOpen selects inside the training step)
+    SPLIT_BACKWARD = True
 
     def forward(self, x):
         if not self._eligible(x):

@@ -438,3 +438,31 @@ def test_conv1_full_training_path_wrw():
                           rtol=0.05), \
         (m.weight.grad - m2.weight.grad).abs().max()
     assert torch.allclose(m.bias.grad, m2.bias.grad, atol=0.5, rtol=0.05)
+
+
+def test_split_backward_conv2_grads_match():
+    """conv2 path with SPLIT_BACKWARD (ATen fwd + custom wrw): all three
+    grads vs plain autograd."""
+    from geomx_amd.ops.conv import GeoConv5
+    torch.manual_seed(50)
+    m = GeoConv5(16, 32).to(DEV)  # not in DEFAULT_ENABLED -> split path
+    assert m.SPLIT_BACKWARD
+    x = torch.randn(4, 16, 40, 40, device=DEV, dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    y = m(x)
+    g = torch.randn_like(y)
+    y.backward(g)
+    m2 = torch.nn.Conv2d(16, 32, 5).to(DEV)
+    with torch.no_grad():
+        m2.weight.copy_(m.weight); m2.bias.copy_(m.bias)
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = torch.nn.functional.conv2d(x2, m2.weight.to(torch.bfloat16),
+                                    m2.bias.to(torch.bfloat16))
+    y2.backward(g)
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=0.1,
+                          rtol=0.05)
+    assert torch.allclose(m.weight.grad, m2.weight.grad.float(), atol=1.0,
+                          rtol=0.05), \
+        (m.weight.grad - m2.weight.grad.float()).abs().max()
+    assert torch.allclose(m.bias.grad, m2.bias.grad.float(), atol=0.5,
+                          rtol=0.05)
