@@ -264,8 +264,8 @@ class _AtenSplitConvFn(torch.autograd.Function):
                 grad_w = wrw_via_kernel(
                     x.contiguous(memory_format=torch.channels_last), go,
                     idx, t16, w.shape)
-                grad_b = go.float().sum(dim=(0, 2, 3)) if ctx.has_bias \
-                    else None
+                grad_b = go.sum(dim=(0, 2, 3), dtype=torch.float32) \
+                    if ctx.has_bias else None
             else:
                 _, gw, gb = torch.ops.aten.convolution_backward(
                     go, x, w, [CO] if ctx.has_bias else None, [1, 1],
