@@ -300,9 +300,11 @@ class GeoConv5(torch.nn.Conv2d):
                 and self.kernel_size == (5, 5) and self.groups == 1)
 
     # split-backward ATen path on non-custom shapes: dgrad by ATen,
-    # weight grad by the custom wrw kernel (conv2: 1.10 ms vs the
-    # 1.62 ms igemm MIOpen selects inside the training step)
-    SPLIT_BACKWARD = True
+    # weight grad by the custom wrw kernel. Off by default: the custom
+    # wrw wins standalone (1.10 vs 1.62 ms) but the split costs more
+    # elsewhere in the full step (bench 69.9k vs 74.4k samples/s) —
+    # see ROADMAP item 1.
+    SPLIT_BACKWARD = False
 
     def forward(self, x):
         if not self._eligible(x):
