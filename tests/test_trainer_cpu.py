@@ -235,3 +235,67 @@ def _trainer_async_wan_overlap(rank, world):
 
 def test_async_wan_overlap_ws2():
     run_dist(2, _trainer_async_wan_overlap)
+
+
+def _trainer_hips_fp16(rank, world):
+    cfg = Config.from_env(num_parties=2, backend="gloo", device="cpu",
+                          bucket_mb=1, compression="fp16")
+    topo = init_topology(2, None, "gloo", "cpu")
+    model = _tiny_model()
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec(name="sgd", lr=0.05),
+                    mode="hips")
+    for s in range(2):
+        x, y = _make_data(seed=30 + s)
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+    import torch.distributed as dist
+    for p in model.parameters():
+        ref = p.data.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(p.data, ref, atol=1e-6)
+
+
+def test_hips_fp16_consistent_ws4():
+    run_dist(4, _trainer_hips_fp16)
+
+
+def _trainer_dist_checkpoint(rank, world, tmpdir):
+    import os
+    from geomx_amd.utils import checkpoint as ckpt
+    cfg = Config.from_env(num_parties=2, backend="gloo", device="cpu",
+                          bucket_mb=1)
+    topo = init_topology(2, None, "gloo", "cpu")
+    model = _tiny_model()
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec("adam", lr=0.01),
+                    mode="hips")
+    for s in range(2):
+        x, y = _make_data(seed=60 + s)
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        tr.zero_grad(); loss.backward(); tr.step()
+    prefix = os.path.join(tmpdir, f"ck")
+    if rank == 0:
+        ckpt.save_checkpoint(model, tr, prefix, 7)
+    import torch.distributed as dist
+    dist.barrier()
+    # every rank resumes from rank 0's checkpoint
+    model2 = _tiny_model()
+    tr2 = GeoTrainer(model2, cfg, topo, OptimizerSpec("adam", lr=0.01),
+                     mode="hips")
+    ckpt.load_checkpoint(model2, tr2, prefix, 7)
+    for a, b in zip(model.parameters(), model2.parameters()):
+        assert torch.allclose(a.data, b.data, atol=1e-7)
+    # training continues consistently across ranks
+    x, y = _make_data(seed=99)
+    loss = torch.nn.functional.cross_entropy(model2(x), y)
+    tr2.zero_grad(); loss.backward(); tr2.step()
+    for p in model2.parameters():
+        ref = p.data.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(p.data, ref, atol=1e-6)
+
+
+def test_dist_checkpoint_ws4(tmp_path_factory):
+    d = str(tmp_path_factory.mktemp("ck"))
+    run_dist(4, _trainer_dist_checkpoint, d)
