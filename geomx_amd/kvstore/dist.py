@@ -513,6 +513,25 @@ class KVStoreDist(KVStoreBase):
         dist.broadcast(st.stored, src=owner_leader, group=group)
         self.wan.charge(cross_party_bytes("broadcast", st.numel * 4, P))
 
+    def push_row_sparse(self, key, row_ids: torch.Tensor,
+                        values: torch.Tensor, priority: int = 0) -> None:
+        """Push a row-sparse gradient (rows named by row_ids, one value
+        row each — the reference's row_sparse push storage,
+        kvstore_dist.h:900 EncodeRowSparseKey). Duplicate ids
+        accumulate. Densified before the wire: in-node xGMI bandwidth
+        makes dense collectives the faster transport for these sizes."""
+        st = self._state(key)
+        if len(st.shape) < 2:
+            raise ValueError("push_row_sparse needs a >=2d key")
+        rows = st.shape[0]
+        width = st.numel // rows
+        dense = torch.zeros(rows, width, dtype=torch.float32,
+                            device=self._device)
+        dense.index_add_(0, row_ids.to(self._device).long(),
+                         values.detach().reshape(-1, width).float()
+                         .to(self._device))
+        self.push(key, dense, priority)
+
     def row_sparse_pull(self, key, out: torch.Tensor, row_ids: torch.Tensor,
                         priority: int = 0) -> None:
         """Pull only the rows named by row_ids (python/mxnet/kvstore.py:316;

@@ -163,3 +163,16 @@ def test_set_updater_and_server_command():
     kv.set_server_command_handler(lambda h, b: cmds.append((h, b)))
     kv._send_command_to_servers(7, "hello")
     assert cmds == [(7, "hello")]
+
+
+def test_push_row_sparse():
+    kv = make_kv()
+    kv.init("emb", torch.zeros(6, 3))
+    ids = torch.tensor([1, 4, 1])
+    vals = torch.ones(3, 3)
+    kv.push_row_sparse("emb", ids, vals)
+    out = torch.empty(6, 3)
+    kv.pull("emb", out)
+    assert torch.allclose(out[1], torch.full((3,), 2.0))  # dup id accumulated
+    assert torch.allclose(out[4], torch.ones(3))
+    assert out[0].abs().sum() == 0
