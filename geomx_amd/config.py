@@ -60,6 +60,11 @@ class Config:
     # "dist_sync"  -> FSA: both tiers synchronous (kvstore.cc:55-62)
     # "dist_async" -> MixedSync: async global tier (DataHandleAsyncDefault)
     mode: str = "dist_sync"
+    # dist_async transport: "lockstep" reproduces the async UPDATE MATH
+    # over synchronous collectives; "store" is the true-async
+    # parameter server (kvstore/async_ps.py) where parties proceed at
+    # their own pace
+    async_transport: str = "lockstep"
     # HFA (hierarchical frequency aggregation, examples/cnn_hfa.py):
     # workers run K1 local steps between pushes; leaders forward to the
     # global tier only every K2-th aggregation (kvstore_dist_server.h:1324-1343)
@@ -153,4 +158,6 @@ class Config:
             raise ValueError("num_parties >= 1")
         if self.hfa_k1 < 1 or self.hfa_k2 < 1:
             raise ValueError("hfa_k1/k2 >= 1")
+        if self.async_transport not in ("lockstep", "store"):
+            raise ValueError("async_transport must be lockstep|store")
         return self

@@ -110,6 +110,7 @@ class KVStoreDist(KVStoreBase):
         self.compression: Optional[Dict] = None
         self.wan = TokenBucket(cfg.wan_gbps)
         self._device = self.topo.device
+        self._aps = None  # true-async global tier (async_transport=store)
 
     # ------------------------------------------------------------------
     # properties (GeoMX API parity: kvstore.py:501-565)
@@ -169,6 +170,8 @@ class KVStoreDist(KVStoreBase):
             raise TypeError("set_optimizer expects OptimizerSpec or dict")
         # every rank creates it; only leaders/owners apply it (state is lazy)
         self.optimizer = ServerOptimizer(spec)
+        if self._aps is not None:
+            self._aps.optimizer = ServerOptimizer(spec)
 
     def set_updater(self, updater) -> None:
         """Install a custom python updater fn(key, grad, stored) that
@@ -205,6 +208,27 @@ class KVStoreDist(KVStoreBase):
             params.setdefault("size_lower_bound", self.cfg.size_lower_bound)
         self.compression = params
 
+    # -- true-async global tier (async_transport="store") ---------------
+    def _use_aps(self) -> bool:
+        return (self.cfg.mode == "dist_async"
+                and self.cfg.async_transport == "store"
+                and self.topo.num_parties > 1
+                and dist.is_initialized())
+
+    def _ensure_aps(self):
+        if self._aps is None:
+            from .async_ps import AsyncPSGlobal
+            store = dist.distributed_c10d._get_default_store()
+            self._aps = AsyncPSGlobal(store, self.topo, self._device,
+                                      wan=self.wan)
+            if self.optimizer is not None:
+                self._aps.optimizer = ServerOptimizer(self.optimizer.spec)
+        return self._aps
+
+    def close(self):
+        if self._aps is not None:
+            self._aps.stop()
+
     # ------------------------------------------------------------------
     # init
     # ------------------------------------------------------------------
@@ -228,6 +252,10 @@ class KVStoreDist(KVStoreBase):
         if dist.is_initialized() and self.topo.world_size > 1:
             dist.broadcast(flat, src=0)
         st.stored = flat.clone()
+        if self._use_aps() and self.topo.is_leader:
+            aps = self._ensure_aps()
+            aps.register(key, st.stored)
+            aps.start()
 
     def _state(self, key) -> _KeyState:
         st = self.keys.get(key)
@@ -254,6 +282,13 @@ class KVStoreDist(KVStoreBase):
 
         if self.cfg.use_hfa:
             self._push_hfa(key, st, party_sum)
+            return
+
+        if self._use_aps():
+            # true-async: the leader hands its party sum to the global
+            # server and returns immediately — no cross-party wait
+            if topo.is_leader:
+                self._ensure_aps().push(key, party_sum)
             return
 
         # inter-party (WAN) tier — leaders only
@@ -451,7 +486,11 @@ class KVStoreDist(KVStoreBase):
     def pull(self, key, out: torch.Tensor, priority: int = 0) -> None:
         st = self._state(key)
         topo = self.topo
-        self._global_exchange_pull(key, st)
+        if self._use_aps():
+            if topo.is_leader:
+                st.stored = self._ensure_aps().pull(key)
+        else:
+            self._global_exchange_pull(key, st)
         # intra-party: leader broadcasts authoritative value to its workers
         if topo.world_size > 1 and topo.num_workers > 1:
             dist.broadcast(st.stored, src=topo.leader_rank,

@@ -1,0 +1,82 @@
+"""True-async (store-transport) global tier: parties proceed at their
+own pace; the global server applies every push on arrival."""
+
+import time
+
+import torch
+
+from dist_helpers import run_dist
+
+from geomx_amd import Config
+from geomx_amd.kvstore import create
+from geomx_amd.kvstore.optimizer import OptimizerSpec
+from geomx_amd.topology import init_topology
+
+
+def _mk_async(**over):
+    cfg = Config.from_env(num_parties=2, backend="gloo", device="cpu",
+                          mode="dist_async", async_transport="store", **over)
+    topo = init_topology(2, None, "gloo", "cpu")
+    return create("dist_async", cfg=cfg, topo=topo)
+
+
+def _fast_party_not_stalled(rank, world):
+    kv = _mk_async()
+    n = 256
+    kv.init("w", torch.zeros(n))
+    party = kv.topo.party_id
+    out = torch.empty(n)
+
+    fast_iters, slow_iters = 10, 3
+    t0 = time.perf_counter()
+    if party == 0:          # fast party
+        for _ in range(fast_iters):
+            kv.push("w", torch.ones(n))
+            kv.pull("w", out)
+        fast_dt = time.perf_counter() - t0
+        # slow party sleeps 0.25 s per iteration; synchronous collectives
+        # would drag us to >= 0.75 s
+        assert fast_dt < 0.6, fast_dt
+    else:                   # slow party
+        for _ in range(slow_iters):
+            time.sleep(0.25)
+            kv.push("w", torch.ones(n))
+            kv.pull("w", out)
+    kv.barrier()
+    # drain: server applies every push (no optimizer -> accumulate)
+    if kv._aps is not None and kv._aps.is_server:
+        assert kv._aps.drain(timeout_s=30)
+        assert kv._aps.applied == fast_iters + slow_iters
+    kv.barrier()
+    kv.pull("w", out)
+    # total = 13 pushes of ones, each a party sum over 2 workers = 2.0
+    assert torch.allclose(out, torch.full((n,), 2.0 * 13)), out[0]
+    kv.close()
+
+
+def test_async_store_no_stall_ws4():
+    run_dist(4, _fast_party_not_stalled)
+
+
+def _async_with_optimizer(rank, world):
+    kv = _mk_async()
+    kv.set_optimizer(OptimizerSpec("dcasgd", lr=0.01))
+    n = 64
+    kv.init("w", torch.ones(n))
+    out = torch.empty(n)
+    for _ in range(4):
+        kv.push("w", torch.full((n,), 0.5))
+        kv.pull("w", out)
+    kv.barrier()
+    if kv._aps is not None and kv._aps.is_server:
+        assert kv._aps.drain(timeout_s=30)
+        assert kv._aps.applied == 8  # 4 per party
+    kv.barrier()
+    kv.pull("w", out)
+    assert torch.isfinite(out).all()
+    assert (out != 1.0).any()  # moved from init
+    kv.close()
+
+
+def test_async_store_dcasgd_ws4():
+    run_dist(4, _async_with_optimizer)
