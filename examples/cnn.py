@@ -16,10 +16,14 @@ def main():
     p = base_parser()
     p.add_argument("--mixed-sync", action="store_true")
     p.add_argument("--dcasgd", action="store_true")
+    p.add_argument("--async-transport", type=str, default="lockstep",
+                   choices=["lockstep", "store"],
+                   help="store = true-async parameter server")
     args = p.parse_args()
 
     mode = "dist_async" if (args.mixed_sync or args.dcasgd) else "dist_sync"
-    kv, net, device = setup(args, mode=mode)
+    kv, net, device = setup(args, mode=mode,
+                            async_transport=args.async_transport)
     if args.dcasgd:
         kv.set_optimizer(OptimizerSpec("dcasgd", lr=args.learning_rate))
     else:
