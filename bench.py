@@ -23,6 +23,11 @@ import os
 import sys
 import time
 
+# fast MIOpen find: measured identical step perf to the exhaustive
+# default (75.0k vs 74.3k samples/s) with far cheaper warmup — matters
+# when 8 ranks run find concurrently on the same node
+os.environ.setdefault("MIOPEN_FIND_MODE", "3")
+
 import torch
 import torch.distributed as distmod
 
