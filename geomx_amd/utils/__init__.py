@@ -1,0 +1,1 @@
+from . import checkpoint, data, health, metrics, profiler  # noqa: F401
