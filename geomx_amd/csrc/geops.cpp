@@ -69,6 +69,14 @@ hipStream_t cur_stream() {
   return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
 }
 
+// surface async launch errors at the call site instead of a later,
+// unrelated synchronize
+void launch_check(const char* what) {
+  const hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, what, ": kernel launch failed: ",
+              hipGetErrorString(e));
+}
+
 void check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
@@ -88,6 +96,7 @@ void quantize_2bit(torch::Tensor grad, torch::Tensor residual,
   geops_quantize_2bit(grad.data_ptr<float>(), residual.data_ptr<float>(),
                       (uint32_t*)out.data_ptr<int32_t>(), n, (float)thr,
                       cur_stream());
+  launch_check("quantize_2bit");
 }
 
 void dequantize_2bit(torch::Tensor packed, torch::Tensor out, double thr) {
@@ -97,6 +106,7 @@ void dequantize_2bit(torch::Tensor packed, torch::Tensor out, double thr) {
   TORCH_CHECK(packed.numel() >= (n + 15) / 16);
   geops_dequantize_2bit((const uint32_t*)packed.data_ptr<int32_t>(),
                         out.data_ptr<float>(), n, (float)thr, cur_stream());
+  launch_check("dequantize_2bit");
 }
 
 void bsc_momentum(torch::Tensor g, torch::Tensor u, torch::Tensor v,
@@ -106,6 +116,7 @@ void bsc_momentum(torch::Tensor g, torch::Tensor u, torch::Tensor v,
   TORCH_CHECK(u.numel() == n && v.numel() == n);
   geops_bsc_momentum(g.data_ptr<float>(), u.data_ptr<float>(),
                      v.data_ptr<float>(), (float)mu, n, cur_stream());
+  launch_check("bsc_momentum");
 }
 
 torch::Tensor make_workspace(const torch::Tensor& like) {
@@ -125,6 +136,7 @@ void bsc_pack(torch::Tensor v, torch::Tensor u, torch::Tensor vals,
                  idx.data_ptr<int32_t>(), (long long*)ws.data_ptr<int64_t>(),
                  (float)boundary, v.numel(), vals.numel(),
                  (float)placeholder, /*zero_uv=*/true, cur_stream());
+  launch_check("bsc_pack");
 }
 
 void bsc_pull_pack(torch::Tensor x, torch::Tensor vals, torch::Tensor idx,
@@ -136,6 +148,7 @@ void bsc_pull_pack(torch::Tensor x, torch::Tensor vals, torch::Tensor idx,
                       idx.data_ptr<int32_t>(), (long long*)ws.data_ptr<int64_t>(),
                       x.numel(), vals.numel(), (float)placeholder,
                       cur_stream());
+  launch_check("bsc_pull_pack");
 }
 
 void bsc_unpack(torch::Tensor vals, torch::Tensor idx, torch::Tensor out,
@@ -149,6 +162,7 @@ void bsc_unpack(torch::Tensor vals, torch::Tensor idx, torch::Tensor out,
   geops_bsc_unpack(vals.data_ptr<float>(), idx.data_ptr<int32_t>(),
                    out.data_ptr<float>(), vals.numel(), accumulate,
                    cur_stream());
+  launch_check("bsc_unpack");
 }
 
 void dgt_contribution(torch::Tensor g, torch::Tensor out, int64_t chunk) {
@@ -158,6 +172,7 @@ void dgt_contribution(torch::Tensor g, torch::Tensor out, int64_t chunk) {
   TORCH_CHECK(out.numel() == nchunks);
   geops_dgt_contribution(g.data_ptr<float>(), out.data_ptr<float>(), n,
                          (int)chunk, nchunks, cur_stream());
+  launch_check("dgt_contribution");
 }
 
 void quantize_4bit(torch::Tensor x, torch::Tensor residual,
@@ -178,6 +193,7 @@ void quantize_4bit(torch::Tensor x, torch::Tensor residual,
                       has_res ? residual.data_ptr<float>() : nullptr,
                       packed.data_ptr<uint8_t>(), minmax.data_ptr<float>(),
                       n, (int)chunk, nchunks, has_res, cur_stream());
+  launch_check("quantize_4bit");
 }
 
 void dequantize_4bit(torch::Tensor packed, torch::Tensor minmax,
@@ -189,6 +205,7 @@ void dequantize_4bit(torch::Tensor packed, torch::Tensor minmax,
   geops_dequantize_4bit(packed.data_ptr<uint8_t>(), minmax.data_ptr<float>(),
                         out.data_ptr<float>(), out.numel(), (int)chunk,
                         cur_stream());
+  launch_check("dequantize_4bit");
 }
 
 void relu_maxpool2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor idx,
@@ -205,6 +222,7 @@ void relu_maxpool2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor idx,
                           (unsigned short*)out.data_ptr(),
                           idx.data_ptr<uint8_t>(), n_vec, Ho, Wo, (int)C,
                           cur_stream());
+  launch_check("relu_maxpool2_fwd");
 }
 
 void relu_maxpool2_bwd(torch::Tensor grad_out, torch::Tensor idx,
@@ -218,6 +236,7 @@ void relu_maxpool2_bwd(torch::Tensor grad_out, torch::Tensor idx,
                           idx.data_ptr<uint8_t>(),
                           (unsigned short*)grad_in.data_ptr(), n_vec_in,
                           (int)Hi, (int)Wi, (int)C, cur_stream());
+  launch_check("relu_maxpool2_bwd");
 }
 
 void conv5_nhwc(torch::Tensor in, torch::Tensor w_frags, torch::Tensor bias,
@@ -239,6 +258,7 @@ void conv5_nhwc(torch::Tensor in, torch::Tensor w_frags, torch::Tensor bias,
       (int)Wo, (int)CI, (int)CO, (int)pad, cur_stream());
   TORCH_CHECK(rc == 0, "conv5_nhwc: unsupported geometry CI=", CI,
               " CO=", CO, " pad=", pad);
+  launch_check("conv5_nhwc");
 }
 
 void conv5_wrw_nhwc(torch::Tensor in, torch::Tensor gout, torch::Tensor part,
@@ -255,6 +275,7 @@ void conv5_wrw_nhwc(torch::Tensor in, torch::Tensor gout, torch::Tensor part,
       (int)n_wg, cur_stream());
   TORCH_CHECK(rc > 0, "conv5_wrw_nhwc: unsupported geometry CI=", CI,
               " CO=", CO, " Wo=", Wo);
+  launch_check("conv5_wrw_nhwc");
 }
 
 void pad_ch3to4_nhwc(torch::Tensor in, torch::Tensor out, int64_t npix) {
@@ -264,6 +285,7 @@ void pad_ch3to4_nhwc(torch::Tensor in, torch::Tensor out, int64_t npix) {
   TORCH_CHECK(fp32 || in.scalar_type() == torch::kBFloat16);
   geops_pad_ch3to4_nhwc(in.data_ptr(), (unsigned short*)out.data_ptr(),
                         npix, fp32 ? 1 : 0, cur_stream());
+  launch_check("pad_ch3to4_nhwc");
 }
 
 void sgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
@@ -272,6 +294,7 @@ void sgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
   TORCH_CHECK(w.numel() == g.numel());
   geops_sgd_update(w.data_ptr<float>(), g.data_ptr<float>(), (float)lr,
                    (float)wd, (float)rescale, w.numel(), cur_stream());
+  launch_check("sgd_update");
 }
 
 void sgd_mom_update(torch::Tensor w, torch::Tensor g, torch::Tensor mom,
@@ -280,6 +303,7 @@ void sgd_mom_update(torch::Tensor w, torch::Tensor g, torch::Tensor mom,
   geops_sgd_mom_update(w.data_ptr<float>(), g.data_ptr<float>(),
                        mom.data_ptr<float>(), (float)lr, (float)momentum,
                        (float)wd, (float)rescale, w.numel(), cur_stream());
+  launch_check("sgd_mom_update");
 }
 
 void adam_update(torch::Tensor w, torch::Tensor g, torch::Tensor m,
@@ -293,6 +317,7 @@ void adam_update(torch::Tensor w, torch::Tensor g, torch::Tensor m,
                     m.data_ptr<float>(), v.data_ptr<float>(), (float)lr_t,
                     (float)beta1, (float)beta2, (float)eps, (float)wd,
                     (float)rescale, w.numel(), cur_stream());
+  launch_check("adam_update");
 }
 
 void dcasgd_update(torch::Tensor w, torch::Tensor g, torch::Tensor prev_w,
@@ -306,6 +331,7 @@ void dcasgd_update(torch::Tensor w, torch::Tensor g, torch::Tensor prev_w,
                       has_mom ? mom.data_ptr<float>() : nullptr, (float)lr,
                       (float)lamda, (float)momentum, (float)wd,
                       (float)rescale, w.numel(), has_mom, cur_stream());
+  launch_check("dcasgd_update");
 }
 
 void rmsprop_update(torch::Tensor w, torch::Tensor g, torch::Tensor n,
@@ -316,6 +342,7 @@ void rmsprop_update(torch::Tensor w, torch::Tensor g, torch::Tensor n,
                        n.data_ptr<float>(), (float)lr, (float)rho,
                        (float)eps, (float)wd, (float)rescale, w.numel(),
                        cur_stream());
+  launch_check("rmsprop_update");
 }
 
 void adagrad_update(torch::Tensor w, torch::Tensor g, torch::Tensor h,
@@ -324,6 +351,7 @@ void adagrad_update(torch::Tensor w, torch::Tensor g, torch::Tensor h,
   geops_adagrad_update(w.data_ptr<float>(), g.data_ptr<float>(),
                        h.data_ptr<float>(), (float)lr, (float)eps,
                        (float)wd, (float)rescale, w.numel(), cur_stream());
+  launch_check("adagrad_update");
 }
 
 void signsgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
@@ -331,6 +359,7 @@ void signsgd_update(torch::Tensor w, torch::Tensor g, double lr, double wd,
   check_f32(w, "w"); check_f32(g, "g");
   geops_signsgd_update(w.data_ptr<float>(), g.data_ptr<float>(), (float)lr,
                        (float)wd, (float)rescale, w.numel(), cur_stream());
+  launch_check("signsgd_update");
 }
 
 void signum_update(torch::Tensor w, torch::Tensor g, torch::Tensor mom,
@@ -339,6 +368,7 @@ void signum_update(torch::Tensor w, torch::Tensor g, torch::Tensor mom,
   geops_signum_update(w.data_ptr<float>(), g.data_ptr<float>(),
                       mom.data_ptr<float>(), (float)lr, (float)momentum,
                       (float)wd, (float)rescale, w.numel(), cur_stream());
+  launch_check("signum_update");
 }
 
 }  // namespace
