@@ -1,0 +1,90 @@
+"""Property-based tests (hypothesis) for the compression reference ops:
+shape-independent invariants that golden tests at fixed sizes miss."""
+
+import torch
+from hypothesis import given, settings, strategies as st
+
+from geomx_amd.ops import reference as ref
+
+
+@settings(max_examples=30, deadline=None)
+@given(n=st.integers(1, 2000), thr=st.floats(0.1, 4.0), seed=st.integers(0, 99))
+def test_2bit_error_feedback_invariant(n, thr, seed):
+    """emitted + residual == grad, for any size/threshold."""
+    g = torch.randn(n, generator=torch.Generator().manual_seed(seed)) * 2
+    r = torch.zeros(n)
+    packed = ref.quantize_2bit(g, r, thr)
+    d = ref.dequantize_2bit(packed, n, thr)
+    assert torch.allclose(d + r, g, atol=1e-5)
+    # emitted values are exactly {-thr, 0, +thr} up to fp32 rounding of thr
+    for u in d.unique().tolist():
+        assert u == 0.0 or abs(abs(u) - thr) < 1e-5
+
+
+@settings(max_examples=30, deadline=None)
+@given(n=st.integers(1, 2000), thr=st.floats(0.1, 4.0), seed=st.integers(0, 99),
+       steps=st.integers(1, 5))
+def test_2bit_multi_step_conservation(n, thr, seed, steps):
+    """over K steps: sum(emitted) + residual == sum(grads)."""
+    gen = torch.Generator().manual_seed(seed)
+    r = torch.zeros(n)
+    total_g = torch.zeros(n)
+    total_d = torch.zeros(n)
+    for _ in range(steps):
+        g = torch.randn(n, generator=gen)
+        total_g += g
+        total_d += ref.dequantize_2bit(ref.quantize_2bit(g, r, thr), n, thr)
+    assert torch.allclose(total_d + r, total_g, atol=1e-4)
+
+
+@settings(max_examples=25, deadline=None)
+@given(n=st.integers(200, 5000), ratio=st.floats(0.005, 0.3),
+       seed=st.integers(0, 99))
+def test_bsc_invariants(n, ratio, seed):
+    g = torch.randn(n, generator=torch.Generator().manual_seed(seed))
+    u = torch.zeros(n)
+    v = torch.zeros(n)
+    vals, idx = ref.bsc_compress(g, u, v, ratio)
+    k = ref.bsc_capacity(n, ratio)
+    assert vals.numel() == k and idx.numel() == k
+    sent = idx >= 0
+    ii = idx[sent].long()
+    # indices strictly increasing (index-order pack)
+    if ii.numel() > 1:
+        assert (ii[1:] > ii[:-1]).all()
+    # conservation on the first step (u=v=0 before): sent + v == g
+    out = ref.bsc_decompress(vals, idx, n)
+    assert torch.allclose(out + v, g, atol=1e-5)
+    # u zeroed exactly at sent coords
+    assert torch.all(u[ii] == 0)
+    # placeholders after the sent prefix
+    m = int(sent.sum())
+    assert (idx[m:] == -1).all()
+
+
+@settings(max_examples=25, deadline=None)
+@given(n=st.integers(2, 3000), seed=st.integers(0, 99))
+def test_4bit_bound(n, seed):
+    x = torch.randn(n, generator=torch.Generator().manual_seed(seed)) * 5
+    packed, lo, hi = ref.quantize_4bit(x)
+    y = ref.dequantize_4bit(packed, n, lo, hi)
+    step = max(hi - lo, 1e-30) / 16
+    assert (y - x).abs().max() <= step * 0.5 + 1e-5
+
+
+@settings(max_examples=20, deadline=None)
+@given(n=st.integers(10, 3000), cap_frac=st.floats(0.01, 1.5),
+       seed=st.integers(0, 99))
+def test_pull_pack_roundtrip(n, cap_frac, seed):
+    gen = torch.Generator().manual_seed(seed)
+    x = torch.zeros(n)
+    nz = max(1, n // 10)
+    pos = torch.randperm(n, generator=gen)[:nz]
+    x[pos] = torch.randn(nz, generator=gen) + 10  # nonzero for sure
+    cap = max(1, int(nz * cap_frac))
+    vals, idx = ref.bsc_pull_compress(x, cap)
+    y = ref.bsc_decompress(vals, idx, n)
+    sent = idx[idx >= 0].long()
+    assert torch.allclose(y[sent], x[sent])
+    if cap >= nz:  # full capacity -> lossless
+        assert torch.allclose(y, x)
