@@ -44,22 +44,35 @@ class DGTState:
         if self.contrib is None or self.contrib.numel() != contrib.numel():
             self.contrib = contrib
         else:
-            self.contrib = self.alpha * contrib + (1 - self.alpha) * self.contrib
+            # EWMA per the reference (kv_app.h:875): alpha weights the
+            # OLD value: c = alpha*c_old + (1-alpha)*mean|g|
+            self.contrib = (self.alpha * self.contrib
+                            + (1 - self.alpha) * contrib)
         n_keep = max(1, int(math.ceil(self.k * self.nchunks)))
         keep = torch.topk(self.contrib, n_keep).indices
         keep_mask = torch.zeros(self.nchunks, dtype=torch.bool,
                                 device=flat.device)
         keep_mask[keep] = True
+        # zero-contribution chunks are not transmitted at all (the
+        # reference drops them from the send, kv_app.h:973; the
+        # receiver zero-fills)
+        dead_mask = self.contrib == 0
+        n_dead = int(dead_mask.sum())
         packed, minmax = ops.quantize_4bit_chunked(flat, self.chunk,
                                                    self.residual)
         deq = ops.dequantize_4bit_chunked(packed, minmax.to(flat.device),
                                           self.numel, self.chunk)
-        elem_mask = keep_mask.repeat_interleave(self.chunk)[:self.numel]
-        out = torch.where(elem_mask, flat, deq.to(flat.device))
+        elem_keep = keep_mask.repeat_interleave(self.chunk)[:self.numel]
+        elem_dead = dead_mask.repeat_interleave(self.chunk)[:self.numel]
+        out = torch.where(elem_keep, flat, deq.to(flat.device))
+        out = torch.where(elem_dead, torch.zeros((), device=flat.device),
+                          out)
         # residual only meaningful for quantized chunks; zero it for
         # exact chunks (they carried no error)
-        self.residual[elem_mask] = 0.0
+        self.residual[elem_keep] = 0.0
+        n_lossy = self.nchunks - n_keep - max(0, n_dead - int(
+            (dead_mask & keep_mask).sum()))
         exact_bytes = n_keep * self.chunk * 4
-        lossy_bytes = (self.nchunks - n_keep) * (self.chunk // 2 + 8)
-        wire = min(self.numel * 4, exact_bytes) + max(0, lossy_bytes)
+        lossy_bytes = max(0, n_lossy) * (self.chunk // 2 + 8)
+        wire = min(self.numel * 4, exact_bytes) + lossy_bytes
         return out, wire

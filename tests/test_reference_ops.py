@@ -259,3 +259,37 @@ def test_signsgd_signum():
     ref.signum_update(w, torch.tensor([1.0, -1.0]), mom, lr=0.1, momentum=0.9)
     assert torch.allclose(mom, torch.tensor([0.1, -0.1]))
     assert torch.allclose(w, torch.tensor([-0.1, 0.1]))
+
+
+def test_dgt_state_semantics():
+    from geomx_amd.kvstore.dgt import DGTState
+    torch.manual_seed(6)
+    n = 1024
+    st = DGTState(n, "cpu", chunk_elems=128, k=0.5, alpha=0.3)
+    x = torch.randn(n)
+    out, wire = st.transform(x)
+    # important half exact
+    keep = torch.topk(st.contrib, 4).indices
+    for c in keep.tolist():
+        sl = slice(c * 128, (c + 1) * 128)
+        assert torch.allclose(out[sl], x[sl])
+    # lossy half close but not exact, bounded by chunk quantization step
+    assert torch.isfinite(out).all()
+    assert 0 < wire < n * 4
+    # EWMA weights OLD by alpha: after a second call with zero grads,
+    # contrib = 0.3 * old + 0.7 * 0
+    old = st.contrib.clone()
+    st.transform(torch.zeros(n))
+    assert torch.allclose(st.contrib, 0.3 * old, atol=1e-6)
+
+
+def test_dgt_zero_contribution_chunks_dropped():
+    from geomx_amd.kvstore.dgt import DGTState
+    n = 512
+    st = DGTState(n, "cpu", chunk_elems=128, k=0.25, alpha=0.3)
+    x = torch.zeros(n)
+    x[:128] = 1.0  # only chunk 0 carries signal
+    out, wire = st.transform(x)
+    # chunks 1..3 have zero contribution -> transmitted as zeros
+    assert torch.all(out[128:] == 0)
+    assert torch.allclose(out[:128], x[:128])
