@@ -90,8 +90,20 @@ class Config:
     # quantized at enable_dgt==3).
     enable_dgt: int = 0
     dgt_k: float = 0.5           # fraction of chunks deemed important (DMLC_K)
+    dgt_k_min: float = 0.2       # DMLC_K_MIN (lower bound; ADAPTIVE_K_FLAG is
+                                 # parsed but never acted on in the reference —
+                                 # kv_app.h:848 dead config — kept for parity)
+    dgt_channels: int = 1        # DMLC_UDP_CHANNEL_NUM (priority tiers)
     dgt_block_size: int = 4096   # bytes per chunk (DGT_BLOCK_SIZE)
     dgt_alpha: float = 0.3       # EWMA contribution factor (DGT_CONTRIBUTION_ALPHA)
+
+    # TSEngine toggle (ENABLE_INTER_TS / ENABLE_INTRA_TS): selects the
+    # incast-free replicated global tier (see kvstore/dist.py docstring)
+    enable_ts: bool = False
+    # P3 toggle (ENABLE_P3): priority scheduling is structural here
+    # (reverse-order buckets; big-key slicing via bigarray_bound), the
+    # flag is accepted for launch-script compatibility
+    enable_p3: bool = False
 
     # --- WAN emulation -------------------------------------------------
     # Bandwidth cap (Gbit/s) applied to inter-party (leader<->leader)
@@ -129,6 +141,11 @@ class Config:
             ),
             enable_dgt=_env_int(["ENABLE_DGT", "GEOMX_ENABLE_DGT"], 0),
             dgt_k=_env_float(["DMLC_K", "GEOMX_DGT_K"], 0.5),
+            dgt_k_min=_env_float(["DMLC_K_MIN"], 0.2),
+            dgt_channels=_env_int(["DMLC_UDP_CHANNEL_NUM"], 1),
+            enable_ts=bool(_env_int(["ENABLE_INTER_TS"], 0)
+                           or _env_int(["ENABLE_INTRA_TS"], 0)),
+            enable_p3=bool(_env_int(["ENABLE_P3"], 0)),
             dgt_block_size=_env_int(["DGT_BLOCK_SIZE", "GEOMX_DGT_BLOCK_SIZE"], 4096),
             dgt_alpha=_env_float(["DGT_CONTRIBUTION_ALPHA", "GEOMX_DGT_ALPHA"], 0.3),
             wan_gbps=_env_float(["GEOMX_WAN_GBPS"], 0.0),

@@ -27,9 +27,13 @@ def create(name: str = "dist_sync", cfg: Optional[Config] = None,
     Extra keyword arguments override Config fields (which themselves
     default from the GeoMX-compatible environment variables).
     """
-    global_mode = overrides.pop("global_mode", "sharded")
+    global_mode = overrides.pop("global_mode", None)
     if cfg is None:
         cfg = Config.from_env(**overrides)
+    if global_mode is None:
+        # ENABLE_INTER_TS/ENABLE_INTRA_TS select the incast-free
+        # replicated tier (the TSEngine role)
+        global_mode = "replicated" if cfg.enable_ts else "sharded"
     cfg.mode = name if name != "local" else "dist_sync"
     cfg.validate()
     return KVStoreDist(cfg, topo=topo, global_mode=global_mode)

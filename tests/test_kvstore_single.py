@@ -176,3 +176,12 @@ def test_push_row_sparse():
     assert torch.allclose(out[1], torch.full((3,), 2.0))  # dup id accumulated
     assert torch.allclose(out[4], torch.ones(3))
     assert out[0].abs().sum() == 0
+
+
+def test_ts_env_selects_replicated(monkeypatch):
+    monkeypatch.setenv("ENABLE_INTER_TS", "1")
+    kv = geomx_amd.kv.create("dist_sync")
+    assert kv.global_mode == "replicated"
+    monkeypatch.delenv("ENABLE_INTER_TS")
+    kv2 = geomx_amd.kv.create("dist_sync")
+    assert kv2.global_mode == "sharded"
