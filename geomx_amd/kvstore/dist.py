@@ -60,7 +60,7 @@ class _KeyState:
     __slots__ = ("shape", "numel", "dtype", "stored", "update_buf",
                  "bsc_u", "bsc_v", "residual_2bit", "milestone",
                  "push_count", "owner_party", "dgt_contrib", "dgt_residual",
-                 "sliced", "padded")
+                 "sliced", "padded", "residual_2bit_wan")
 
     def __init__(self, shape, numel, dtype):
         self.shape = shape
@@ -78,6 +78,7 @@ class _KeyState:
         self.owner_party = 0
         self.sliced = False   # P3/MultiGPS big-tensor slicing across leaders
         self.padded = 0       # padded numel (multiple of P) when sliced
+        self.residual_2bit_wan: Optional[torch.Tensor] = None  # leader WAN tier
 
 
 class _UpdaterAdapter:
@@ -352,8 +353,6 @@ class KVStoreDist(KVStoreBase):
             bound = int(self.compression.get("size_lower_bound",
                                              self.cfg.size_lower_bound))
             return "fp16" if st.numel < bound else "bsc"
-        if ctype == "2bit":
-            return None  # 2bit applies to the intra-party tier only
         return ctype
 
     def _global_exchange_push(self, key, st: _KeyState,
@@ -382,6 +381,25 @@ class KVStoreDist(KVStoreBase):
             dense = []
             for v_, i_ in zip(vlist, ilist):
                 dense.append(ops.bsc_decompress(v_, i_, st.numel))
+            return dense
+
+        if ctype == "2bit":
+            # the reference also compresses the local->global re-push
+            # (DataPushToGlobalServersCompressed,
+            # kvstore_dist_server.h:786): leaders keep their own WAN-side
+            # residual and exchange packed 2-bit words
+            thr = float(self.compression.get("threshold", self.cfg.threshold))
+            if st.residual_2bit_wan is None:
+                st.residual_2bit_wan = torch.zeros(st.numel,
+                                                   device=self._device)
+            packed = ops.quantize_2bit(party_sum, st.residual_2bit_wan, thr)
+            plist = [torch.empty_like(packed) for _ in range(P)]
+            dist.all_gather(plist, packed, group=group)
+            self.wan.charge(cross_party_bytes("all_gather",
+                                              packed.numel() * 4, P))
+            dense = []
+            for p_ in plist:
+                dense.append(ops.dequantize_2bit(p_, st.numel, thr))
             return dense
 
         if ctype == "fp16":
@@ -509,7 +527,8 @@ class KVStoreDist(KVStoreBase):
             return
         ctype = self._effective_ctype(st)
         need_wire = self.global_mode == "sharded" and not (
-            ctype in ("bsc", "fp16", "dgt") or self.cfg.mode == "dist_async")
+            ctype in ("bsc", "fp16", "dgt", "2bit")
+            or self.cfg.mode == "dist_async")
         if not need_wire:
             return  # all leaders already hold the result (replayed update)
         if not topo.is_leader:

@@ -370,8 +370,18 @@ def _2bit_two_tier(rank, world):
     kv.push("w", torch.full((n,), 0.7))  # each worker quantizes to +0.5
     out = torch.empty(n)
     kv.pull("w", out)
-    # party sums 2*0.5, inter-party dense sum = 2.0
-    assert torch.allclose(out, torch.full((n,), 0.5 * world)), (rank, out)
+    # party sum = 2*0.5 = 1.0; the WAN tier re-quantizes with thr 0.5:
+    # emits 2 codes worth 0.5 (residual 0); inter-party sum = 2.0...
+    # 2bit can only emit +-thr once per element per push, so 1.0 emits
+    # 0.5 and keeps 0.5 in the leader residual; sum over parties = 1.0
+    assert torch.allclose(out, torch.full((n,), 0.5 * 2)), (rank, out)
+    # second push flushes the residual: leaders now emit 0.5 (residual)
+    # + 0.5 (new party sum of 2x0.5... each worker: residual .2+.4=.6)
+    kv.push("w", torch.full((n,), 0.4))
+    kv.pull("w", out)
+    # workers: residual 0.2+0.4=0.6 -> emit 0.5 each; party sum 1.0;
+    # leader residual 0.5+1.0=1.5 -> emit 0.5 (single shot), keep 1.0
+    assert torch.allclose(out, torch.full((n,), 0.5 * 2)), (rank, out)
 
 
 def test_2bit_two_tier_ws4():
