@@ -555,8 +555,11 @@ class KVStoreDist(KVStoreBase):
             # gradient_compression.cc:271-308) — only meaningful for the
             # sharded aggregated-gradient pull; replicated mode skips the
             # wire entirely. Kept for the sharded path.
-            cap = ops.ref.bsc_capacity(st.numel,
-                                       float(self.compression["threshold"]), P)
+            # capacity N*thr*numWorkers*2 (kvstore_dist_server.h:1194 —
+            # the x2 slack absorbs non-overlapping worker supports),
+            # clamped to the dense size so the wire never regresses
+            cap = min(st.numel, ops.ref.bsc_capacity(
+                st.numel, float(self.compression["threshold"]), 2 * P))
             if topo.party_id == st.owner_party:
                 vals, idx = ops.bsc_pull_compress(st.stored, cap)
             else:
