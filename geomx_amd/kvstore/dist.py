@@ -112,6 +112,16 @@ class KVStoreDist(KVStoreBase):
         self.wan = TokenBucket(cfg.wan_gbps)
         self._device = self.topo.device
         self._aps = None  # true-async global tier (async_transport=store)
+        # TSEngine: throughput-scheduled relay tier replaces the
+        # all_gather for dense/fp16 payloads in replicated dist_sync
+        self._ts = None
+        if (cfg.enable_ts and global_mode == "replicated"
+                and self.topo.num_parties > 1 and self.topo.is_leader
+                and cfg.mode == "dist_sync"):
+            from .tsengine import TSExchange
+            self._ts = TSExchange(self.topo.leader_group,
+                                  self.topo.party_id,
+                                  self.topo.leader_ranks, wan=self.wan)
 
     # ------------------------------------------------------------------
     # properties (GeoMX API parity: kvstore.py:501-565)
@@ -403,6 +413,11 @@ class KVStoreDist(KVStoreBase):
             return dense
 
         if ctype == "fp16":
+            if self._ts is not None:
+                # TSEngine relay tree, fp16 wire: 2(P-1) point-to-point
+                # hops, no incast link
+                return [self._ts.allreduce_sum(party_sum,
+                                               wire_dtype=torch.float16)]
             h = party_sum.to(torch.float16)
             hlist = [torch.empty_like(h) for _ in range(P)]
             dist.all_gather(hlist, h, group=group)
@@ -443,6 +458,10 @@ class KVStoreDist(KVStoreBase):
             return None  # pull all_gathers the slices
 
         # dense fp32
+        if self._ts is not None:
+            # TSEngine (replicated dist_sync): scheduler-driven relay
+            # merge + spread instead of the all_gather
+            return [self._ts.allreduce_sum(party_sum)]
         if self.global_mode == "replicated" or self.cfg.mode == "dist_async":
             flist = [torch.empty_like(party_sum) for _ in range(P)]
             dist.all_gather(flist, party_sum, group=group)
