@@ -90,10 +90,32 @@ class AsyncPSGlobal:
 
     # -- lifecycle ------------------------------------------------------
     def register(self, key, init_value: torch.Tensor):
+        """Register a key. Elastic rejoin is implicit (ps-lite
+        `is_recovery` ADD_NODE semantics, van.cc:372-394: a recovered
+        node skips barriers and re-pulls live state): if the store
+        already carries state for this key, a restarted leader resumes
+        its push sequence where it left off, and a restarted server
+        adopts the last published parameters and the per-party consumed
+        counters instead of re-initializing. Optimizer state is not in
+        the store — reload it via load_optimizer_states if it matters."""
         n = init_value.numel()
         self.keys[key] = n
-        self._push_seq[key] = 0
+        # resume the outgoing counter: the store's seq is authoritative
+        # (0 for a fresh start). Without this a restarted leader would
+        # emit seq 1 against a server that already consumed past it,
+        # and every subsequent push would be silently ignored.
+        self._push_seq[key] = self._get_int(
+            self._k(key, f"p{self.topo.party_id}|seq"))
         if self.is_server:
+            ver = self._get_int(self._k(key, "ver"))
+            if ver >= 1:
+                # server rejoin: published params + acks survive us
+                payload = self.store.get(self._k(key, f"v{ver % 2}"))
+                self._stored[key] = _bytes_tensor(payload, n, self.device)
+                for p in range(self.topo.num_parties):
+                    self._seen[(key, p)] = self._get_int(
+                        self._k(key, f"p{p}|ack"))
+                return
             flat = init_value.detach().reshape(-1).float().to(self.device)
             self._stored[key] = flat.clone()
             self._publish(key)

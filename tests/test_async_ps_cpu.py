@@ -80,3 +80,56 @@ def _async_with_optimizer(rank, world):
 
 def test_async_store_dcasgd_ws4():
     run_dist(4, _async_with_optimizer)
+
+
+def _elastic_rejoin(rank, world):
+    """ps-lite is_recovery parity (van.cc:372-394): a restarted leader
+    resumes its push sequence; a restarted server adopts the published
+    params + consumed counters — nothing is lost or double-applied."""
+    import torch.distributed as dist
+    from geomx_amd.kvstore.async_ps import AsyncPSGlobal
+
+    kv = _mk_async()
+    n = 32
+    kv.init("w", torch.zeros(n))
+    out = torch.empty(n)
+    for _ in range(2):
+        kv.push("w", torch.ones(n))
+        kv.pull("w", out)
+    kv.barrier()
+
+    store = dist.distributed_c10d._get_default_store()
+    if kv._aps is not None and kv._aps.is_server:
+        assert kv._aps.drain(timeout_s=30)
+        kv._aps.stop()
+        # server restarts: a fresh endpoint on the same store must adopt
+        # the live state, NOT the (different) init value it is given
+        srv2 = AsyncPSGlobal(store, kv.topo, kv._device)
+        srv2.register("w", torch.full((n,), 5.0))
+        assert not torch.allclose(srv2._stored["w"], torch.full((n,), 5.0))
+        assert srv2._seen[("w", 0)] == 2 and srv2._seen[("w", 1)] == 2
+        srv2.start()
+        kv._aps = srv2
+    else:
+        # party-1 leader restarts: push counter resumes where it left off
+        cli2 = AsyncPSGlobal(store, kv.topo, kv._device)
+        cli2.register("w", torch.zeros(n))
+        assert cli2._push_seq["w"] == 2, cli2._push_seq
+        kv._aps = cli2
+    kv.barrier()
+
+    kv.push("w", torch.ones(n))
+    kv.pull("w", out)
+    kv.barrier()
+    if kv._aps.is_server:
+        assert kv._aps.drain(timeout_s=30)
+        assert kv._aps.applied == 2  # only the two post-restart pushes
+    kv.barrier()
+    kv.pull("w", out)
+    # 6 total pushes of a ones party-sum, accumulate mode
+    assert torch.allclose(out, torch.full((n,), 6.0)), out[0]
+    kv.close()
+
+
+def test_async_store_elastic_rejoin_ws2():
+    run_dist(2, _elastic_rejoin)
