@@ -192,17 +192,57 @@ class KVStoreDist(KVStoreBase):
             raise TypeError("updater must be callable")
         self.optimizer = _UpdaterAdapter(updater)
 
+    # CommandType, kvstore_dist_server.h:49-52 (the C++ enum; the
+    # reference's Python mirror kvstore.py:89-97 predates the inserted
+    # kSyncGlobalMode and is off by one from 4 up — we follow the C++
+    # server, which is what actually interprets the codes)
+    CMD_CONTROLLER = 0
+    CMD_SET_MULTI_PRECISION = 1
+    CMD_STOP_SERVER = 2
+    CMD_SYNC_MODE = 3
+    CMD_SYNC_GLOBAL_MODE = 4
+    CMD_SET_GRADIENT_COMPRESSION = 5
+    CMD_SET_PROFILER_PARAMS = 6
+
     def _send_command_to_servers(self, head: int, body: str) -> None:
         """Generic in-band server command (kvstore.py:644-661 parity).
         SPMD transport: rank 0 broadcasts; leader ranks (the 'servers')
-        receive it via the registered handler."""
+        interpret built-in CommandType codes, then any registered
+        handler sees the command too (kController user payloads)."""
         obj = [int(head), str(body)]
         if dist.is_initialized() and self.topo.world_size > 1:
             dist.broadcast_object_list(obj, src=0)
-        if self.topo.is_leader and self._server_command_handler is not None:
-            self._server_command_handler(obj[0], obj[1])
+        if self.topo.is_leader:
+            self._handle_builtin_command(obj[0], obj[1])
+            if self._server_command_handler is not None:
+                self._server_command_handler(obj[0], obj[1])
+
+    def _handle_builtin_command(self, head: int, body: str) -> None:
+        """Server-side CommandType dispatch (DataHandleEx command arm,
+        kvstore_dist_server.h:320-345)."""
+        if head == self.CMD_SYNC_MODE:
+            # servers boot async and are switched to sync by command
+            # (:329-330); body "0" flips back for symmetry
+            self.cfg.mode = "dist_sync" if body != "0" else "dist_async"
+        elif head == self.CMD_SYNC_GLOBAL_MODE:
+            # global tier sync toggle (:332-333). Our lockstep global
+            # tier shares cfg.mode; replaying the same flip keeps the
+            # two commands equivalent here.
+            self.cfg.mode = "dist_sync" if body != "0" else "dist_async"
+        elif head == self.CMD_STOP_SERVER:
+            self.close()
+        elif head == self.CMD_SET_MULTI_PRECISION:
+            # fp32 master copies are structural in this design (flat
+            # fp32 buckets / fp32 stored values) — accept and record
+            self._multi_precision = True
+        elif head == self.CMD_SET_GRADIENT_COMPRESSION:
+            import json as _json
+            self.set_gradient_compression(_json.loads(body))
+        # kController / kSetProfilerParams carry user payloads; the
+        # registered handler (below) is the consumer
 
     _server_command_handler = None
+    _multi_precision = False
 
     def set_server_command_handler(self, fn) -> None:
         self._server_command_handler = fn

@@ -165,6 +165,38 @@ def test_set_updater_and_server_command():
     assert cmds == [(7, "hello")]
 
 
+def test_builtin_server_commands():
+    """CommandType dispatch (kvstore_dist_server.h:320-345): sync-mode
+    flips, multi-precision flag, gradient compression by command."""
+    kv = make_kv()
+    kv.init("w", torch.ones(4))
+
+    kv._send_command_to_servers(kv.CMD_SYNC_MODE, "0")
+    assert kv.cfg.mode == "dist_async"
+    kv._send_command_to_servers(kv.CMD_SYNC_MODE, "1")
+    assert kv.cfg.mode == "dist_sync"
+    kv._send_command_to_servers(kv.CMD_SYNC_GLOBAL_MODE, "0")
+    assert kv.cfg.mode == "dist_async"
+    kv._send_command_to_servers(kv.CMD_SYNC_GLOBAL_MODE, "1")
+    assert kv.cfg.mode == "dist_sync"
+
+    assert not kv._multi_precision
+    kv._send_command_to_servers(kv.CMD_SET_MULTI_PRECISION, "1")
+    assert kv._multi_precision
+
+    kv._send_command_to_servers(kv.CMD_SET_GRADIENT_COMPRESSION,
+                                '{"type": "fp16"}')
+    assert kv.compression == {"type": "fp16"}
+
+    # built-ins still reach a registered handler (kController consumers)
+    cmds = []
+    kv.set_server_command_handler(lambda h, b: cmds.append(h))
+    kv._send_command_to_servers(kv.CMD_SET_MULTI_PRECISION, "1")
+    assert cmds == [kv.CMD_SET_MULTI_PRECISION]
+
+    kv._send_command_to_servers(kv.CMD_STOP_SERVER, "")
+
+
 def test_push_row_sparse():
     kv = make_kv()
     kv.init("emb", torch.zeros(6, 3))
