@@ -303,6 +303,15 @@ class KVStoreDist(KVStoreBase):
         if dist.is_initialized() and self.topo.world_size > 1:
             dist.broadcast(flat, src=0)
         st.stored = flat.clone()
+        if self.cfg.use_hfa:
+            # the reference seeds the milestone with the initial params
+            # (HandleHFAAccumulate first call, kvstore_dist_server.h:963
+            # — the init pull response copies stored into the
+            # milestone), so the first K2 exchange ships (params - w0)/P
+            # rather than full params. Results are identical under the
+            # synchronous exchange (milestone terms cancel), but the
+            # delta magnitude — what compression sees — matches.
+            st.milestone = flat.clone()
         if self._use_aps() and self.topo.is_leader:
             aps = self._ensure_aps()
             aps.register(key, st.stored)

@@ -217,3 +217,17 @@ def test_ts_env_selects_replicated(monkeypatch):
     monkeypatch.delenv("ENABLE_INTER_TS")
     kv2 = geomx_amd.kv.create("dist_sync")
     assert kv2.global_mode == "sharded"
+
+
+def test_hfa_milestone_seeded_from_init():
+    """The milestone starts at the initial params (reference
+    HandleHFAAccumulate first call, kvstore_dist_server.h:963), so the
+    first K2 exchange ships (params - w0)/P, not raw params."""
+    kv = make_kv(use_hfa=True, hfa_k2=1)
+    w0 = torch.full((6,), 3.0)
+    kv.init("w", w0)
+    assert torch.equal(kv.keys["w"].milestone, w0)
+    kv.push("w", torch.full((6,), 5.0))   # party param average
+    out = torch.empty(6)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((6,), 5.0))  # single party: avg
