@@ -361,3 +361,34 @@ def _dist_convergence(rank, world):
 
 def test_dist_convergence_ws2():
     run_dist(2, _dist_convergence, timeout=240)
+
+
+def test_resnet50_trainer_cpu_step():
+    """ResNet-50 (BASELINE.json config 5) steps through GeoTrainer on
+    CPU: shapes, bucket aliasing, and the optimizer path all hold for a
+    deep residual net, not just the example CNN."""
+    import torch
+    from geomx_amd import Config
+    from geomx_amd.kvstore.optimizer import OptimizerSpec
+    from geomx_amd.models import create_model
+    from geomx_amd.parallel import GeoTrainer
+    from geomx_amd.topology import init_topology
+
+    torch.manual_seed(0)
+    model = create_model("resnet50", num_classes=10)
+    cfg = Config.from_env(num_parties=1, backend="gloo", device="cpu")
+    topo = init_topology(1, None, "gloo", "cpu")
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec("sgd_mom", lr=0.01),
+                    mode="flat")
+    x = torch.randn(2, 3, 64, 64)
+    y = torch.randint(0, 10, (2,))
+    before = [p.clone() for p in list(model.parameters())[:3]]
+    for _ in range(2):
+        out = model(x)
+        loss = torch.nn.functional.cross_entropy(out, y)
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+    assert torch.isfinite(loss)
+    for p0, p in zip(before, list(model.parameters())[:3]):
+        assert not torch.equal(p0, p)
