@@ -133,3 +133,37 @@ def _elastic_rejoin(rank, world):
 
 def test_async_store_elastic_rejoin_ws2():
     run_dist(2, _elastic_rejoin)
+
+
+def _stress(rank, world):
+    """Race smoke (SURVEY §5.2 stance): many keys, interleaved pushes
+    and pulls from both parties while the server's consumer thread
+    applies concurrently — totals must come out exact."""
+    kv = _mk_async()
+    nkeys, iters, n = 6, 12, 17
+    for k in range(nkeys):
+        kv.init(k, torch.zeros(n))
+    out = torch.empty(n)
+    for it in range(iters):
+        for k in range(nkeys):
+            kv.push(k, torch.full((n,), float(k + 1)))
+        # pulls are party-collective (intra-DC stays synchronous in
+        # MixedSync) — the condition must be party-uniform
+        party = kv.topo.party_id
+        if it % 3 == party % 3:
+            kv.pull((it + party) % nkeys, out)  # reads racing the consumer
+    kv.barrier()
+    if kv._aps is not None and kv._aps.is_server:
+        assert kv._aps.drain(timeout_s=60)
+        assert kv._aps.applied == 2 * nkeys * iters
+    kv.barrier()
+    for k in range(nkeys):
+        kv.pull(k, out)
+        # 2 parties x iters pushes, party sum = 2 workers x (k+1)
+        expect = 2 * iters * 2.0 * (k + 1)
+        assert torch.allclose(out, torch.full((n,), expect)), (k, out[0])
+    kv.close()
+
+
+def test_async_store_stress_ws4():
+    run_dist(4, _stress)
