@@ -111,6 +111,16 @@ class GeoTrainer:
         self.spec = optimizer or OptimizerSpec(name="sgd", lr=0.01)
         self.server_opt = ServerOptimizer(self.spec)
         self._step = 0
+        # TSEngine relay tier for the leader exchange (dense/fp16 wires,
+        # synchronous mode only — the pipelined async tier needs async_op
+        # collectives)
+        self._ts = None
+        if (cfg.enable_ts and self.mode == "hips"
+                and topo.num_parties > 1 and topo.is_leader
+                and cfg.mode == "dist_sync"):
+            from ..kvstore.tsengine import TSExchange
+            self._ts = TSExchange(topo.leader_group, topo.party_id,
+                                  topo.leader_ranks, wan=self.wan)
         self._build_buckets()
         self._register_hooks()
 
@@ -323,11 +333,18 @@ class GeoTrainer:
                     dist.all_reduce(b.flat, group=topo.leader_group)
                     self.wan.charge(cross_party_bytes("all_reduce", wire, P))
                 elif ctype in ("fp16", "mpq"):
-                    h = b.flat.to(torch.float16)
-                    dist.all_reduce(h, group=topo.leader_group)
-                    self.wan.charge(cross_party_bytes(
-                        "all_reduce", h.numel() * 2, P))
-                    b.flat.copy_(h.float())
+                    if self._ts is not None:
+                        # TSEngine relay (charges per hop internally)
+                        b.flat.copy_(self._ts.allreduce_sum(
+                            b.flat, wire_dtype=torch.float16))
+                    else:
+                        h = b.flat.to(torch.float16)
+                        dist.all_reduce(h, group=topo.leader_group)
+                        self.wan.charge(cross_party_bytes(
+                            "all_reduce", h.numel() * 2, P))
+                        b.flat.copy_(h.float())
+                elif self._ts is not None:
+                    b.flat.copy_(self._ts.allreduce_sum(b.flat))
                 else:
                     dist.all_reduce(b.flat, group=topo.leader_group)
                     self.wan.charge(cross_party_bytes(

@@ -40,9 +40,9 @@ def _single_process_reference(steps=3, lr=0.05, world=2):
     return {k: v.detach().clone() for k, v in model.state_dict().items()}
 
 
-def _trainer_run(rank, world, mode, num_parties):
+def _trainer_run(rank, world, mode, num_parties, over=None):
     cfg = Config.from_env(num_parties=num_parties, backend="gloo",
-                          device="cpu", bucket_mb=1)
+                          device="cpu", bucket_mb=1, **(over or {}))
     topo = init_topology(cfg.num_parties, cfg.party_sizes, "gloo", "cpu")
     model = _tiny_model()
     tr = GeoTrainer(model, cfg, topo, OptimizerSpec(name="sgd", lr=0.05),
@@ -70,6 +70,12 @@ def test_flat_matches_single_process_ws2():
 def test_hips_sync_matches_single_process_ws4():
     # FSA with no compression is numerically identical to flat DP
     run_dist(4, _trainer_run, "hips", 2)
+
+
+def test_hips_tsengine_matches_single_process_ws4():
+    # ENABLE_TS swaps the leader all_reduce for the relay tree; the
+    # dense sum is the same numbers in a different order
+    run_dist(4, _trainer_run, "hips", 2, {"enable_ts": True})
 
 
 def _trainer_grad_views(rank, world):
@@ -392,3 +398,28 @@ def test_resnet50_trainer_cpu_step():
     assert torch.isfinite(loss)
     for p0, p in zip(before, list(model.parameters())[:3]):
         assert not torch.equal(p0, p)
+
+
+def _trainer_hips_ts_fp16(rank, world):
+    cfg = Config.from_env(num_parties=2, backend="gloo", device="cpu",
+                          bucket_mb=1, compression="fp16", enable_ts=True)
+    topo = init_topology(2, None, "gloo", "cpu")
+    model = _tiny_model()
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec(name="sgd", lr=0.05),
+                    mode="hips")
+    assert (tr._ts is not None) == topo.is_leader
+    for s in range(3):
+        x, y = _make_data(seed=11 + s)
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+    import torch.distributed as dist
+    for p in model.parameters():
+        ref = p.data.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(p.data, ref, atol=1e-6)
+
+
+def test_hips_tsengine_fp16_consistent_ws4():
+    run_dist(4, _trainer_hips_ts_fp16)
