@@ -238,8 +238,23 @@ class KVStoreDist(KVStoreBase):
         elif head == self.CMD_SET_GRADIENT_COMPRESSION:
             import json as _json
             self.set_gradient_compression(_json.loads(body))
-        # kController / kSetProfilerParams carry user payloads; the
-        # registered handler (below) is the consumer
+        elif head == self.CMD_SET_PROFILER_PARAMS:
+            # body "<verb>:<params>" with the KVStoreServerProfilerCommand
+            # verbs (set_config/state/pause/dump — kvstore_dist_server.h
+            # :409-456); executes on this leader's default profiler
+            from ..utils import profiler as _prof
+            verb, _, payload = body.partition(":")
+            v = int(verb)
+            if v == _prof.ServerProfilerCommand.SET_CONFIG:
+                _prof._default.set_config(filename=payload or None)
+            elif v == _prof.ServerProfilerCommand.STATE:
+                _prof._default.set_state(payload or "run")
+            elif v == _prof.ServerProfilerCommand.PAUSE:
+                _prof._default.pause()
+            elif v == _prof.ServerProfilerCommand.DUMP:
+                _prof._default.dump(rank=self.topo.rank)
+        # kController carries user payloads; the registered handler
+        # (below) is the consumer
 
     _server_command_handler = None
     _multi_precision = False

@@ -231,3 +231,23 @@ def test_hfa_milestone_seeded_from_init():
     out = torch.empty(6)
     kv.pull("w", out)
     assert torch.allclose(out, torch.full((6,), 5.0))  # single party: avg
+
+
+def test_profiler_params_command(tmp_path):
+    """kSetProfilerParams in-band command drives the server profiler
+    (kvstore_dist_server.h:409-456 rank-prefixed trace layout)."""
+    from geomx_amd.utils import profiler as prof
+
+    kv = make_kv()
+    trace = str(tmp_path / "srv.json")
+    kv._send_command_to_servers(
+        kv.CMD_SET_PROFILER_PARAMS,
+        f"{prof.ServerProfilerCommand.SET_CONFIG}:{trace}")
+    kv._send_command_to_servers(
+        kv.CMD_SET_PROFILER_PARAMS, f"{prof.ServerProfilerCommand.STATE}:run")
+    kv.init("w", torch.ones(8))
+    kv.push("w", torch.ones(8))
+    kv._send_command_to_servers(
+        kv.CMD_SET_PROFILER_PARAMS, f"{prof.ServerProfilerCommand.DUMP}:")
+    import os
+    assert os.path.exists(str(tmp_path / "rank0_srv.json"))
