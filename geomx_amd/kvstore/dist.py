@@ -298,7 +298,13 @@ class KVStoreDist(KVStoreBase):
     # ------------------------------------------------------------------
     # init
     # ------------------------------------------------------------------
-    def init(self, key, value: torch.Tensor) -> None:
+    def init(self, key, value) -> None:
+        # list-of-keys form (kvstore.py:118: "a string or int, or a
+        # list of them")
+        if isinstance(key, (list, tuple)):
+            for k, v in zip(key, value):
+                self.init(k, v)
+            return
         if key in self.keys:
             raise ValueError(f"key {key!r} already initialised")
         st = _KeyState(tuple(value.shape), value.numel(), value.dtype)
@@ -341,7 +347,20 @@ class KVStoreDist(KVStoreBase):
     # ------------------------------------------------------------------
     # push
     # ------------------------------------------------------------------
-    def push(self, key, value: torch.Tensor, priority: int = 0) -> None:
+    def push(self, key, value, priority: int = 0) -> None:
+        # list-of-keys form (kvstore.py:162)
+        if isinstance(key, (list, tuple)):
+            for k, v in zip(key, value):
+                self.push(k, v, priority)
+            return
+        # per-key list of values = this worker's multi-device grads,
+        # merged before the wire (the reference groups and sums
+        # per-device values, kvstore.py:205 / comm.h Reduce)
+        if isinstance(value, (list, tuple)):
+            acc = value[0].detach().reshape(-1).float().clone()
+            for v in value[1:]:
+                acc += v.detach().reshape(-1).float().to(acc.device)
+            value = acc
         st = self._state(key)
         if value.numel() != st.numel:
             raise ValueError(f"push size mismatch for {key!r}")
@@ -584,7 +603,19 @@ class KVStoreDist(KVStoreBase):
     # ------------------------------------------------------------------
     # pull
     # ------------------------------------------------------------------
-    def pull(self, key, out: torch.Tensor, priority: int = 0) -> None:
+    def pull(self, key, out, priority: int = 0) -> None:
+        # list-of-keys form (kvstore.py:242); a per-key list of outs
+        # broadcasts the same value into each (multi-device pull)
+        if isinstance(key, (list, tuple)):
+            for k, o in zip(key, out):
+                self.pull(k, o, priority)
+            return
+        if isinstance(out, (list, tuple)):
+            self.pull(key, out[0], priority)
+            for o in out[1:]:
+                with torch.no_grad():
+                    o.copy_(out[0].to(o.device))
+            return
         st = self._state(key)
         topo = self.topo
         if self._use_aps():

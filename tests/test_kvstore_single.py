@@ -251,3 +251,29 @@ def test_profiler_params_command(tmp_path):
         kv.CMD_SET_PROFILER_PARAMS, f"{prof.ServerProfilerCommand.DUMP}:")
     import os
     assert os.path.exists(str(tmp_path / "rank0_srv.json"))
+
+
+def test_list_key_and_multi_device_forms():
+    """List-of-keys and per-key value-list forms (kvstore.py:118,162,242:
+    keys may be lists; a value list for one key is this worker's
+    multi-device grads, merged before the wire)."""
+    kv = make_kv()
+    kv.init([0, 1], [torch.zeros(4), torch.zeros(6)])
+    kv.push([0, 1], [torch.ones(4), torch.full((6,), 2.0)])
+    o0, o1 = torch.empty(4), torch.empty(6)
+    kv.pull([0, 1], [o0, o1])
+    assert torch.allclose(o0, torch.ones(4))
+    assert torch.allclose(o1, torch.full((6,), 2.0))
+
+    # multi-device push: list of values for ONE key sums
+    kv2 = make_kv()
+    kv2.init("w", torch.zeros(5))
+    kv2.push("w", [torch.ones(5), torch.full((5,), 3.0)])
+    out = torch.empty(5)
+    kv2.pull("w", out)
+    assert torch.allclose(out, torch.full((5,), 4.0))
+
+    # multi-device pull: same value into every out
+    outs = [torch.empty(5), torch.empty(5)]
+    kv2.pull("w", outs)
+    assert torch.allclose(outs[0], outs[1])
