@@ -707,12 +707,19 @@ class KVStoreDist(KVStoreBase):
                          .to(self._device))
         self.push(key, dense, priority)
 
-    def row_sparse_pull(self, key, out: torch.Tensor, row_ids: torch.Tensor,
+    def row_sparse_pull(self, key, out, row_ids,
                         priority: int = 0) -> None:
         """Pull only the rows named by row_ids (python/mxnet/kvstore.py:316;
         GPU row-id dedup was cub Unique in the reference,
         kvstore_utils.cu:44-111 — torch.unique is the rocPRIM-backed
-        equivalent). `out` must be [len(row_ids), row_width]."""
+        equivalent). `out` must be [len(row_ids), row_width]; `out` and
+        `row_ids` may be aligned lists (multi-device form)."""
+        if isinstance(out, (list, tuple)):
+            if not isinstance(row_ids, (list, tuple)):
+                row_ids = [row_ids] * len(out)
+            for o, r in zip(out, row_ids):
+                self.row_sparse_pull(key, o, r, priority)
+            return
         st = self._state(key)
         if len(st.shape) < 2:
             raise ValueError("row_sparse_pull needs a >=2d key")
