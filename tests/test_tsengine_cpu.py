@@ -198,3 +198,73 @@ def _kv_ts_fp16(rank, world):
 
 def test_kv_tsengine_fp16_ws4():
     run_dist(4, _kv_ts_fp16)
+
+
+# ---------------------------------------------------------------------------
+# schedule validity properties (no dist): any P, any seed, any A state
+# ---------------------------------------------------------------------------
+
+def _mk_offline(P, seed):
+    """TSExchange with no process group — schedule methods only."""
+    ts = TSExchange.__new__(TSExchange)
+    ts.me = 0
+    ts.P = P
+    ts.sched = TSScheduler(P)
+    rng = random.Random(seed)
+    # random known/unknown throughput state
+    for i in range(P):
+        for j in range(P):
+            if i != j and rng.random() < 0.5:
+                ts.sched.update(i, j, rng.random() * 1e6)
+    return ts
+
+
+@pytest.mark.parametrize("P", [2, 3, 4, 5, 8, 16])
+@pytest.mark.parametrize("seed", [0, 1, 2])
+def test_spread_schedule_valid(P, seed):
+    ts = _mk_offline(P, seed)
+    src = seed % P
+    rounds = ts._spread_schedule(src, random.Random(seed))
+    have = {src}
+    for sends in rounds:
+        used = set()
+        for s, r in sends:
+            assert s in have          # only holders forward
+            assert r not in have      # receivers are new
+            assert s not in used and r not in used  # one role per round
+            used.add(s)
+            used.add(r)
+        have |= {r for _, r in sends}
+    assert have == set(range(P))      # everyone reached
+    # optimal depth: each round at most doubles the holders
+    import math
+    assert len(rounds) == math.ceil(math.log2(P)) if P > 1 else not rounds
+
+
+@pytest.mark.parametrize("P", [2, 3, 4, 5, 8, 16])
+@pytest.mark.parametrize("seed", [0, 1, 2])
+def test_merge_schedule_valid(P, seed):
+    ts = _mk_offline(P, seed)
+    rounds, root = ts._merge_schedule(random.Random(seed))
+    holders = set(range(P))
+    total_sends = 0
+    for sends in rounds:
+        used = set()
+        for s, r in sends:
+            assert s in holders and r in holders
+            assert s not in used and r not in used
+            used.add(s)
+            used.add(r)
+        holders -= {s for s, _ in sends}
+        total_sends += len(sends)
+    assert holders == {root}
+    assert total_sends == P - 1       # minimal message count
+
+
+def test_schedules_deterministic_for_same_state():
+    a = _mk_offline(8, 42)
+    b = _mk_offline(8, 42)
+    assert a._spread_schedule(3, random.Random(7)) == \
+        b._spread_schedule(3, random.Random(7))
+    assert a._merge_schedule(random.Random(7)) == \
+        b._merge_schedule(random.Random(7))
