@@ -268,3 +268,20 @@ def test_schedules_deterministic_for_same_state():
         b._spread_schedule(3, random.Random(7))
     assert a._merge_schedule(random.Random(7)) == \
         b._merge_schedule(random.Random(7))
+
+
+def _kv_ts_uneven(rank, world):
+    # parties [1, 3]: leaders are ranks 0 and 1; workers 2,3 in party 1
+    kv = _mk_kv(num_parties=2, party_sizes=[1, 3])
+    assert (kv._ts is not None) == kv.topo.is_leader
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    kv.init("w", torch.ones(7))
+    kv.push("w", torch.full((7,), 1.0))
+    out = torch.empty(7)
+    kv.pull("w", out)
+    # 4 workers total: w = 1 - 0.1*4
+    assert torch.allclose(out, torch.full((7,), 0.6), atol=1e-6), (rank, out)
+
+
+def test_kv_tsengine_uneven_parties_ws4():
+    run_dist(4, _kv_ts_uneven)
