@@ -31,10 +31,24 @@ peer can never hang training (the heartbeat layer reports it instead).
 This is intra-datacenter-synchronous / inter-datacenter-asynchronous —
 exactly the reference's MixedSync split. The WAN token bucket still
 prices every payload that crosses the party boundary.
+
+Message-loss tolerance (ps-lite Resender / PS_DROP_MSG parity,
+src/resender.h + van.cc:871-877): every push payload is tagged with its
+sequence number; `GEOMX_DROP_MSG=<pct>` makes the producer "lose" that
+fraction of transmissions (fault injection). The server validates the
+tag before consuming — a lost or stale slot stalls that party's stream
+(never corrupts it) — and the producer retransmits un-ACKed pushes
+after `GEOMX_RESEND_TIMEOUT_MS` (checked on every push/pull/flush). If
+a gap can never be repaired (the producer died with payloads in
+flight), the server skips it after `skip_timeout_s` and counts it in
+`self.lost` — the reference's lossy-channel stance: a lost gradient
+delays nobody.
 """
 
 from __future__ import annotations
 
+import os
+import random
 import threading
 import time
 from typing import Dict, Optional
@@ -60,7 +74,10 @@ class AsyncPSGlobal:
     party 0 additionally hosts the server (consumer thread + optimizer
     + authoritative parameters)."""
 
-    def __init__(self, store, topo, device, wan=None, poll_s: float = 0.001):
+    def __init__(self, store, topo, device, wan=None, poll_s: float = 0.001,
+                 drop_pct: Optional[float] = None,
+                 resend_timeout_s: Optional[float] = None,
+                 skip_timeout_s: float = 5.0):
         self.store = store
         self.topo = topo
         self.device = device
@@ -75,6 +92,20 @@ class AsyncPSGlobal:
         self.applied = 0
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
+        # Resender / fault injection (PS_DROP_MSG is a percentage)
+        if drop_pct is None:
+            drop_pct = float(os.environ.get("GEOMX_DROP_MSG", "0")) / 100.0
+        if resend_timeout_s is None:
+            resend_timeout_s = float(
+                os.environ.get("GEOMX_RESEND_TIMEOUT_MS", "50")) / 1000.0
+        self.drop_pct = drop_pct
+        self.resend_timeout_s = resend_timeout_s
+        self.skip_timeout_s = skip_timeout_s
+        self.lost = 0                 # server: pushes skipped as unrepairable
+        self._rng = random.Random(0xA5 + getattr(topo, "rank", 0))
+        # producer: (key, seq) -> [payload, last_tx_time]
+        self._unacked: Dict[tuple, list] = {}
+        self._stall: Dict[tuple, float] = {}  # server: (key,party) stall t0
 
     # -- key helpers ----------------------------------------------------
     def _k(self, key, suffix):
@@ -148,14 +179,61 @@ class AsyncPSGlobal:
             if time.time() > end:
                 raise TimeoutError(
                     f"async PS server lagging > {DEPTH} pushes on {key!r}")
+            # keep the resender alive while blocked: the window can be
+            # full precisely BECAUSE an earlier transmission was lost
+            self._resend_unacked()
             time.sleep(self.poll_s)
-        payload = _tensor_bytes(party_sum)
+        payload = seq.to_bytes(8, "little") + _tensor_bytes(party_sum)
         if self.wan is not None:
             self.wan.charge(len(payload), sync_device=False)
-        self.store.set(self._k(key, f"p{pid}|d{seq % DEPTH}"), payload)
-        self.store.set(self._k(key, f"p{pid}|seq"), str(seq).encode())
+        self._unacked[(key, seq)] = [payload, 0.0]
+        self._transmit(key, seq)
+        self._resend_unacked()
+
+    def _transmit(self, key, seq):
+        """One transmission attempt, subject to injected loss. The seq
+        announcement always carries the producer's HIGH-WATER mark so a
+        retransmission of an old slot never regresses the counter."""
+        pid = self.topo.party_id
+        rec = self._unacked.get((key, seq))
+        if rec is None:
+            return
+        rec[1] = time.time()
+        if self.drop_pct > 0 and self._rng.random() < self.drop_pct:
+            return  # "lost on the WAN" — announcement and data both
+        self.store.set(self._k(key, f"p{pid}|d{seq % DEPTH}"), rec[0])
+        self.store.set(self._k(key, f"p{pid}|seq"),
+                       str(self._push_seq[key]).encode())
+
+    def _resend_unacked(self):
+        """Resender (src/resender.h): retransmit pushes un-ACKed after
+        the timeout. Called on every push/pull/flush."""
+        if not self._unacked:
+            return
+        pid = self.topo.party_id
+        now = time.time()
+        acks: Dict[object, int] = {}
+        for (key, seq) in list(self._unacked):
+            ack = acks.get(key)
+            if ack is None:
+                ack = acks[key] = self._get_int(self._k(key, f"p{pid}|ack"))
+            if seq <= ack:
+                del self._unacked[(key, seq)]
+            elif now - self._unacked[(key, seq)][1] > self.resend_timeout_s:
+                self._transmit(key, seq)
+
+    def flush(self, timeout_s: float = 30.0) -> bool:
+        """Producer: block until every push of this party is ACKed
+        (resending as needed)."""
+        end = time.time() + timeout_s
+        while self._unacked and time.time() < end:
+            self._resend_unacked()
+            if self._unacked:
+                time.sleep(self.poll_s)
+        return not self._unacked
 
     def pull(self, key, timeout_s: float = 60.0) -> torch.Tensor:
+        self._resend_unacked()
         ver = self._get_int(self._k(key, "ver"))
         end = time.time() + timeout_s
         while ver < 1:  # server has not published the initial params yet
@@ -187,20 +265,37 @@ class AsyncPSGlobal:
                     while seen < seq:
                         # consume EVERY push in order (the async server
                         # applies each arriving push)
-                        seen += 1
-                        payload = self.store.get(
-                            self._k(key, f"p{p}|d{seen % DEPTH}"))
-                        grad = _bytes_tensor(payload, n, self.device)
-                        st = self._stored[key]
-                        if self.optimizer is not None:
-                            self.optimizer.update(key, st, grad)
-                        else:
-                            st.add_(grad)
+                        nxt = seen + 1
+                        slot = self._k(key, f"p{p}|d{nxt % DEPTH}")
+                        ok = False
+                        if self.store.check([slot]):
+                            payload = self.store.get(slot)
+                            tag = int.from_bytes(payload[:8], "little")
+                            ok = tag == nxt
+                        if not ok:
+                            # lost or stale slot: wait for the resender;
+                            # give up only on an unrepairable gap (dead
+                            # producer) and count the loss
+                            t0 = self._stall.setdefault((key, p), time.time())
+                            if time.time() - t0 > self.skip_timeout_s:
+                                self.lost += 1
+                            else:
+                                break
+                        self._stall.pop((key, p), None)
+                        seen = nxt
+                        if ok:
+                            grad = _bytes_tensor(payload[8:], n, self.device)
+                            st = self._stored[key]
+                            if self.optimizer is not None:
+                                self.optimizer.update(key, st, grad)
+                            else:
+                                st.add_(grad)
+                            self.applied += 1
                         self._seen[(key, p)] = seen
                         self.store.set(self._k(key, f"p{p}|ack"),
                                        str(seen).encode())
-                        self.applied += 1
-                        self._publish(key)
+                        if ok:
+                            self._publish(key)
                         progressed = True
             if not progressed:
                 time.sleep(self.poll_s)

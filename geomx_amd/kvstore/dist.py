@@ -293,6 +293,7 @@ class KVStoreDist(KVStoreBase):
 
     def close(self):
         if self._aps is not None:
+            self._aps.flush(timeout_s=10)  # deliver any un-ACKed pushes
             self._aps.stop()
 
     # ------------------------------------------------------------------

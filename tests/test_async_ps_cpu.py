@@ -167,3 +167,36 @@ def _stress(rank, world):
 
 def test_async_store_stress_ws4():
     run_dist(4, _stress)
+
+
+def _lossy_transport(rank, world):
+    """PS_DROP_MSG parity: 30% of push transmissions are dropped; the
+    resender (src/resender.h analog) retransmits un-ACKed pushes until
+    the server has applied every one — totals come out exact."""
+    kv = _mk_async()
+    n = 64
+    kv.init("w", torch.zeros(n))
+    aps = kv._aps
+    if aps is not None:
+        aps.drop_pct = 0.3
+        aps.resend_timeout_s = 0.02
+    out = torch.empty(n)
+    iters = 8
+    for _ in range(iters):
+        kv.push("w", torch.ones(n))
+        kv.pull("w", out)
+    if aps is not None:
+        assert aps.flush(timeout_s=30)   # resend until all ACKed
+    kv.barrier()
+    if aps is not None and aps.is_server:
+        assert aps.drain(timeout_s=30)
+        assert aps.applied == 2 * iters  # nothing lost, nothing doubled
+        assert aps.lost == 0
+    kv.barrier()
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((n,), 2.0 * 2 * iters)), out[0]
+    kv.close()
+
+
+def test_async_store_lossy_transport_ws4():
+    run_dist(4, _lossy_transport)
