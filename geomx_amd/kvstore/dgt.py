@@ -26,11 +26,19 @@ from .. import ops
 
 class DGTState:
     def __init__(self, numel: int, device, chunk_elems: int = 1024,
-                 k: float = 0.5, alpha: float = 0.3):
+                 k: float = 0.5, alpha: float = 0.3, mode: int = 3):
+        """mode mirrors ENABLE_DGT (van.cc:736-748 Unimportant_send):
+        1 = UDP lossy full-precision, 2 = TCP full-precision low
+        priority, 3 = TCP + 4-bit encode. Modes 1/2 keep unimportant
+        chunks exact on the wire (full bytes charged); only mode 3
+        quantizes them. Mode 1's packet loss has no RCCL analog — its
+        deterministic stand-in is the zero-contribution chunk drop that
+        all modes share (the receiver zero-fills, van.cc:356-366)."""
         self.numel = numel
         self.chunk = chunk_elems
         self.k = k
         self.alpha = alpha
+        self.mode = mode if mode in (1, 2, 3) else 3
         self.nchunks = (numel + chunk_elems - 1) // chunk_elems
         self.contrib = None
         self.residual = torch.zeros(numel, device=device)
@@ -58,21 +66,29 @@ class DGTState:
         # receiver zero-fills)
         dead_mask = self.contrib == 0
         n_dead = int(dead_mask.sum())
-        packed, minmax = ops.quantize_4bit_chunked(flat, self.chunk,
-                                                   self.residual)
-        deq = ops.dequantize_4bit_chunked(packed, minmax.to(flat.device),
-                                          self.numel, self.chunk)
         elem_keep = keep_mask.repeat_interleave(self.chunk)[:self.numel]
         elem_dead = dead_mask.repeat_interleave(self.chunk)[:self.numel]
-        out = torch.where(elem_keep, flat, deq.to(flat.device))
+        if self.mode >= 3:
+            packed, minmax = ops.quantize_4bit_chunked(flat, self.chunk,
+                                                       self.residual)
+            deq = ops.dequantize_4bit_chunked(packed,
+                                              minmax.to(flat.device),
+                                              self.numel, self.chunk)
+            out = torch.where(elem_keep, flat, deq.to(flat.device))
+            # residual only meaningful for quantized chunks; zero it for
+            # exact chunks (they carried no error)
+            self.residual[elem_keep] = 0.0
+            lossy_elem_bytes = self.chunk // 2 + 8
+        else:
+            # modes 1/2: unimportant chunks travel exact (low priority /
+            # lossy channel, but full precision on the wire)
+            out = flat.clone()
+            lossy_elem_bytes = self.chunk * 4
         out = torch.where(elem_dead, torch.zeros((), device=flat.device),
                           out)
-        # residual only meaningful for quantized chunks; zero it for
-        # exact chunks (they carried no error)
-        self.residual[elem_keep] = 0.0
         n_lossy = self.nchunks - n_keep - max(0, n_dead - int(
             (dead_mask & keep_mask).sum()))
         exact_bytes = n_keep * self.chunk * 4
-        lossy_bytes = max(0, n_lossy) * (self.chunk // 2 + 8)
-        wire = min(self.numel * 4, exact_bytes) + lossy_bytes
+        wire = min(self.numel * 4, exact_bytes) \
+            + max(0, n_lossy) * lossy_elem_bytes
         return out, wire

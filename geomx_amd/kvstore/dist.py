@@ -765,7 +765,8 @@ class KVStoreDist(KVStoreBase):
             from .dgt import DGTState
             dg = DGTState(st.numel, x.device,
                           chunk_elems=max(64, self.cfg.dgt_block_size // 4),
-                          k=self.cfg.dgt_k, alpha=self.cfg.dgt_alpha)
+                          k=self.cfg.dgt_k, alpha=self.cfg.dgt_alpha,
+                          mode=self.cfg.enable_dgt or 3)
             self._dgt_states[key] = dg
         out, wire = dg.transform(x)
         self._dgt_wire = wire

@@ -327,7 +327,8 @@ class GeoTrainer:
                                          chunk_elems=max(
                                              64, self.cfg.dgt_block_size // 4),
                                          k=self.cfg.dgt_k,
-                                         alpha=self.cfg.dgt_alpha)
+                                         alpha=self.cfg.dgt_alpha,
+                                         mode=self.cfg.enable_dgt or 3)
                     lossy, wire = b.dgt.transform(b.flat)
                     b.flat.copy_(lossy)
                     dist.all_reduce(b.flat, group=topo.leader_group)

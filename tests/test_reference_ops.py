@@ -293,3 +293,33 @@ def test_dgt_zero_contribution_chunks_dropped():
     # chunks 1..3 have zero contribution -> transmitted as zeros
     assert torch.all(out[128:] == 0)
     assert torch.allclose(out[:128], x[:128])
+
+
+def test_dgt_mode_distinctions():
+    """ENABLE_DGT modes (van.cc:736-748): 1/2 keep unimportant chunks
+    exact (full wire bytes), 3 quantizes them to 4 bits."""
+    from geomx_amd.kvstore.dgt import DGTState
+    torch.manual_seed(5)
+    n, chunk = 1024, 128
+    g = torch.randn(n) * torch.repeat_interleave(
+        torch.tensor([5.0, 0.1, 4.0, 0.2, 3.0, 0.3, 2.0, 0.4]), chunk)
+
+    st2 = DGTState(n, "cpu", chunk_elems=chunk, k=0.5, mode=2)
+    out2, wire2 = st2.transform(g)
+    # mode 2: everything full precision (important exact, unimportant
+    # exact-but-low-priority) — reconstruction is exact
+    assert torch.allclose(out2, g)
+    assert wire2 == n * 4
+
+    st3 = DGTState(n, "cpu", chunk_elems=chunk, k=0.5, mode=3)
+    out3, wire3 = st3.transform(g)
+    keep = (out3 == g).float().mean()
+    assert 0.4 < keep < 0.9          # important chunks exact
+    assert not torch.allclose(out3, g)   # unimportant chunks quantized
+    assert wire3 < wire2                 # 4-bit tier shrinks the wire
+    # quantization error bounded by the per-chunk 4-bit step
+    assert (out3 - g).abs().max() < g.abs().max() / 4
+
+    st1 = DGTState(n, "cpu", chunk_elems=chunk, k=0.5, mode=1)
+    out1, wire1 = st1.transform(g)
+    assert torch.allclose(out1, g) and wire1 == n * 4
