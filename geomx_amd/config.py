@@ -110,6 +110,13 @@ class Config:
     # traffic via a token bucket, so Bi-Sparse/MPQ/DGT speedups are
     # measurable in-node. 0 = uncapped.
     wan_gbps: float = 0.0
+    # Optional per-party uplink rates (Gbit/s, one per party) for
+    # HETEROGENEOUS WAN emulation — the regime TSEngine's relay
+    # scheduling targets (a slow data center's link is visited exactly
+    # once per relay instead of sitting on the collective's ring).
+    # Overrides wan_gbps for each listed party; GEOMX_PARTY_WAN_GBPS is
+    # the comma-separated env form.
+    party_wan_gbps: Optional[List[float]] = None
 
     # --- runtime -------------------------------------------------------
     # P3/MultiGPS big-tensor slicing: keys with numel >= bigarray_bound
@@ -157,6 +164,9 @@ class Config:
         sizes = _env_str(["GEOMX_PARTY_SIZES"], None)
         if sizes:
             cfg.party_sizes = [int(x) for x in sizes.split(",") if x]
+        rates = _env_str(["GEOMX_PARTY_WAN_GBPS"], None)
+        if rates:
+            cfg.party_wan_gbps = [float(x) for x in rates.split(",") if x]
         for k, v in overrides.items():
             if not hasattr(cfg, k):
                 raise ValueError(f"unknown config field {k!r}")
@@ -177,4 +187,14 @@ class Config:
             raise ValueError("hfa_k1/k2 >= 1")
         if self.async_transport not in ("lockstep", "store"):
             raise ValueError("async_transport must be lockstep|store")
+        if self.party_wan_gbps is not None \
+                and len(self.party_wan_gbps) != self.num_parties:
+            raise ValueError("party_wan_gbps needs one rate per party")
         return self
+
+    def wan_rate_for(self, party_id: int) -> float:
+        """This party's emulated WAN uplink rate (Gbit/s; 0 = uncapped).
+        Heterogeneous per-party rates override the global wan_gbps."""
+        if self.party_wan_gbps is not None:
+            return float(self.party_wan_gbps[party_id])
+        return self.wan_gbps

@@ -109,7 +109,7 @@ class KVStoreDist(KVStoreBase):
         self._key_order: List[object] = []
         self.optimizer: Optional[ServerOptimizer] = None
         self.compression: Optional[Dict] = None
-        self.wan = TokenBucket(cfg.wan_gbps)
+        self.wan = TokenBucket(cfg.wan_rate_for(self.topo.party_id))
         self._device = self.topo.device
         self._aps = None  # true-async global tier (async_transport=store)
         # TSEngine: throughput-scheduled relay tier replaces the

@@ -106,7 +106,7 @@ class GeoTrainer:
         self.cfg = cfg
         self.topo = topo
         self.mode = mode if topo.world_size > 1 else "flat"
-        self.wan = TokenBucket(cfg.wan_gbps)
+        self.wan = TokenBucket(cfg.wan_rate_for(topo.party_id))
         self.device = topo.device
         self.spec = optimizer or OptimizerSpec(name="sgd", lr=0.01)
         self.server_opt = ServerOptimizer(self.spec)

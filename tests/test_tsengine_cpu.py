@@ -285,3 +285,24 @@ def _kv_ts_uneven(rank, world):
 
 def test_kv_tsengine_uneven_parties_ws4():
     run_dist(4, _kv_ts_uneven)
+
+
+def _kv_ts_hetero_wan(rank, world):
+    """Per-party WAN rates feed the scheduler: the slow party's uplink
+    measures slower, so greedy rounds route around it."""
+    kv = _mk_kv(num_parties=4,
+                party_wan_gbps=[1.0, 1.0, 1.0, 0.02])
+    kv.init("w", torch.zeros(4096))
+    for s in range(6):
+        kv.push("w", torch.full((4096,), float(s)))
+    A = kv._ts.sched.A
+    slow_tx = [A[3][j] for j in range(4) if j != 3 and A[3][j] >= 0]
+    fast_tx = [A[i][j] for i in range(3) for j in range(4)
+               if i != j and A[i][j] >= 0]
+    assert slow_tx and fast_tx
+    import statistics
+    assert statistics.median(fast_tx) > 5 * max(slow_tx), (slow_tx, fast_tx)
+
+
+def test_kv_tsengine_heterogeneous_wan_ws4():
+    run_dist(4, _kv_ts_hetero_wan)

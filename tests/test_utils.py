@@ -255,3 +255,23 @@ def test_recordio_corruption_detected(tmp_path):
     rd = RecordDataset(path)
     with pytest.raises(IOError):
         rd[0]
+
+
+def test_party_wan_gbps_config(monkeypatch):
+    """Heterogeneous per-party WAN rates: env parsing, validation, and
+    rate lookup."""
+    import pytest
+    from geomx_amd import Config
+
+    monkeypatch.setenv("GEOMX_PARTY_WAN_GBPS", "1.0,0.25")
+    cfg = Config.from_env(num_parties=2)
+    assert cfg.party_wan_gbps == [1.0, 0.25]
+    assert cfg.wan_rate_for(0) == 1.0 and cfg.wan_rate_for(1) == 0.25
+    monkeypatch.delenv("GEOMX_PARTY_WAN_GBPS")
+
+    cfg2 = Config.from_env(num_parties=2, wan_gbps=3.0)
+    assert cfg2.wan_rate_for(0) == 3.0 == cfg2.wan_rate_for(1)
+
+    with pytest.raises(ValueError):
+        Config.from_env(num_parties=3,
+                        party_wan_gbps=[1.0, 2.0]).validate()
