@@ -57,6 +57,9 @@ def parse_args():
                    choices=[None, "bsc", "fp16", "mpq", "2bit", "dgt"])
     p.add_argument("--bsc-ratio", type=float, default=0.01)
     p.add_argument("--wan-gbps", type=float, default=0.0)
+    p.add_argument("--party-wan-gbps", type=str, default=None,
+                   help="comma-separated per-party uplink Gbit/s "
+                        "(heterogeneous WAN; overrides --wan-gbps)")
     p.add_argument("--bucket-mb", type=int, default=25)
     p.add_argument("--optimizer", type=str, default="sgd_mom")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
@@ -91,11 +94,13 @@ def main():
     # same emulated inter-DC link
     topo_parties = parties if args.mode == "hips" and world > 1 else 1
 
+    pw = [float(x) for x in args.party_wan_gbps.split(",")] \
+        if args.party_wan_gbps else None
     cfg = Config.from_env(
         num_parties=parties, backend=backend, mode=args.sync_mode,
         compression=args.compress, bsc_ratio=args.bsc_ratio,
-        wan_gbps=args.wan_gbps, bucket_mb=args.bucket_mb,
-        comm_dtype=args.comm_dtype)
+        wan_gbps=args.wan_gbps, party_wan_gbps=pw,
+        bucket_mb=args.bucket_mb, comm_dtype=args.comm_dtype)
     topo = init_topology(topo_parties, None, backend)
     device = topo.device
 
