@@ -88,3 +88,42 @@ def test_pull_pack_roundtrip(n, cap_frac, seed):
     assert torch.allclose(y[sent], x[sent])
     if cap >= nz:  # full capacity -> lossless
         assert torch.allclose(y, x)
+
+
+@settings(max_examples=20, deadline=None)
+@given(n=st.integers(64, 4000), chunk=st.sampled_from([64, 128, 256]),
+       k=st.floats(0.1, 0.9), seed=st.integers(0, 99))
+def test_dgt_state_invariants(n, chunk, k, seed):
+    """Any size/chunk/k: important chunks pass through exactly, wire
+    bytes never exceed dense, zero chunks drop to zero."""
+    from geomx_amd.kvstore.dgt import DGTState
+    g = torch.randn(n, generator=torch.Generator().manual_seed(seed))
+    dgt = DGTState(n, "cpu", chunk_elems=chunk, k=k)
+    out, wire = dgt.transform(g)
+    assert out.shape == g.shape
+    assert wire <= n * 4 + dgt.nchunks * 8  # dense + per-chunk minmax
+    # at least ceil(k * nchunks) chunks exact (ignoring ties): the
+    # element-exact fraction is at least k minus one chunk
+    exact_elems = int((out == g).sum())
+    assert exact_elems >= int(k * dgt.nchunks - 1) * min(chunk, n)
+    assert torch.isfinite(out).all()
+
+
+@settings(max_examples=20, deadline=None)
+@given(n=st.integers(1, 200), shape=st.sampled_from([(3,), (2, 5), (1, 2, 3)]),
+       seed=st.integers(0, 99))
+def test_recordio_property_roundtrip(n, shape, seed, tmp_path_factory):
+    """Arbitrary record counts/shapes round-trip exactly."""
+    from geomx_amd.utils.recordio import RecordDataset, RecordWriter
+    d = tmp_path_factory.mktemp("rec")
+    path = str(d / "t.rec")
+    gen = torch.Generator().manual_seed(seed)
+    ts = [torch.randn(*shape, generator=gen) for _ in range(n)]
+    with RecordWriter(path) as w:
+        for i, t in enumerate(ts):
+            w.write(t, label=i)
+    rd = RecordDataset(path)
+    assert len(rd) == n
+    for i in [0, n // 2, n - 1]:
+        x, y = rd[i]
+        assert torch.equal(x, ts[i]) and y == i
