@@ -665,27 +665,18 @@ class KVStoreDist(KVStoreBase):
             self.wan.charge(cross_party_bytes("all_gather", chunk * 4, P))
             st.stored = torch.cat(parts)[:st.numel]
             return
-        if ctype == "bsc" and self.optimizer is None:
-            # pull-side re-sparsification (BSCPullCompress,
-            # gradient_compression.cc:271-308) — only meaningful for the
-            # sharded aggregated-gradient pull; replicated mode skips the
-            # wire entirely. Kept for the sharded path.
-            # capacity N*thr*numWorkers*2 (kvstore_dist_server.h:1194 —
-            # the x2 slack absorbs non-overlapping worker supports),
-            # clamped to the dense size so the wire never regresses
-            cap = min(st.numel, ops.ref.bsc_capacity(
-                st.numel, float(self.compression["threshold"]), 2 * P))
-            if topo.party_id == st.owner_party:
-                vals, idx = ops.bsc_pull_compress(st.stored, cap)
-            else:
-                vals = torch.empty(cap, device=self._device)
-                idx = torch.empty(cap, dtype=torch.int32, device=self._device)
-            dist.broadcast(vals, src=owner_leader, group=group)
-            dist.broadcast(idx, src=owner_leader, group=group)
-            self.wan.charge(cross_party_bytes("broadcast", cap * 8, P))
-            if topo.party_id != st.owner_party:
-                ops.bsc_decompress(vals, idx, st.numel, out=st.stored)
-            return
+        # NOTE on pull-side BSC (BSCPullCompress,
+        # gradient_compression.cc:271-308, response sizing
+        # kvstore_dist_server.h:1194): the reference needs it because
+        # its global server is a star — the aggregated gradient must
+        # travel BACK over the WAN to every local server. Here every
+        # compressed exchange is an all_gather of the per-party
+        # payloads, so all leaders already hold the aggregate and the
+        # pull direction has NO wire at all (need_wire above) — the
+        # "bi-directional" saving is structural. The op itself
+        # (ops.bsc_pull_compress + k_bsc_pull_pack) remains for star
+        # topologies and is golden-tested against the reference
+        # semantics on CPU and gfx950.
         dist.broadcast(st.stored, src=owner_leader, group=group)
         self.wan.charge(cross_party_bytes("broadcast", st.numel * 4, P))
 
