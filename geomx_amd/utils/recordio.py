@@ -29,6 +29,11 @@ import numpy as np
 import torch
 from torch.utils.data import Dataset
 
+try:  # native reader (csrc/recordio.cpp): mmap views + threaded batch
+    from .. import _geoio
+except ImportError:  # pragma: no cover - built by setup.py
+    _geoio = None
+
 _MAGIC = 0xCED7EC0D
 _HEADER = struct.Struct("<IIQ")          # magic, flag, payload_len
 _DTYPES = {
@@ -112,16 +117,44 @@ def pack_dataset(dataset, path: str,
 
 
 class RecordDataset(Dataset):
-    """mmap-backed random-access reader; safe to use from forked/spawned
-    DataLoader workers (each lazily opens its own map)."""
+    """Random-access reader; safe to use from forked/spawned DataLoader
+    workers (each lazily opens its own map). Reads go through the
+    native C++ reader (`geomx_amd._geoio`, csrc/recordio.cpp — the
+    reference's C++ IO path analog) when it is built; `native=False`
+    forces the pure-Python mmap fallback. `read_batch` assembles a
+    whole same-shape batch in one call with a multi-threaded copy."""
 
-    def __init__(self, path: str):
+    def __init__(self, path: str, native: Optional[bool] = None):
         self.path = path
         idx_path = path + ".idx"
         if not os.path.exists(idx_path):
             raise FileNotFoundError(idx_path)
         self.offsets = np.fromfile(idx_path, dtype=np.int64)
         self._mm = None
+        self.native = (_geoio is not None) if native is None else native
+        if self.native and _geoio is None:
+            raise RuntimeError("native reader requested but _geoio "
+                               "extension is not built")
+        self._nf = None
+
+    def _native_file(self):
+        if self._nf is None:
+            self._nf = _geoio.RecordFile(self.path)
+        return self._nf
+
+    def read_batch(self, indices, threads: int = 4,
+                   pin_memory: bool = False):
+        """Assemble records `indices` (same shape/dtype) into one
+        [N, *shape] tensor + [N] int64 labels."""
+        idx = [int(i) for i in indices]
+        if self.native:
+            return self._native_file().read_batch(idx, threads,
+                                                  pin_memory)
+        xs, ys = zip(*(self[i] for i in idx))
+        batch = torch.stack(xs)
+        if pin_memory:
+            batch = batch.pin_memory()
+        return batch, torch.tensor(ys, dtype=torch.int64)
 
     def _map(self) -> mmap.mmap:
         if self._mm is None:
@@ -134,6 +167,8 @@ class RecordDataset(Dataset):
         return len(self.offsets)
 
     def __getitem__(self, i: int):
+        if self.native:
+            return self._native_file().read(int(i))
         mm = self._map()
         off = int(self.offsets[i])
         magic, _flag, plen = _HEADER.unpack_from(mm, off)
@@ -146,6 +181,8 @@ class RecordDataset(Dataset):
     def labels(self) -> torch.Tensor:
         """All labels (one header read per record) — feeds
         ClassSplitSampler without materializing the tensors."""
+        if self.native:
+            return self._native_file().labels()
         mm = self._map()
         out = torch.empty(len(self), dtype=torch.int64)
         for i, off in enumerate(self.offsets):
@@ -153,8 +190,10 @@ class RecordDataset(Dataset):
             out[i] = struct.unpack_from("<q", mm, start)[0]
         return out
 
-    # pickling (DataLoader spawn workers): drop the map, reopen lazily
+    # pickling (DataLoader spawn workers): drop the handles, reopen
+    # lazily in the worker
     def __getstate__(self):
         d = dict(self.__dict__)
         d["_mm"] = None
+        d["_nf"] = None
         return d

@@ -12,7 +12,8 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+from torch.utils.cpp_extension import (BuildExtension,  # noqa: E402
+                                       CppExtension, CUDAExtension)
 
 here = os.path.dirname(os.path.abspath(__file__))
 
@@ -30,11 +31,18 @@ ext = CUDAExtension(
     },
 )
 
+# host-side IO runtime (record-file reader) — plain C++, no HIP
+io_ext = CppExtension(
+    name="geomx_amd._geoio",
+    sources=["geomx_amd/csrc/recordio.cpp"],
+    extra_compile_args=["-O3", "-std=c++17"],
+)
+
 setup(
     name="geomx_amd",
     version="0.1.0",
     packages=["geomx_amd", "geomx_amd.kvstore", "geomx_amd.ops",
               "geomx_amd.parallel", "geomx_amd.models", "geomx_amd.utils"],
-    ext_modules=[ext],
+    ext_modules=[ext, io_ext],
     cmdclass={"build_ext": BuildExtension},
 )

@@ -252,9 +252,12 @@ def test_recordio_corruption_detected(tmp_path):
     with open(path, "r+b") as f:
         f.seek(0)
         f.write(b"\x00\x00\x00\x00")  # clobber magic
-    rd = RecordDataset(path)
+    # both readers must refuse the corrupt frame (native raises
+    # RuntimeError via pybind, the python path raises IOError)
+    with pytest.raises((IOError, RuntimeError)):
+        RecordDataset(path)[0]
     with pytest.raises(IOError):
-        rd[0]
+        RecordDataset(path, native=False)[0]
 
 
 def test_party_wan_gbps_config(monkeypatch):
@@ -275,3 +278,41 @@ def test_party_wan_gbps_config(monkeypatch):
     with pytest.raises(ValueError):
         Config.from_env(num_parties=3,
                         party_wan_gbps=[1.0, 2.0]).validate()
+
+
+def test_recordio_native_matches_python(tmp_path):
+    """The C++ reader (_geoio) and the python mmap path return
+    identical tensors/labels; read_batch assembles correctly."""
+    import pytest
+    from geomx_amd.utils.data import SyntheticImageDataset
+    from geomx_amd.utils.recordio import (RecordDataset, _geoio,
+                                          pack_dataset)
+    if _geoio is None:
+        pytest.skip("_geoio not built")
+
+    ds = SyntheticImageDataset(n=10, shape=(3, 6, 6), num_classes=3)
+    path = str(tmp_path / "n.rec")
+    pack_dataset(ds, path)
+    nat = RecordDataset(path, native=True)
+    py = RecordDataset(path, native=False)
+    assert len(nat) == len(py) == 10
+    for i in range(10):
+        xn, yn = nat[i]
+        xp, yp = py[i]
+        assert torch.equal(xn, xp) and yn == yp
+    assert torch.equal(nat.labels, py.labels)
+
+    idx = [3, 0, 7, 7, 1]
+    bn, ln = nat.read_batch(idx, threads=3)
+    bp, lp = py.read_batch(idx)
+    assert torch.equal(bn, bp) and torch.equal(ln, lp)
+    assert bn.shape == (5, 3, 6, 6)
+
+    # mixed shapes in one batch must fail loudly, not corrupt
+    from geomx_amd.utils.recordio import RecordWriter
+    p2 = str(tmp_path / "mix.rec")
+    with RecordWriter(p2) as w:
+        w.write(torch.ones(4), 0)
+        w.write(torch.ones(5), 1)
+    with pytest.raises(RuntimeError):
+        RecordDataset(p2, native=True).read_batch([0, 1])
