@@ -197,3 +197,41 @@ class RecordDataset(Dataset):
         d["_mm"] = None
         d["_nf"] = None
         return d
+
+
+class RecordBatchLoader:
+    """Iterable of (batch, labels) from a record file — the fast host
+    path for feeding one training rank from a node-local shard.
+    Uses `RecordDataset.read_batch` (native threaded assembly when
+    `_geoio` is built), so one call produces the whole [B, *shape]
+    tensor instead of B python-object round-trips. `sampler` orders the
+    epoch (e.g. SplitSampler / ClassSplitSampler for IID / non-IID
+    sharding); default is sequential."""
+
+    def __init__(self, dataset: "RecordDataset", batch_size: int,
+                 sampler=None, threads: int = 4, pin_memory: bool = False,
+                 drop_last: bool = False):
+        self.ds = dataset
+        self.bs = int(batch_size)
+        self.sampler = sampler
+        self.threads = threads
+        self.pin_memory = pin_memory
+        self.drop_last = drop_last
+
+    def _indices(self):
+        if self.sampler is not None:
+            return [int(i) for i in self.sampler]
+        return list(range(len(self.ds)))
+
+    def __len__(self):
+        n = len(self.sampler) if self.sampler is not None else len(self.ds)
+        return n // self.bs if self.drop_last else (n + self.bs - 1) // self.bs
+
+    def __iter__(self):
+        idx = self._indices()
+        for lo in range(0, len(idx), self.bs):
+            chunk = idx[lo:lo + self.bs]
+            if self.drop_last and len(chunk) < self.bs:
+                return
+            yield self.ds.read_batch(chunk, threads=self.threads,
+                                     pin_memory=self.pin_memory)

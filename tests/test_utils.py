@@ -316,3 +316,31 @@ def test_recordio_native_matches_python(tmp_path):
         w.write(torch.ones(5), 1)
     with pytest.raises(RuntimeError):
         RecordDataset(p2, native=True).read_batch([0, 1])
+
+
+def test_record_batch_loader(tmp_path):
+    from geomx_amd.utils.data import SplitSampler, SyntheticImageDataset
+    from geomx_amd.utils.recordio import (RecordBatchLoader, RecordDataset,
+                                          pack_dataset)
+
+    ds = SyntheticImageDataset(n=20, shape=(2, 4, 4), num_classes=3)
+    path = str(tmp_path / "bl.rec")
+    pack_dataset(ds, path)
+    rd = RecordDataset(path)
+    sampler = SplitSampler(len(rd), num_parts=2, part_index=1,
+                           shuffle=False)
+    dl = RecordBatchLoader(rd, batch_size=4, sampler=sampler)
+    assert len(dl) == 3  # 10 records of worker 1 in batches of 4
+    seen = 0
+    for xb, yb in dl:
+        for j in range(xb.shape[0]):
+            i = 10 + seen + j  # worker 1's contiguous shard
+            assert torch.equal(xb[j], ds.x[i])
+            assert yb[j] == ds.y[i]
+        seen += xb.shape[0]
+    assert seen == 10
+
+    # drop_last
+    dl2 = RecordBatchLoader(rd, batch_size=3, drop_last=True)
+    assert len(dl2) == 6
+    assert sum(x.shape[0] for x, _ in dl2) == 18
