@@ -256,6 +256,16 @@ class KVStoreDist(KVStoreBase):
         # kController carries user payloads; the registered handler
         # (below) is the consumer
 
+    def set_learning_rate(self, lr: float) -> None:
+        """Runtime LR update on the server optimizer (the reference
+        re-sends the pickled optimizer via kController; here it is a
+        direct call — state is preserved, unlike set_optimizer)."""
+        if self.optimizer is None:
+            raise RuntimeError("no optimizer set")
+        self.optimizer.set_learning_rate(lr)
+        if self._aps is not None and self._aps.optimizer is not None:
+            self._aps.optimizer.set_learning_rate(lr)
+
     _server_command_handler = None
     _multi_precision = False
 

@@ -43,6 +43,16 @@ class OptimizerSpec:
 class ServerOptimizer:
     """Per-key optimizer state + fused update dispatch on the owner rank."""
 
+    @property
+    def learning_rate(self) -> float:
+        return self.spec.lr
+
+    def set_learning_rate(self, lr: float) -> None:
+        """Runtime LR update (gluon Trainer.set_learning_rate /
+        Optimizer lr_scheduler parity): takes effect on the next
+        update; optimizer state is untouched."""
+        self.spec.lr = float(lr)
+
     def __init__(self, spec: OptimizerSpec):
         self.spec = spec.validate()
         self.state: Dict[object, Dict[str, torch.Tensor]] = {}

@@ -354,6 +354,14 @@ class GeoTrainer:
                 dist.broadcast(b.flat, src=topo.leader_rank,
                                group=topo.party_group)
 
+    @property
+    def learning_rate(self) -> float:
+        return self.server_opt.spec.lr
+
+    def set_learning_rate(self, lr: float) -> None:
+        """LR scheduling hook (gluon Trainer.set_learning_rate parity)."""
+        self.server_opt.set_learning_rate(lr)
+
     def zero_grad(self):
         for b in self.buckets:
             b.flat.zero_()

@@ -277,3 +277,20 @@ def test_list_key_and_multi_device_forms():
     outs = [torch.empty(5), torch.empty(5)]
     kv2.pull("w", outs)
     assert torch.allclose(outs[0], outs[1])
+
+
+def test_set_learning_rate_runtime():
+    """LR scheduling (gluon Trainer.set_learning_rate parity): takes
+    effect immediately, optimizer state preserved."""
+    kv = make_kv()
+    kv.set_optimizer(OptimizerSpec("sgd", lr=0.1))
+    kv.init("w", torch.ones(4))
+    kv.push("w", torch.ones(4))
+    out = torch.empty(4)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((4,), 0.9))
+    kv.set_learning_rate(0.5)
+    assert kv.optimizer.learning_rate == 0.5
+    kv.push("w", torch.ones(4))
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.full((4,), 0.4))
