@@ -128,6 +128,12 @@ class TSExchange:
             if extra > 0:
                 time.sleep(extra)
         dist.send(wire, dst=self.ranks[dst_party], group=self.group)
+        if wire.is_cuda:
+            # NCCL send only enqueues on the comm stream; sync so the
+            # measured interval covers the actual transfer (on the
+            # emulated WAN the token-bucket wait above dominates either
+            # way, so A reflects the configured per-party rates)
+            torch.cuda.synchronize()
         dt = max(time.perf_counter() - t0, 1e-9)
         tput = nbytes / dt
         self._my_row[dst_party] = tput
