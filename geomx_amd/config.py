@@ -117,6 +117,9 @@ class Config:
     # Overrides wan_gbps for each listed party; GEOMX_PARTY_WAN_GBPS is
     # the comma-separated env form.
     party_wan_gbps: Optional[List[float]] = None
+    # Propagation delay (round-trip, ms) added per emulated transfer —
+    # inter-DC WANs are tens of ms; deep relay overlays pay it per hop.
+    wan_rtt_ms: float = 0.0
 
     # --- runtime -------------------------------------------------------
     # P3/MultiGPS big-tensor slicing: keys with numel >= bigarray_bound
@@ -156,6 +159,7 @@ class Config:
             dgt_block_size=_env_int(["DGT_BLOCK_SIZE", "GEOMX_DGT_BLOCK_SIZE"], 4096),
             dgt_alpha=_env_float(["DGT_CONTRIBUTION_ALPHA", "GEOMX_DGT_ALPHA"], 0.3),
             wan_gbps=_env_float(["GEOMX_WAN_GBPS"], 0.0),
+            wan_rtt_ms=_env_float(["GEOMX_WAN_RTT_MS"], 0.0),
             bucket_mb=_env_int(["GEOMX_BUCKET_MB"], 25),
             bigarray_bound=_env_int(
                 ["MXNET_KVSTORE_BIGARRAY_BOUND", "GEOMX_BIGARRAY_BOUND"],

@@ -109,7 +109,8 @@ class KVStoreDist(KVStoreBase):
         self._key_order: List[object] = []
         self.optimizer: Optional[ServerOptimizer] = None
         self.compression: Optional[Dict] = None
-        self.wan = TokenBucket(cfg.wan_rate_for(self.topo.party_id))
+        self.wan = TokenBucket(cfg.wan_rate_for(self.topo.party_id),
+                               rtt_ms=cfg.wan_rtt_ms)
         self._device = self.topo.device
         self._aps = None  # true-async global tier (async_transport=store)
         # TSEngine: throughput-scheduled relay tier replaces the

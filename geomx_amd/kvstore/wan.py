@@ -29,18 +29,32 @@ import torch
 
 
 class TokenBucket:
-    """Pace traffic to `gbps` gigabit/s. Thread-unsafe by design (one per
-    kvstore; kvstore calls are already serialized per rank)."""
+    """Pace traffic to `gbps` gigabit/s, plus `rtt_ms` of propagation
+    delay per transfer. Thread-unsafe by design (one per kvstore;
+    kvstore calls are already serialized per rank).
 
-    def __init__(self, gbps: float):
+    Each charge() models ONE transfer over the party's uplink: it
+    completes after a full RTT (handshake + propagation — optimistic
+    one-RTT for a pipelined collective, and naturally one RTT PER HOP
+    for relay-tree exchanges, which is exactly the latency cost a deep
+    overlay pays on a real WAN) plus the serialization time
+    bytes/rate. rtt_ms=0 (default) keeps the bandwidth-only model."""
+
+    def __init__(self, gbps: float, rtt_ms: float = 0.0):
         self.gbps = float(gbps)
+        self.rtt_s = float(rtt_ms) / 1e3
         self._debt_until = 0.0  # monotonic time when the link is free again
         self.total_bytes = 0
         self.total_wait = 0.0
 
     @property
     def enabled(self) -> bool:
-        return self.gbps > 0
+        return self.gbps > 0 or self.rtt_s > 0
+
+    def _serialize_s(self, nbytes: float) -> float:
+        if self.gbps <= 0:
+            return 0.0
+        return nbytes * 8.0 / (self.gbps * 1e9)
 
     def charge(self, nbytes: float, sync_device: bool = True):
         """Block until the emulated link would have finished moving nbytes."""
@@ -49,11 +63,12 @@ class TokenBucket:
         if sync_device and torch.cuda.is_available():
             torch.cuda.synchronize()
         now = time.perf_counter()
+        # propagation does not occupy the link; serialization does
         start = max(now, self._debt_until)
-        duration = nbytes * 8.0 / (self.gbps * 1e9)
-        self._debt_until = start + duration
+        self._debt_until = start + self._serialize_s(nbytes)
         self.total_bytes += nbytes
-        wait = self._debt_until - now
+        done = self._debt_until + self.rtt_s
+        wait = done - now
         if wait > 0:
             self.total_wait += wait
             _sleep_precise(wait)
@@ -68,9 +83,9 @@ class TokenBucket:
             return 0.0
         now = time.perf_counter()
         start = max(now, self._debt_until)
-        self._debt_until = start + nbytes * 8.0 / (self.gbps * 1e9)
+        self._debt_until = start + self._serialize_s(nbytes)
         self.total_bytes += nbytes
-        return self._debt_until
+        return self._debt_until + self.rtt_s
 
     def wait_until(self, ready_time: float):
         if ready_time <= 0:

@@ -106,7 +106,8 @@ class GeoTrainer:
         self.cfg = cfg
         self.topo = topo
         self.mode = mode if topo.world_size > 1 else "flat"
-        self.wan = TokenBucket(cfg.wan_rate_for(topo.party_id))
+        self.wan = TokenBucket(cfg.wan_rate_for(topo.party_id),
+                               rtt_ms=cfg.wan_rtt_ms)
         self.device = topo.device
         self.spec = optimizer or OptimizerSpec(name="sgd", lr=0.01)
         self.server_opt = ServerOptimizer(self.spec)

@@ -53,3 +53,28 @@ def test_cross_party_bytes_model():
     assert cross_party_bytes("send", 100, 4) == 100.0
     with pytest.raises(ValueError):
         cross_party_bytes("nope", 1, 2)
+
+
+def test_token_bucket_rtt():
+    """RTT adds per-transfer latency on top of serialization; rtt with
+    zero bandwidth still delays (latency-only link)."""
+    import time
+
+    from geomx_amd.kvstore.wan import TokenBucket
+
+    tb = TokenBucket(gbps=1.0, rtt_ms=20.0)
+    t0 = time.perf_counter()
+    tb.charge(125_000, sync_device=False)   # 1 ms serialization + 20 ms RTT
+    dt = time.perf_counter() - t0
+    assert 0.019 <= dt < 0.08, dt
+
+    lat_only = TokenBucket(gbps=0.0, rtt_ms=10.0)
+    assert lat_only.enabled
+    t0 = time.perf_counter()
+    lat_only.charge(1_000_000, sync_device=False)
+    assert 0.009 <= time.perf_counter() - t0 < 0.05
+
+    # async reservation reports completion incl. RTT
+    tb2 = TokenBucket(gbps=1.0, rtt_ms=50.0)
+    ready = tb2.charge_async(125_000)
+    assert ready - time.perf_counter() > 0.045
