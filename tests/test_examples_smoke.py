@@ -10,7 +10,7 @@ import pytest
 HERE = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 EXAMPLES = ["cnn.py", "cnn_bsc.py", "cnn_fp16.py", "cnn_mpq.py",
-            "cnn_hfa.py", "cnn_dgt.py"]
+            "cnn_hfa.py", "cnn_dgt.py", "resnet50.py"]
 
 
 @pytest.mark.parametrize("script", EXAMPLES)
