@@ -4,6 +4,11 @@ shape-independent invariants that golden tests at fixed sizes miss."""
 import torch
 from hypothesis import given, settings, strategies as st
 
+# deterministic examples: a CI runner (the driver) must never trip on
+# a freshly-generated edge case — new examples are explored in-session
+settings.register_profile("ci", derandomize=True)
+settings.load_profile("ci")
+
 from geomx_amd.ops import reference as ref
 
 
