@@ -200,3 +200,50 @@ def _lossy_transport(rank, world):
 
 def test_async_store_lossy_transport_ws4():
     run_dist(4, _lossy_transport)
+
+
+def _dead_producer_gap(rank, world):
+    """A producer that dies with a push in flight leaves an
+    unrepairable gap: the server skips it after skip_timeout_s, counts
+    it in `lost`, and the stream continues — a lost gradient delays
+    nobody (the reference's lossy-channel stance)."""
+    kv = _mk_async()
+    n = 16
+    kv.init("w", torch.zeros(n))
+    aps = kv._aps
+    out = torch.empty(n)
+
+    kv.push("w", torch.ones(n))
+    kv.pull("w", out)
+    kv.barrier()
+
+    if aps is not None:
+        aps.skip_timeout_s = 0.5
+        if not aps.is_server:
+            # party 1's leader: this push is "lost with the producer" —
+            # 100% drop and then wipe the resend queue (process death)
+            aps.drop_pct = 1.0
+            kv.push("w", torch.ones(n))
+            aps._unacked.clear()
+            aps.drop_pct = 0.0
+    kv.barrier()
+    # a later push announces a higher seq, exposing the gap
+    kv.push("w", torch.ones(n))
+    kv.pull("w", out)
+    if aps is not None:
+        assert aps.flush(timeout_s=30)
+    kv.barrier()
+    if aps is not None and aps.is_server:
+        assert aps.drain(timeout_s=30)
+        assert aps.lost == 1, aps.lost
+        # pushes: 2 parties x (1 + 1) + party1's lost one never applied
+        assert aps.applied == 4, aps.applied
+    kv.barrier()
+    kv.pull("w", out)
+    # 4 applied pushes, each a single-worker party sum of ones
+    assert torch.allclose(out, torch.full((n,), 4.0)), out[0]
+    kv.close()
+
+
+def test_async_store_dead_producer_gap_ws2():
+    run_dist(2, _dead_producer_gap)
