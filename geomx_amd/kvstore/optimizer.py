@@ -32,6 +32,10 @@ class OptimizerSpec:
     lamda: float = 0.04          # DCASGD delay-compensation scale
     rho: float = 0.9             # RMSProp decay
     rescale_grad: float = 1.0
+    clip_gradient: Optional[float] = None  # clamp |g| per element AFTER
+                                           # rescale (optimizer.py:912 —
+                                           # every reference optimizer
+                                           # clips this way)
 
     def validate(self):
         if self.name not in ("sgd", "sgd_mom", "adam", "dcasgd", "rmsprop",
@@ -84,6 +88,12 @@ class ServerOptimizer:
         """Apply one fused update of `w` (fp32, flat) with aggregated grad."""
         s = self.spec
         rs = s.rescale_grad if rescale is None else rescale
+        if s.clip_gradient is not None:
+            # reference clips the RESCALED gradient, so fold the scale
+            # in here and hand the kernels rs=1
+            grad = torch.clamp(grad * rs, -s.clip_gradient,
+                               s.clip_gradient)
+            rs = 1.0
         st = self._get_state(key, w)
         self.step_count[key] += 1
         t = self.step_count[key]

@@ -294,3 +294,27 @@ def test_set_learning_rate_runtime():
     kv.push("w", torch.ones(4))
     kv.pull("w", out)
     assert torch.allclose(out, torch.full((4,), 0.4))
+
+
+def test_clip_gradient():
+    """clip_gradient clamps the rescaled grad per element before the
+    update (reference optimizer.py:912)."""
+    kv = make_kv()
+    kv.set_optimizer(OptimizerSpec("sgd", lr=1.0, clip_gradient=0.5,
+                                   rescale_grad=1.0))
+    kv.init("w", torch.zeros(4))
+    kv.push("w", torch.tensor([10.0, -10.0, 0.2, -0.2]))
+    out = torch.empty(4)
+    kv.pull("w", out)
+    assert torch.allclose(out, torch.tensor([-0.5, 0.5, -0.2, 0.2]))
+
+    # clip applies AFTER rescale
+    kv2 = make_kv()
+    kv2.set_optimizer(OptimizerSpec("sgd", lr=1.0, clip_gradient=0.5))
+    kv2.init("w", torch.zeros(2))
+    kv2.push("w", torch.tensor([100.0, 0.01]))
+    kv2.optimizer.update("w2", torch.zeros(2),
+                         torch.tensor([100.0, 0.01]), rescale=0.001)
+    out2 = torch.empty(2)
+    kv2.pull("w", out2)
+    assert torch.allclose(out2, torch.tensor([-0.5, -0.01]))
