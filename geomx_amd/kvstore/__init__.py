@@ -24,6 +24,13 @@ def create(name: str = "dist_sync", cfg: Optional[Config] = None,
     """Create a kvstore.
 
     name: "dist_sync" (FSA), "dist_async" (MixedSync), or "local".
+    The reference resolves types by SUBSTRING (kvstore.cc:41-80:
+    "dist" selects the distributed store, "_sync" within it selects
+    synchronous mode, "device"/"nccl" pick intra-node comm variants) —
+    so names like "dist_device_sync", "device", "nccl" are accepted
+    here the same way. Device/NCCL distinctions collapse on MI355X:
+    RCCL over xGMI IS the (only) intra-node fabric.
+
     Extra keyword arguments override Config fields (which themselves
     default from the GeoMX-compatible environment variables).
     """
@@ -34,6 +41,12 @@ def create(name: str = "dist_sync", cfg: Optional[Config] = None,
         # ENABLE_INTER_TS/ENABLE_INTRA_TS select the incast-free
         # replicated tier (the TSEngine role)
         global_mode = "replicated" if cfg.enable_ts else "sharded"
-    cfg.mode = name if name != "local" else "dist_sync"
+    t = str(name).lower()
+    if "dist" in t:
+        cfg.mode = "dist_sync" if "_sync" in t else "dist_async"
+    else:  # local / device / nccl: single-node synchronous store
+        cfg.mode = "dist_sync"
     cfg.validate()
-    return KVStoreDist(cfg, topo=topo, global_mode=global_mode)
+    kv = KVStoreDist(cfg, topo=topo, global_mode=global_mode)
+    kv._type_name = t
+    return kv

@@ -129,7 +129,9 @@ class KVStoreDist(KVStoreBase):
     # ------------------------------------------------------------------
     @property
     def type(self) -> str:
-        return self.cfg.mode
+        # the reference reports the ORIGINAL type string (kvstore.cc:80
+        # kv->type_ = tname), e.g. "dist_device_sync"
+        return getattr(self, "_type_name", None) or self.cfg.mode
 
     @property
     def rank(self) -> int:

@@ -318,3 +318,18 @@ def test_clip_gradient():
     out2 = torch.empty(2)
     kv2.pull("w", out2)
     assert torch.allclose(out2, torch.tensor([-0.5, -0.01]))
+
+
+def test_type_string_substring_dispatch():
+    """Type names resolve by substring like the reference
+    (kvstore.cc:41-80), and kv.type echoes the original string."""
+    for name, mode in [("dist_sync", "dist_sync"),
+                       ("dist_device_sync", "dist_sync"),
+                       ("dist_async", "dist_async"),
+                       ("dist_device_async", "dist_async"),
+                       ("local", "dist_sync"),
+                       ("device", "dist_sync"),
+                       ("nccl", "dist_sync")]:
+        kv = create(name, cfg=Config.from_env())
+        assert kv.type == name
+        assert kv.cfg.mode == mode, (name, kv.cfg.mode)
