@@ -617,7 +617,10 @@ class KVStoreDist(KVStoreBase):
     # ------------------------------------------------------------------
     # pull
     # ------------------------------------------------------------------
-    def pull(self, key, out, priority: int = 0) -> None:
+    def pull(self, key, out, priority: int = 0,
+             ignore_sparse: bool = True) -> None:
+        # ignore_sparse accepted for signature parity (kvstore.py:242);
+        # row-sparse values live behind row_sparse_pull here.
         # list-of-keys form (kvstore.py:242); a per-key list of outs
         # broadcasts the same value into each (multi-device pull)
         if isinstance(key, (list, tuple)):
