@@ -211,21 +211,40 @@ class GeoTrainer:
                                      async_op=True)
 
     # ------------------------------------------------------------------
-    def step(self, lr: Optional[float] = None):
+    def step(self, lr: Optional[float] = None,
+             batch_size: Optional[int] = None):
         """Finish outstanding communication, run the WAN tier (hips), and
-        apply the fused optimizer update. Call after loss.backward()."""
+        apply the fused optimizer update. Call after loss.backward().
+
+        `batch_size` normalizes the reduced gradient by 1/batch_size at
+        update time (gluon Trainer.step semantics, trainer.py:258 —
+        pass it instead of dividing the loss). For gradient surgery
+        between the reduce and the update (e.g. custom clipping), call
+        `allreduce_grads()` and `update()` separately — same contract
+        as the reference's split; not available in the pipelined
+        dist_async mode, where the stale update happens inside the
+        exchange."""
         self._step += 1
-        if lr is not None:
-            self.spec.lr = lr
+        if self.mode == "hips" and self.topo.num_parties > 1 \
+                and self.cfg.mode == "dist_async" \
+                and self.cfg.compression in (None, "fp16"):
+            self._wait_buckets()
+            self._wan_tier_async()
+            return
+        self.allreduce_grads()
+        self.update(batch_size=batch_size, lr=lr)
+
+    def _wait_buckets(self):
         for b in self.buckets:
             if b.work is not None:
                 b.work.wait()
                 b.work = None
-        if self.mode == "hips" and self.topo.num_parties > 1 \
-                and self.cfg.mode == "dist_async" \
-                and self.cfg.compression in (None, "fp16"):
-            self._wan_tier_async()
-            return
+
+    def allreduce_grads(self):
+        """Finish the bucket collectives and the (synchronous) WAN tier;
+        after this, `p.grad` views hold the fully-reduced gradients
+        (gluon Trainer.allreduce_grads, trainer.py:293)."""
+        self._wait_buckets()
         if self.mode == "hips" and self.topo.num_parties > 1:
             self._wan_tier()
         elif self.mode == "flat" and self.wan.enabled \
@@ -237,8 +256,18 @@ class GeoTrainer:
             for b in self.buckets:
                 self.wan.charge(cross_party_bytes(
                     "all_reduce", b.flat.numel() * 4, self.cfg.num_parties))
+
+    def update(self, batch_size: Optional[int] = None,
+               lr: Optional[float] = None):
+        """Apply the fused optimizer to the reduced gradients
+        (gluon Trainer.update, trainer.py:326)."""
+        if lr is not None:
+            self.spec.lr = lr
+        rs = None if batch_size is None \
+            else self.spec.rescale_grad / batch_size
         for b in self.buckets:
-            self.server_opt.update(("bucket", b.index), b.param_flat, b.flat)
+            self.server_opt.update(("bucket", b.index), b.param_flat,
+                                   b.flat, rescale=rs)
             b.ready = 0
 
     # -- pipelined WAN tier (dist_async / MixedSync realized) -----------

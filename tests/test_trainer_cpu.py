@@ -423,3 +423,35 @@ def _trainer_hips_ts_fp16(rank, world):
 
 def test_hips_tsengine_fp16_consistent_ws4():
     run_dist(4, _trainer_hips_ts_fp16)
+
+
+def test_step_batch_size_and_split():
+    """gluon Trainer parity: step(batch_size=B) rescales by 1/B;
+    allreduce_grads()/update() split allows gradient surgery between
+    reduce and update."""
+    cfg = Config.from_env(backend="gloo", device="cpu", bucket_mb=1)
+    topo = init_topology(1, None, "gloo", "cpu")
+    model = torch.nn.Linear(4, 2, bias=False)
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec("sgd", lr=1.0))
+    x = torch.ones(8, 4)
+    w0 = model.weight.detach().clone()
+
+    loss = model(x).sum()
+    tr.zero_grad()
+    loss.backward()
+    g = model.weight.grad.detach().clone()
+    tr.step(batch_size=8)
+    assert torch.allclose(model.weight.detach(), w0 - g / 8, atol=1e-6)
+
+    # split form with surgery: clamp the reduced grad before update
+    w1 = model.weight.detach().clone()
+    loss = model(x).sum()
+    tr.zero_grad()
+    loss.backward()
+    tr.allreduce_grads()
+    with torch.no_grad():
+        model.weight.grad.clamp_(-0.1, 0.1)
+    tr.update()
+    assert torch.allclose(model.weight.detach(),
+                          w1 - model.weight.grad, atol=1e-6)
+    assert model.weight.grad.abs().max() <= 0.1
