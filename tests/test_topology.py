@@ -41,3 +41,43 @@ def test_party_of():
     assert t.is_leader
     assert t.leader_index == 2
     assert not t.is_master_worker
+
+
+def test_config_validation_errors():
+    import pytest
+
+    from geomx_amd import Config
+
+    with pytest.raises(ValueError):
+        Config.from_env(mode="nope").validate()
+    with pytest.raises(ValueError):
+        Config.from_env(compression="zip").validate()
+    with pytest.raises(ValueError):
+        Config.from_env(bsc_ratio=1.5).validate()
+    with pytest.raises(ValueError):
+        Config.from_env(num_parties=0).validate()
+    with pytest.raises(ValueError):
+        Config.from_env(hfa_k2=0).validate()
+    with pytest.raises(ValueError):
+        Config.from_env(async_transport="udp").validate()
+
+
+def test_kvstore_invalid_global_mode():
+    import pytest
+
+    from geomx_amd import Config
+    from geomx_amd.kvstore.dist import KVStoreDist
+
+    with pytest.raises(ValueError):
+        KVStoreDist(Config.from_env(), global_mode="ring")
+
+
+def test_explicit_sharded_overrides_ts():
+    """An explicit global_mode wins over the ENABLE_TS default; the
+    relay tier only attaches to the replicated tier."""
+    from geomx_amd import Config
+    from geomx_amd.kvstore import create
+
+    kv = create("dist_sync", cfg=Config.from_env(enable_ts=True),
+                global_mode="sharded")
+    assert kv.global_mode == "sharded" and kv._ts is None
