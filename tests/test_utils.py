@@ -344,3 +344,14 @@ def test_record_batch_loader(tmp_path):
     dl2 = RecordBatchLoader(rd, batch_size=3, drop_last=True)
     assert len(dl2) == 6
     assert sum(x.shape[0] for x, _ in dl2) == 18
+
+
+def test_recordio_empty_file(tmp_path):
+    from geomx_amd.utils.recordio import RecordDataset, RecordWriter
+
+    path = str(tmp_path / "e.rec")
+    with RecordWriter(path):
+        pass
+    for native in (None, False):
+        rd = RecordDataset(path, native=native)
+        assert len(rd) == 0
