@@ -83,3 +83,23 @@ def test_resnet50_smoke():
         loss = y.float().square().mean()
     loss.backward()
     assert torch.isfinite(loss).item()
+
+
+def test_recordio_pinned_batch_to_gpu(tmp_path):
+    """Native record reader feeding the GPU: threaded batch assembly
+    into pinned host memory, then H2D — the intended data path on a
+    training rank."""
+    from geomx_amd.utils.data import SyntheticImageDataset
+    from geomx_amd.utils.recordio import RecordDataset, _geoio, pack_dataset
+
+    assert _geoio is not None, "_geoio extension not built"
+    ds = SyntheticImageDataset(n=32, shape=(3, 16, 16), num_classes=4)
+    path = str(tmp_path / "g.rec")
+    pack_dataset(ds, path)
+    rd = RecordDataset(path, native=True)
+    xb, yb = rd.read_batch(range(16), threads=4, pin_memory=True)
+    assert xb.is_pinned()
+    dev = xb.to("cuda:0", non_blocking=True)
+    torch.cuda.synchronize()
+    assert torch.equal(dev.cpu(), ds.x[:16])
+    assert torch.equal(yb, ds.y[:16])
