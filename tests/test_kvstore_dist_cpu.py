@@ -386,3 +386,23 @@ def _2bit_two_tier(rank, world):
 
 def test_2bit_two_tier_ws4():
     run_dist(4, _2bit_two_tier)
+
+
+# ---------------------------------------------------------------------------
+# list-form API across ranks
+# ---------------------------------------------------------------------------
+
+def _list_forms(rank, world):
+    kv = _mk(num_parties=2)
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    kv.init([0, 1], [torch.ones(4), torch.zeros(6)])
+    kv.push([0, 1], [torch.ones(4), torch.full((6,), 0.5)])
+    o0, o1 = torch.empty(4), torch.empty(6)
+    kv.pull([0, 1], [o0, o1])
+    # 4 workers: w0 = 1 - 0.1*4*1, w1 = 0 - 0.1*4*0.5
+    assert torch.allclose(o0, torch.full((4,), 0.6), atol=1e-6), (rank, o0)
+    assert torch.allclose(o1, torch.full((6,), -0.2), atol=1e-6), (rank, o1)
+
+
+def test_list_forms_ws4():
+    run_dist(4, _list_forms)
