@@ -236,6 +236,13 @@ class TSExchange:
         self._sync_rows(acc.device)
         return root, acc.reshape(x.shape)
 
+    def stats(self) -> dict:
+        """Observability: the learned throughput matrix (bytes/s, -1 =
+        unmeasured) and exchange count — the data a scheduler operator
+        would watch on a real WAN."""
+        return {"A_bytes_per_s": [row[:] for row in self.sched.A],
+                "exchanges": self._seq}
+
     def allreduce_sum(self, x: torch.Tensor,
                       wire_dtype: torch.dtype = torch.float32
                       ) -> torch.Tensor:

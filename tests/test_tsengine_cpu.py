@@ -306,3 +306,17 @@ def _kv_ts_hetero_wan(rank, world):
 
 def test_kv_tsengine_heterogeneous_wan_ws4():
     run_dist(4, _kv_ts_hetero_wan)
+
+
+def _stats_surface(rank, world):
+    ts = _mk_ts(rank, world)
+    ts.allreduce_sum(torch.randn(64))
+    st = ts.stats()
+    assert st["exchanges"] == 2  # merge + spread
+    assert len(st["A_bytes_per_s"]) == world
+    measured = [v for row in st["A_bytes_per_s"] for v in row if v >= 0]
+    assert measured  # at least the hops of one allreduce
+
+
+def test_ts_stats_ws3():
+    run_dist(3, _stats_surface)
