@@ -96,7 +96,10 @@ _WRW_NWG = 768  # partial slabs (3 workgroups per CU)
 # measures SLOWER than MIOpen's igemm_wrw (conv2: 2.07 vs 0.67 ms;
 # conv1: 2.41 vs 1.64 ms) — LDS gather-bound. Off by default until the
 # A-fragment gather is restructured.
-WRW_ENABLED = False
+# opt-in via env for A/B runs (GEOPS_WRW=1); module default stays off
+# because v1 loses net in-step (see docs/kernels.md wrw v2 notes)
+import os as _os
+WRW_ENABLED = _os.environ.get("GEOPS_WRW", "0") == "1"
 
 
 def build_wrw_unpack_index(weight_shape) -> torch.Tensor:
@@ -304,7 +307,9 @@ class GeoConv5(torch.nn.Conv2d):
     # wrw wins standalone (1.10 vs 1.62 ms) but the split costs more
     # elsewhere in the full step (bench 69.9k vs 74.4k samples/s) —
     # see ROADMAP item 1.
-    SPLIT_BACKWARD = False
+    # opt-in via env for A/B runs (GEOPS_SPLIT_BWD=1)
+    SPLIT_BACKWARD = __import__("os").environ.get(
+        "GEOPS_SPLIT_BWD", "0") == "1"
 
     def forward(self, x):
         if not self._eligible(x):
