@@ -2,10 +2,11 @@
 # One-call GPU validation sweep — run on a GPU box (e.g. under gpurun):
 #   gpurun --timeout 1500 -- 'bash scripts/gpu_validation.sh'
 # Covers: the gpu test suite, the flagship bench, a kernel bandwidth
-# table, and 2-process distributed smokes (two ranks sharing the one
-# GPU over gloo: flat, HiPS+BSC under a WAN cap, and the TSEngine
-# relay) so every distributed code path executes on real hardware
-# before the multi-GPU scaling run.
+# table, and 2-process distributed smokes (gloo keeps tensors on CPU —
+# these validate the DISTRIBUTED LOGIC of flat, HiPS+BSC under a WAN
+# cap, and the TSEngine relay on the box; GPU compute itself is
+# covered by the gpu suite and the 1-proc bench above) before the
+# driver's multi-GPU RCCL scaling run.
 set -euo pipefail
 cd "$(dirname "$0")/.."
 mkdir -p gpurun_out
