@@ -1,1 +1,1 @@
-from . import checkpoint, data, health, metrics, profiler  # noqa: F401
+from . import checkpoint, data, health, metrics, profiler, recordio  # noqa: F401
