@@ -455,3 +455,33 @@ def test_step_batch_size_and_split():
     assert torch.allclose(model.weight.detach(),
                           w1 - model.weight.grad, atol=1e-6)
     assert model.weight.grad.abs().max() <= 0.1
+
+
+def _async_hetero_wan(rank, world):
+    """Pipelined async WAN tier composed with heterogeneous per-party
+    uplinks: steps proceed, reservations complete, replicas stay
+    finite and party-consistent."""
+    cfg = Config.from_env(num_parties=2, backend="gloo", device="cpu",
+                          bucket_mb=1, mode="dist_async",
+                          party_wan_gbps=[5.0, 1.0])
+    topo = init_topology(2, None, "gloo", "cpu")
+    model = _tiny_model()
+    tr = GeoTrainer(model, cfg, topo, OptimizerSpec(name="sgd", lr=0.05),
+                    mode="hips")
+    assert tr.wan.gbps == (5.0 if topo.party_id == 0 else 1.0)
+    for s in range(4):
+        x, y = _make_data(seed=50 + s)
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        tr.zero_grad()
+        loss.backward()
+        tr.step()
+    import torch.distributed as dist
+    for p in model.parameters():
+        assert torch.isfinite(p).all()
+        ref = p.data.clone()
+        dist.broadcast(ref, src=topo.leader_rank, group=topo.party_group)
+        assert torch.allclose(p.data, ref)
+
+
+def test_async_heterogeneous_wan_ws4():
+    run_dist(4, _async_hetero_wan)
