@@ -42,6 +42,93 @@ class DGTState:
         self.nchunks = (numel + chunk_elems - 1) // chunk_elems
         self.contrib = None
         self.residual = torch.zeros(numel, device=device)
+        self.n_keep = max(1, int(math.ceil(self.k * self.nchunks)))
+
+    # ------------------------------------------------------------------
+    # wire form (mode 3): what actually crosses the leader tier.
+    # The reference ships exact bytes for important chunks and 4-bit
+    # codes + a per-chunk codebook for the rest (van.cc:750-824); here
+    # the payload is a fixed-shape tuple so RCCL all_gather can carry it
+    # without variable-length framing:
+    #   exact    f32 [n_keep*chunk]   important chunks, full precision
+    #   keep_idx i32 [n_keep]         which chunks are exact (sorted)
+    #   packed   u8  [n_lossy*chunk/2] 4-bit codes of the rest
+    #   minmax   f32 [n_lossy, 2]     per-chunk codebook (lo, hi)
+    # Zero-contribution chunks reconstruct to ~0 (minmax zeroed), the
+    # deterministic analog of the reference dropping them from the send
+    # (kv_app.h:973, receiver zero-fill van.cc:356-366).
+    # ------------------------------------------------------------------
+    def _update_contrib(self, flat: torch.Tensor) -> None:
+        contrib = ops.dgt_contribution(flat, self.chunk)
+        if self.contrib is None or self.contrib.numel() != contrib.numel():
+            self.contrib = contrib
+        else:
+            self.contrib = (self.alpha * self.contrib
+                            + (1 - self.alpha) * contrib)
+
+    def wire_bytes(self) -> int:
+        """Real bytes per payload (mode 3)."""
+        n_lossy = self.nchunks - self.n_keep
+        return (self.n_keep * self.chunk * 4 + self.n_keep * 4
+                + n_lossy * (self.chunk // 2 + 8))
+
+    def compress(self, x: torch.Tensor):
+        """Mode-3 wire payload. Same EWMA/residual state machine as
+        transform(); shapes are identical on every rank for the same
+        (numel, chunk_elems, k), so the tuple all_gathers directly."""
+        assert self.mode >= 3, "wire payload exists only for mode 3"
+        flat = x.reshape(-1)
+        self._update_contrib(flat)
+        keep_idx = torch.topk(self.contrib, self.n_keep).indices \
+            .sort().values
+        dead = self.contrib == 0
+        pad = self.nchunks * self.chunk
+        if pad != self.numel:
+            fpad = torch.zeros(pad, device=flat.device)
+            fpad[:self.numel] = flat
+            rpad = torch.zeros(pad, device=flat.device)
+            rpad[:self.numel] = self.residual
+        else:
+            fpad, rpad = flat, self.residual
+        packed, minmax = ops.quantize_4bit_chunked(fpad, self.chunk, rpad)
+        minmax = minmax.to(flat.device)
+        if pad != self.numel:
+            self.residual.copy_(rpad[:self.numel])
+        # residual is only meaningful for quantized chunks
+        keep_mask = torch.zeros(self.nchunks, dtype=torch.bool,
+                                device=flat.device)
+        keep_mask[keep_idx] = True
+        elem_keep = keep_mask.repeat_interleave(self.chunk)[:self.numel]
+        self.residual[elem_keep] = 0.0
+
+        lossy_idx = (~keep_mask).nonzero(as_tuple=True)[0]
+        exact = fpad.view(self.nchunks, self.chunk)[keep_idx].clone()
+        exact[dead[keep_idx]] = 0.0
+        pk = packed.view(self.nchunks, self.chunk // 2)[lossy_idx] \
+            .reshape(-1).contiguous()
+        mm = minmax.view(self.nchunks, 2)[lossy_idx].clone()
+        mm[dead[lossy_idx]] = 0.0
+        return (exact.reshape(-1), keep_idx.to(torch.int32), pk, mm)
+
+    def decompress(self, exact: torch.Tensor, keep_idx: torch.Tensor,
+                   packed: torch.Tensor, minmax: torch.Tensor) -> torch.Tensor:
+        """Rebuild a dense tensor from a wire payload (any rank's)."""
+        dev = exact.device
+        n_keep = keep_idx.numel()
+        n_lossy = self.nchunks - n_keep
+        keep_mask = torch.zeros(self.nchunks, dtype=torch.bool, device=dev)
+        keep_mask[keep_idx.long()] = True
+        lossy_idx = (~keep_mask).nonzero(as_tuple=True)[0]
+        out = torch.zeros(self.nchunks, self.chunk, device=dev)
+        if n_lossy:
+            deq = ops.dequantize_4bit_chunked(
+                packed, minmax.view(n_lossy, 2), n_lossy * self.chunk,
+                self.chunk)
+            out[lossy_idx] = deq.to(dev).view(n_lossy, self.chunk)
+            # minmax == (0,0) marks an untransmitted (dead) chunk: the
+            # 1e-30 span dequantizes to ~3e-32, i.e. exact zero fill
+        out[keep_idx.long()] = exact.view(n_keep, self.chunk)
+        return out.reshape(-1)[:self.numel]
 
     def transform(self, x: torch.Tensor) -> Tuple[torch.Tensor, int]:
         """Return (lossy reconstruction, wire bytes). Important chunks

@@ -194,6 +194,10 @@ class KVStoreDist(KVStoreBase):
         if not callable(updater):
             raise TypeError("updater must be callable")
         self.optimizer = _UpdaterAdapter(updater)
+        if self._aps is not None:
+            # the async store-transport server applies updates itself;
+            # give it the same adapter (it only calls .update())
+            self._aps.optimizer = self.optimizer
 
     # CommandType, kvstore_dist_server.h:49-52 (the C++ enum; the
     # reference's Python mirror kvstore.py:89-97 predates the inserted
@@ -282,6 +286,17 @@ class KVStoreDist(KVStoreBase):
             raise ValueError(f"unknown compression type {ctype!r}")
         if ctype == "2bit":
             params.setdefault("threshold", self.cfg.threshold)
+            # untested/meaningless compositions are rejected loudly:
+            # HFA exchanges PARAMETER deltas (not gradients) and the
+            # store-transport async tier applies pushes on arrival —
+            # neither carries a 2bit error-feedback residual contract.
+            if self.cfg.use_hfa:
+                raise ValueError("2bit compression is not supported with "
+                                 "HFA (parameter-delta exchange)")
+            if self.cfg.mode == "dist_async" \
+                    and self.cfg.async_transport == "store":
+                raise ValueError("2bit compression is not supported with "
+                                 "async_transport='store'")
         if ctype in ("bsc", "mpq"):
             params.setdefault("threshold", self.cfg.bsc_ratio)
             params.setdefault("size_lower_bound", self.cfg.size_lower_bound)
@@ -300,7 +315,11 @@ class KVStoreDist(KVStoreBase):
             store = dist.distributed_c10d._get_default_store()
             self._aps = AsyncPSGlobal(store, self.topo, self._device,
                                       wan=self.wan)
-            if self.optimizer is not None:
+            if isinstance(self.optimizer, _UpdaterAdapter):
+                # custom python updater: share the adapter (it is
+                # stateless apart from the user's closure)
+                self._aps.optimizer = self.optimizer
+            elif self.optimizer is not None:
                 self._aps.optimizer = ServerOptimizer(self.optimizer.spec)
         return self._aps
 
@@ -328,9 +347,13 @@ class KVStoreDist(KVStoreBase):
                 and st.numel >= self.cfg.bigarray_bound:
             # P3/MultiGPS: slice the key uniformly across ALL leaders
             # (EncodeP3Key kvstore_dist.h:763-799; server-side sharding
-            # kvstore_dist_server.h:1770-1810)
+            # kvstore_dist_server.h:1770-1810). Chunks are rounded to a
+            # multiple of 64 elements so every leader's slice base stays
+            # 256B-aligned — the fused-optimizer kernels require 16-byte
+            # alignment (csrc/geops.cpp check_f32).
             st.sliced = True
-            st.padded = ((st.numel + P - 1) // P) * P
+            align = 64 * P
+            st.padded = ((st.numel + align - 1) // align) * align
         self.keys[key] = st
         self._key_order.append(key)
         flat = value.detach().reshape(-1).float().to(self._device)
@@ -522,11 +545,26 @@ class KVStoreDist(KVStoreBase):
             return [x.float() for x in hlist]
 
         if ctype == "dgt":
-            contrib_now = self._dgt_transform(st, party_sum)
+            dg = self._dgt_state(st, party_sum.device)
+            if dg.mode >= 3:
+                # real reduced wire: all_gather the 4-bit payload tuple
+                # and decompress each party's contribution on arrival
+                # (van.cc:750-824 end-to-end, not an emulation charge)
+                payload = dg.compress(party_sum)
+                gathered = [[torch.empty_like(t) for _ in range(P)]
+                            for t in payload]
+                for lst, t in zip(gathered, payload):
+                    dist.all_gather(lst, t, group=group)
+                self.wan.charge(cross_party_bytes(
+                    "all_gather", dg.wire_bytes(), P))
+                return [dg.decompress(*(lst[p] for lst in gathered))
+                        for p in range(P)]
+            # modes 1/2: unimportant chunks travel exact (priority-only
+            # semantics; no byte saving exists on a reliable fabric)
+            contrib_now, wire = dg.transform(party_sum)
             hlist = [torch.empty_like(contrib_now) for _ in range(P)]
             dist.all_gather(hlist, contrib_now, group=group)
-            nbytes = self._dgt_wire_bytes(st)
-            self.wan.charge(cross_party_bytes("all_gather", nbytes, P))
+            self.wan.charge(cross_party_bytes("all_gather", wire, P))
             return list(hlist)
 
         if st.sliced and self.cfg.mode == "dist_sync":
@@ -744,12 +782,15 @@ class KVStoreDist(KVStoreBase):
     # state blob — gluon/block.py:315,356 + kvstore.py:566-592)
     # ------------------------------------------------------------------
     def save_optimizer_states(self, fname: str, dump_optimizer: bool = False):
+        """Serialize optimizer STATES; with dump_optimizer=True also the
+        optimizer itself (the spec), matching the reference where the
+        pickled optimizer rides along only on request
+        (python/mxnet/kvstore.py:566-592)."""
         if self.optimizer is None:
             raise RuntimeError("no optimizer set")
         blob = self.optimizer.state_dict()
         if not dump_optimizer:
-            blob = {k: v for k, v in blob.items() if k != "spec"} | \
-                {"spec": blob["spec"]}
+            blob = {k: v for k, v in blob.items() if k != "spec"}
         with open(fname, "wb") as f:
             pickle.dump(blob, f)
 
@@ -757,27 +798,26 @@ class KVStoreDist(KVStoreBase):
         with open(fname, "rb") as f:
             blob = pickle.load(f)
         if self.optimizer is None:
+            if "spec" not in blob:
+                raise RuntimeError(
+                    "states-only blob (saved with dump_optimizer=False) "
+                    "needs set_optimizer() before load_optimizer_states")
             self.optimizer = ServerOptimizer(OptimizerSpec(**blob["spec"]))
         self.optimizer.load_state_dict(blob, device=self._device)
 
     # ------------------------------------------------------------------
     # DGT transform (shared DGTState, kvstore/dgt.py)
     # ------------------------------------------------------------------
-    def _dgt_transform(self, st: _KeyState, x: torch.Tensor) -> torch.Tensor:
+    def _dgt_state(self, st: _KeyState, device):
         if not hasattr(self, "_dgt_states"):
             self._dgt_states = {}
         key = id(st)
         dg = self._dgt_states.get(key)
         if dg is None or dg.numel != st.numel:
             from .dgt import DGTState
-            dg = DGTState(st.numel, x.device,
+            dg = DGTState(st.numel, device,
                           chunk_elems=max(64, self.cfg.dgt_block_size // 4),
                           k=self.cfg.dgt_k, alpha=self.cfg.dgt_alpha,
                           mode=self.cfg.enable_dgt or 3)
             self._dgt_states[key] = dg
-        out, wire = dg.transform(x)
-        self._dgt_wire = wire
-        return out
-
-    def _dgt_wire_bytes(self, st: _KeyState) -> int:
-        return self._dgt_wire
+        return dg

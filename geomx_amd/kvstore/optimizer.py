@@ -127,7 +127,8 @@ class ServerOptimizer:
         }
 
     def load_state_dict(self, d: Dict, device=None):
-        self.spec = OptimizerSpec(**d["spec"]).validate()
+        if "spec" in d:  # absent when saved with dump_optimizer=False
+            self.spec = OptimizerSpec(**d["spec"]).validate()
         self.step_count = dict(d["step_count"])
         self.state = {}
         for k, st in d["state"].items():

