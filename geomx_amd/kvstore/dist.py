@@ -154,6 +154,7 @@ class KVStoreDist(KVStoreBase):
         return self.topo.num_parties
 
     def barrier(self):
+        self.flush()
         if dist.is_initialized():
             dist.barrier()
 
@@ -182,6 +183,7 @@ class KVStoreDist(KVStoreBase):
             spec = OptimizerSpec(**optimizer)
         else:
             raise TypeError("set_optimizer expects OptimizerSpec or dict")
+        self.flush()  # pending pushes apply under the OLD optimizer
         # every rank creates it; only leaders/owners apply it (state is lazy)
         self.optimizer = ServerOptimizer(spec)
         if self._aps is not None:
@@ -216,6 +218,7 @@ class KVStoreDist(KVStoreBase):
         SPMD transport: rank 0 broadcasts; leader ranks (the 'servers')
         interpret built-in CommandType codes, then any registered
         handler sees the command too (kController user payloads)."""
+        self.flush()
         obj = [int(head), str(body)]
         if dist.is_initialized() and self.topo.world_size > 1:
             dist.broadcast_object_list(obj, src=0)
@@ -280,6 +283,7 @@ class KVStoreDist(KVStoreBase):
         self._server_command_handler = fn
 
     def set_gradient_compression(self, compression_params: Dict) -> None:
+        self.flush()  # pending pushes complete under the OLD scheme
         params = dict(compression_params)
         ctype = params.get("type")
         if ctype not in ("2bit", "bsc", "fp16", "mpq", "dgt"):
@@ -324,6 +328,7 @@ class KVStoreDist(KVStoreBase):
         return self._aps
 
     def close(self):
+        self.flush()
         if self._aps is not None:
             self._aps.flush(timeout_s=10)  # deliver any un-ACKed pushes
             self._aps.stop()
@@ -382,7 +387,17 @@ class KVStoreDist(KVStoreBase):
         return st
 
     # ------------------------------------------------------------------
-    # push
+    # push — ASYNC with priority (engine-var semantics of
+    # kvstore_dist.h:553-562 as stream events + a deferred WAN tier).
+    #
+    # push() launches the party-tier collective with async_op=True (RCCL
+    # runs it on its comm stream, ordered after the producing compute,
+    # so it overlaps the rest of backward — the role of the reference's
+    # priority engine pools, threaded_engine_perdevice.cc:82-124) and
+    # defers the leader/WAN tier. The first pull (or any state-reading
+    # API) flushes every pending key in DESCENDING priority order
+    # (ps-lite pops the max meta.priority first, threadsafe_queue.h:50)
+    # — deterministic across ranks, so the deferred collectives match.
     # ------------------------------------------------------------------
     def push(self, key, value, priority: int = 0) -> None:
         # list-of-keys form (kvstore.py:162)
@@ -401,11 +416,46 @@ class KVStoreDist(KVStoreBase):
         st = self._state(key)
         if value.numel() != st.numel:
             raise ValueError(f"push size mismatch for {key!r}")
-        topo = self.topo
+        if key in self._pending:
+            # re-push before pull: complete the previous round first
+            self.flush()
         st.push_count += 1
+        pend = self._party_launch(st, value)
+        if self._use_aps():
+            # the store-transport tier is ALREADY asynchronous end-to-end
+            # (leader hands off and returns); deferring would only delay
+            # the hand-off
+            self._post_aggregate(key, st, self._party_finish(st, pend))
+            return
+        self._pending[key] = (priority, self._push_seq, key, pend)
+        self._push_seq += 1
 
-        party_sum = self._party_aggregate(st, value)
+    _push_seq = 0
 
+    @property
+    def _pending(self):
+        p = getattr(self, "_pending_map", None)
+        if p is None:
+            p = self._pending_map = {}
+        return p
+
+    def flush(self) -> None:
+        """Complete every pending push: wait the party-tier collectives
+        and run the deferred WAN tier + server update, highest priority
+        first (P3 ordering)."""
+        if not self._pending:
+            return
+        entries = sorted(self._pending.values(),
+                         key=lambda e: (-e[0], e[1]))
+        self._pending_map = {}
+        for _prio, _seq, key, pend in entries:
+            st = self.keys[key]
+            party_sum = self._party_finish(st, pend)
+            self._post_aggregate(key, st, party_sum)
+
+    def _post_aggregate(self, key, st: _KeyState,
+                        party_sum: torch.Tensor) -> None:
+        topo = self.topo
         if topo.num_parties == 1:
             if topo.is_leader:
                 self._apply_global(key, st, [party_sum])
@@ -429,13 +479,14 @@ class KVStoreDist(KVStoreBase):
                 self._apply_global(key, st, contribs)
 
     # -- intra-party tier ------------------------------------------------
-    def _party_aggregate(self, st: _KeyState, value: torch.Tensor) -> torch.Tensor:
-        """N workers' pushes -> leader-held sum (local server aggregation,
-        kvstore_dist_server.h:1286-1296)."""
+    def _party_launch(self, st: _KeyState, value: torch.Tensor):
+        """Issue the intra-party aggregation (local server aggregation,
+        kvstore_dist_server.h:1286-1296) as an async collective; returns
+        a pending record for _party_finish."""
         topo = self.topo
         buf = value.detach().reshape(-1).float().to(self._device)
         if topo.world_size == 1 or topo.num_workers == 1:
-            return buf.clone()
+            return ("direct", buf.clone(), None, None)
         ctype = self.compression.get("type") if self.compression else None
         if ctype == "2bit":
             # worker-side quantize; leader gathers packed words and
@@ -445,32 +496,40 @@ class KVStoreDist(KVStoreBase):
             if st.residual_2bit is None:
                 st.residual_2bit = torch.zeros(st.numel, device=self._device)
             packed = ops.quantize_2bit(buf, st.residual_2bit, thr)
-            gathered = self._party_gather(packed)
-            if topo.is_leader:
+            if topo.backend == "nccl":
+                # all_gather is universally supported; party links are xGMI
+                out_all = [torch.empty_like(packed)
+                           for _ in range(topo.num_workers)]
+                work = dist.all_gather(out_all, packed,
+                                       group=topo.party_group, async_op=True)
+                return ("2bit", out_all, work, thr)
+            glist = [torch.empty_like(packed)
+                     for _ in range(topo.num_workers)] \
+                if topo.is_leader else None
+            work = dist.gather(packed, gather_list=glist,
+                               dst=topo.leader_rank,
+                               group=topo.party_group, async_op=True)
+            return ("2bit", glist if glist is not None else buf, work, thr)
+        out = buf.clone()
+        work = dist.reduce(out, dst=topo.leader_rank, op=dist.ReduceOp.SUM,
+                           group=topo.party_group, async_op=True)
+        return ("reduce", out, work, None)
+
+    def _party_finish(self, st: _KeyState, pend) -> torch.Tensor:
+        kind, payload, work, extra = pend
+        if work is not None:
+            work.wait()
+        if kind == "2bit" and isinstance(payload, list):
+            if self.topo.is_leader:
                 acc = torch.zeros(st.numel, device=self._device)
                 tmp = torch.empty(st.numel, device=self._device)
-                for p in gathered:
-                    ops.dequantize_2bit(p, st.numel, thr, out=tmp)
+                for p in payload:
+                    ops.dequantize_2bit(p, st.numel, extra, out=tmp)
                     acc += tmp
                 return acc
-            return buf  # non-leaders' value unused
-        out = buf.clone()
-        dist.reduce(out, dst=topo.leader_rank, op=dist.ReduceOp.SUM,
-                    group=topo.party_group)
-        return out
-
-    def _party_gather(self, t: torch.Tensor) -> List[torch.Tensor]:
-        topo = self.topo
-        out = [torch.empty_like(t) for _ in range(topo.num_workers)] \
-            if topo.is_leader else None
-        if self.topo.backend == "nccl":
-            # all_gather is universally supported; party links are xGMI
-            out_all = [torch.empty_like(t) for _ in range(topo.num_workers)]
-            dist.all_gather(out_all, t, group=topo.party_group)
-            return out_all
-        dist.gather(t, gather_list=out, dst=topo.leader_rank,
-                    group=topo.party_group)
-        return out if out is not None else []
+            return payload[0].new_zeros(st.numel, dtype=torch.float32) \
+                if payload else torch.zeros(st.numel, device=self._device)
+        return payload
 
     # -- global (WAN) tier ----------------------------------------------
     def _effective_ctype(self, st: _KeyState) -> Optional[str]:
@@ -671,6 +730,7 @@ class KVStoreDist(KVStoreBase):
                 with torch.no_grad():
                     o.copy_(out[0].to(o.device))
             return
+        self.flush()
         st = self._state(key)
         topo = self.topo
         if self._use_aps():
@@ -753,29 +813,151 @@ class KVStoreDist(KVStoreBase):
                          .to(self._device))
         self.push(key, dense, priority)
 
+    def _comm_device(self) -> torch.device:
+        return torch.device("cpu") if self.topo.backend == "gloo" \
+            else self._device
+
     def row_sparse_pull(self, key, out, row_ids,
                         priority: int = 0) -> None:
-        """Pull only the rows named by row_ids (python/mxnet/kvstore.py:316;
-        GPU row-id dedup was cub Unique in the reference,
-        kvstore_utils.cu:44-111 — torch.unique is the rocPRIM-backed
-        equivalent). `out` must be [len(row_ids), row_width]; `out` and
-        `row_ids` may be aligned lists (multi-device form)."""
+        """Pull only the rows named by row_ids (python/mxnet/kvstore.py:316).
+
+        Rows-only wire (EncodeRowSparseKey kvstore_dist.h:900 semantics):
+        each worker ships its row-id list to its party leader and
+        receives exactly those rows back — worker-tier bytes are
+        proportional to len(row_ids), not to the key size. On the leader
+        (WAN) tier, sharded-dense keys fetch only the union of the
+        party's requested rows from the owner; replicated / compressed /
+        HFA modes hold the value on every leader already and need no
+        global wire at all. Sliced (P3/MultiGPS) keys reassemble on the
+        leader first (rows stripe across every leader's chunk), the
+        worker tier still moves rows only. Row-id dedup is torch.unique
+        (the reference used cub Unique, kvstore_utils.cu:44-111).
+
+        `out` must be [len(row_ids), row_width]; `out` and `row_ids`
+        may be aligned lists (multi-device form)."""
         if isinstance(out, (list, tuple)):
             if not isinstance(row_ids, (list, tuple)):
                 row_ids = [row_ids] * len(out)
             for o, r in zip(out, row_ids):
                 self.row_sparse_pull(key, o, r, priority)
             return
+        self.flush()
         st = self._state(key)
         if len(st.shape) < 2:
             raise ValueError("row_sparse_pull needs a >=2d key")
         rows = st.shape[0]
         width = st.numel // rows
-        full = torch.empty(st.shape, dtype=out.dtype, device=self._device)
-        self.pull(key, full, priority)
-        sel = full.reshape(rows, width)[row_ids.long()]
+        topo = self.topo
+        ids = row_ids.reshape(-1).long()
+
+        if topo.world_size == 1:
+            sel = st.stored.reshape(rows, width)[ids.to(self._device)]
+            with torch.no_grad():
+                out.reshape(ids.numel(), width).copy_(sel.to(out.dtype))
+            return
+
+        cdev = self._comm_device()
+        # --- party tier, phase 1: workers -> leader row-id lists -------
+        my_ids = torch.unique(ids).to(cdev)
+        if topo.num_workers > 1:
+            counts = [torch.zeros(1, dtype=torch.int64, device=cdev)
+                      for _ in range(topo.num_workers)]
+            cnt = torch.tensor([my_ids.numel()], dtype=torch.int64,
+                               device=cdev)
+            dist.all_gather(counts, cnt, group=topo.party_group)
+            if topo.is_leader:
+                peer_ids = {}
+                for i, r in enumerate(topo.party_ranks):
+                    if r == topo.rank:
+                        continue
+                    buf = torch.empty(int(counts[i].item()),
+                                      dtype=torch.int64, device=cdev)
+                    dist.recv(buf, src=r)
+                    peer_ids[r] = buf
+            else:
+                dist.send(my_ids, dst=topo.leader_rank)
+
+        # --- leader (WAN) tier: make the requested rows authoritative --
+        if topo.is_leader:
+            if topo.num_workers > 1:
+                union = torch.unique(torch.cat(
+                    [my_ids] + list(peer_ids.values())))
+            else:
+                union = my_ids
+            self._global_pull_rows(key, st, union.to(self._device),
+                                   rows, width)
+
+        # --- party tier, phase 2: leader -> workers selected rows ------
+        if topo.num_workers > 1:
+            if topo.is_leader:
+                stored2d = st.stored.reshape(rows, width)
+                for i, r in enumerate(topo.party_ranks):
+                    if r == topo.rank:
+                        continue
+                    sel = stored2d[peer_ids[r].to(self._device)] \
+                        .to(cdev).contiguous()
+                    dist.send(sel, dst=r)
+                sel = stored2d[ids.to(self._device)]
+            else:
+                rbuf = torch.empty(my_ids.numel(), width, device=cdev)
+                dist.recv(rbuf, src=topo.leader_rank)
+                # scatter unique rows back to the (possibly repeated)
+                # requested order
+                pos = torch.searchsorted(my_ids.cpu(), ids.cpu())
+                sel = rbuf[pos.to(cdev)]
+        else:
+            sel = st.stored.reshape(rows, width)[ids.to(self._device)]
         with torch.no_grad():
-            out.reshape(len(row_ids), width).copy_(sel)
+            out.reshape(ids.numel(), width).copy_(sel.to(out.device,
+                                                         out.dtype))
+
+    def _global_pull_rows(self, key, st: _KeyState, union: torch.Tensor,
+                          rows: int, width: int) -> None:
+        """Leader-tier row fetch. Only the sharded dense mode has a wire
+        here (the owner holds the authoritative value); everything else
+        already replays the update on every leader."""
+        topo = self.topo
+        P = topo.num_parties
+        if P == 1 or self.cfg.use_hfa:
+            return
+        if self._use_aps():
+            if topo.is_leader:
+                st.stored = self._ensure_aps().pull(key)
+            return
+        ctype = self._effective_ctype(st)
+        need_wire = self.global_mode == "sharded" and not (
+            ctype in ("bsc", "fp16", "dgt", "2bit")
+            or self.cfg.mode == "dist_async")
+        if not need_wire:
+            return
+        if st.sliced:
+            # rows stripe across every leader's chunk: reassemble once
+            self._global_exchange_pull(key, st)
+            return
+        cdev = self._comm_device()
+        owner_leader = topo.leader_ranks[st.owner_party]
+        if topo.rank == owner_leader:
+            stored2d = st.stored.reshape(rows, width)
+            for r in topo.leader_ranks:
+                if r == owner_leader:
+                    continue
+                cnt = torch.zeros(1, dtype=torch.int64, device=cdev)
+                dist.recv(cnt, src=r)
+                idx = torch.empty(int(cnt.item()), dtype=torch.int64,
+                                  device=cdev)
+                dist.recv(idx, src=r)
+                sel = stored2d[idx.to(self._device)].to(cdev).contiguous()
+                dist.send(sel, dst=r)
+                self.wan.charge(sel.numel() * 4 + idx.numel() * 8)
+        else:
+            u = union.to(cdev)
+            dist.send(torch.tensor([u.numel()], dtype=torch.int64,
+                                   device=cdev), dst=owner_leader)
+            dist.send(u, dst=owner_leader)
+            rbuf = torch.empty(u.numel(), width, device=cdev)
+            dist.recv(rbuf, src=owner_leader)
+            st.stored.reshape(rows, width)[union.to(self._device)] = \
+                rbuf.to(self._device)
 
     # ------------------------------------------------------------------
     # checkpointing (layout parity: named-param dict + separate optimizer
@@ -786,6 +968,7 @@ class KVStoreDist(KVStoreBase):
         optimizer itself (the spec), matching the reference where the
         pickled optimizer rides along only on request
         (python/mxnet/kvstore.py:566-592)."""
+        self.flush()
         if self.optimizer is None:
             raise RuntimeError("no optimizer set")
         blob = self.optimizer.state_dict()

@@ -406,3 +406,54 @@ def _list_forms(rank, world):
 
 def test_list_forms_ws4():
     run_dist(4, _list_forms)
+
+
+# ---------------------------------------------------------------------------
+# Row-sparse pull: rows-only wire (kvstore_dist.h:900 EncodeRowSparseKey)
+# ---------------------------------------------------------------------------
+
+def _row_sparse_hier(rank, world):
+    kv = _mk(num_parties=2)
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    torch.manual_seed(0)
+    w = torch.randn(12, 4)
+    kv.init("emb", w)
+    g = torch.zeros(12, 4)
+    g[rank] = 1.0  # each worker touches a different row
+    kv.push("emb", g)
+    # dense pull = golden model
+    dense = torch.empty(12, 4)
+    kv.pull("emb", dense)
+    # per-worker distinct (repeated, unsorted) id lists
+    ids = torch.tensor([rank, (rank + 5) % 12, rank, 11 - rank])
+    out = torch.empty(len(ids), 4)
+    kv.row_sparse_pull("emb", out, ids)
+    assert torch.allclose(out, dense[ids], atol=1e-6), \
+        (rank, out, dense[ids])
+
+
+def test_row_sparse_pull_rows_only_ws4():
+    run_dist(4, _row_sparse_hier)
+
+
+def _row_sparse_sharded_dense(rank, world):
+    # sharded dense mode (owner leader holds the value): the leader
+    # tier must fetch only the requested-row union from the owner
+    kv = _mk(num_parties=2)
+    kv.global_mode = "sharded"
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.5))
+    torch.manual_seed(1)
+    w = torch.randn(10, 3)
+    kv.init("emb", w)
+    g = torch.ones(10, 3) * (rank + 1)
+    kv.push("emb", g)
+    dense = torch.empty(10, 3)
+    kv.pull("emb", dense)
+    ids = torch.tensor([2, 7, 2])
+    out = torch.empty(3, 3)
+    kv.row_sparse_pull("emb", out, ids)
+    assert torch.allclose(out, dense[ids], atol=1e-6), (rank, out)
+
+
+def test_row_sparse_pull_sharded_ws2():
+    run_dist(2, _row_sparse_sharded_dense)
