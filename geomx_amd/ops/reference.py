@@ -99,13 +99,26 @@ def bsc_sample_size(n: int, ratio: float) -> int:
     return int(10 / ratio)
 
 
+def bsc_sample_indices(n: int, sample_size: int, seed: int = 42
+                       ) -> torch.Tensor:
+    """Seeded random sample positions, shared by the CPU and GPU
+    boundary estimators so a mixed CPU/GPU party computes the SAME
+    threshold (the paths used to diverge — strided vs
+    randperm — which a structured gradient could bias). Sampling is
+    with replacement (torch.randint): for threshold ESTIMATION over
+    0.5% of elements it is statistically indistinguishable from the
+    reference's srand/rand draw (gradient_compression.cc:213-231,
+    itself with replacement)."""
+    gen = torch.Generator(device="cpu").manual_seed(seed)
+    return torch.randint(0, n, (sample_size,), generator=gen)
+
+
 def bsc_boundary(v: torch.Tensor, ratio: float, seed: int = 42) -> float:
     """Sampled top-k threshold estimation over |v|."""
     n = v.numel()
     sample_size = min(bsc_sample_size(n, ratio), n)
     top_k = max(1, int(sample_size * ratio))
-    gen = torch.Generator(device="cpu").manual_seed(seed)
-    idx = torch.randperm(n, generator=gen)[:sample_size].to(v.device)
+    idx = bsc_sample_indices(n, sample_size, seed).to(v.device)
     sample = v.reshape(-1)[idx].abs()
     return torch.topk(sample, min(top_k, sample.numel())).values[-1].item()
 
