@@ -101,7 +101,8 @@ def main():
         compression=args.compress, bsc_ratio=args.bsc_ratio,
         wan_gbps=args.wan_gbps, party_wan_gbps=pw,
         bucket_mb=args.bucket_mb, comm_dtype=args.comm_dtype)
-    topo = init_topology(topo_parties, None, backend)
+    topo = init_topology(topo_parties, None, backend,
+                     "cuda" if use_cuda else None)
     device = topo.device
 
     # fail loudly if the native extension is missing on a GPU machine

@@ -48,7 +48,7 @@ from typing import Dict, List, Optional
 import torch
 import torch.distributed as dist
 
-from .. import ops
+from .. import comm, ops
 from ..config import Config
 from ..topology import Topology, init_topology
 from .base import KVStoreBase
@@ -364,7 +364,7 @@ class KVStoreDist(KVStoreBase):
         flat = value.detach().reshape(-1).float().to(self._device)
         # rank 0's value is authoritative at init (reference: first init wins)
         if dist.is_initialized() and self.topo.world_size > 1:
-            dist.broadcast(flat, src=0)
+            comm.broadcast(flat, src=0)
         st.stored = flat.clone()
         if self.cfg.use_hfa:
             # the reference seeds the milestone with the initial params
@@ -500,18 +500,18 @@ class KVStoreDist(KVStoreBase):
                 # all_gather is universally supported; party links are xGMI
                 out_all = [torch.empty_like(packed)
                            for _ in range(topo.num_workers)]
-                work = dist.all_gather(out_all, packed,
+                work = comm.all_gather(out_all, packed,
                                        group=topo.party_group, async_op=True)
                 return ("2bit", out_all, work, thr)
             glist = [torch.empty_like(packed)
                      for _ in range(topo.num_workers)] \
                 if topo.is_leader else None
-            work = dist.gather(packed, gather_list=glist,
+            work = comm.gather(packed, gather_list=glist,
                                dst=topo.leader_rank,
                                group=topo.party_group, async_op=True)
             return ("2bit", glist if glist is not None else buf, work, thr)
         out = buf.clone()
-        work = dist.reduce(out, dst=topo.leader_rank, op=dist.ReduceOp.SUM,
+        work = comm.reduce(out, dst=topo.leader_rank, op=dist.ReduceOp.SUM,
                            group=topo.party_group, async_op=True)
         return ("reduce", out, work, None)
 
@@ -564,8 +564,8 @@ class KVStoreDist(KVStoreBase):
             payload = vals.numel() * 4 + idx.numel() * 4
             vlist = [torch.empty_like(vals) for _ in range(P)]
             ilist = [torch.empty_like(idx) for _ in range(P)]
-            dist.all_gather(vlist, vals, group=group)
-            dist.all_gather(ilist, idx, group=group)
+            comm.all_gather(vlist, vals, group=group)
+            comm.all_gather(ilist, idx, group=group)
             self.wan.charge(cross_party_bytes("all_gather", payload, P))
             dense = []
             for v_, i_ in zip(vlist, ilist):
@@ -583,7 +583,7 @@ class KVStoreDist(KVStoreBase):
                                                    device=self._device)
             packed = ops.quantize_2bit(party_sum, st.residual_2bit_wan, thr)
             plist = [torch.empty_like(packed) for _ in range(P)]
-            dist.all_gather(plist, packed, group=group)
+            comm.all_gather(plist, packed, group=group)
             self.wan.charge(cross_party_bytes("all_gather",
                                               packed.numel() * 4, P))
             dense = []
@@ -599,7 +599,7 @@ class KVStoreDist(KVStoreBase):
                                                wire_dtype=torch.float16)]
             h = party_sum.to(torch.float16)
             hlist = [torch.empty_like(h) for _ in range(P)]
-            dist.all_gather(hlist, h, group=group)
+            comm.all_gather(hlist, h, group=group)
             self.wan.charge(cross_party_bytes("all_gather", h.numel() * 2, P))
             return [x.float() for x in hlist]
 
@@ -613,7 +613,7 @@ class KVStoreDist(KVStoreBase):
                 gathered = [[torch.empty_like(t) for _ in range(P)]
                             for t in payload]
                 for lst, t in zip(gathered, payload):
-                    dist.all_gather(lst, t, group=group)
+                    comm.all_gather(lst, t, group=group)
                 self.wan.charge(cross_party_bytes(
                     "all_gather", dg.wire_bytes(), P))
                 return [dg.decompress(*(lst[p] for lst in gathered))
@@ -622,7 +622,7 @@ class KVStoreDist(KVStoreBase):
             # semantics; no byte saving exists on a reliable fabric)
             contrib_now, wire = dg.transform(party_sum)
             hlist = [torch.empty_like(contrib_now) for _ in range(P)]
-            dist.all_gather(hlist, contrib_now, group=group)
+            comm.all_gather(hlist, contrib_now, group=group)
             self.wan.charge(cross_party_bytes("all_gather", wire, P))
             return list(hlist)
 
@@ -636,7 +636,7 @@ class KVStoreDist(KVStoreBase):
                 if lo >= hi:
                     continue
                 sl = party_sum[lo:hi].clone()
-                dist.reduce(sl, dst=topo.leader_ranks[c],
+                comm.reduce(sl, dst=topo.leader_ranks[c],
                             op=dist.ReduceOp.SUM, group=group)
                 if topo.party_id == c:
                     my_slice = sl
@@ -658,13 +658,13 @@ class KVStoreDist(KVStoreBase):
             return [self._ts.allreduce_sum(party_sum)]
         if self.global_mode == "replicated" or self.cfg.mode == "dist_async":
             flist = [torch.empty_like(party_sum) for _ in range(P)]
-            dist.all_gather(flist, party_sum, group=group)
+            comm.all_gather(flist, party_sum, group=group)
             self.wan.charge(cross_party_bytes("all_gather", st.numel * 4, P))
             return list(flist)
         # sharded dense: reduce to owner (sum), single contribution
         out = party_sum.clone()
         owner_leader = topo.leader_ranks[st.owner_party]
-        dist.reduce(out, dst=owner_leader, op=dist.ReduceOp.SUM, group=group)
+        comm.reduce(out, dst=owner_leader, op=dist.ReduceOp.SUM, group=group)
         self.wan.charge(cross_party_bytes("reduce", st.numel * 4, P))
         if topo.leader_rank == owner_leader and topo.is_leader \
                 and topo.party_id == st.owner_party:
@@ -706,7 +706,7 @@ class KVStoreDist(KVStoreBase):
             return
         P = topo.num_parties
         delta = (st.stored - st.milestone) / P
-        dist.all_reduce(delta, op=dist.ReduceOp.SUM, group=topo.leader_group)
+        comm.all_reduce(delta, op=dist.ReduceOp.SUM, group=topo.leader_group)
         self.wan.charge(cross_party_bytes("all_reduce", st.numel * 4, P))
         st.stored = st.milestone + delta
         st.milestone = st.stored.clone()
@@ -740,7 +740,7 @@ class KVStoreDist(KVStoreBase):
             self._global_exchange_pull(key, st)
         # intra-party: leader broadcasts authoritative value to its workers
         if topo.world_size > 1 and topo.num_workers > 1:
-            dist.broadcast(st.stored, src=topo.leader_rank,
+            comm.broadcast(st.stored, src=topo.leader_rank,
                            group=topo.party_group)
         result = st.stored.reshape(st.shape).to(out.dtype)
         with torch.no_grad():
@@ -775,7 +775,7 @@ class KVStoreDist(KVStoreBase):
             if lo < hi:
                 mine[:hi - lo] = st.stored[lo:hi]
             parts = [torch.empty_like(mine) for _ in range(P)]
-            dist.all_gather(parts, mine, group=group)
+            comm.all_gather(parts, mine, group=group)
             self.wan.charge(cross_party_bytes("all_gather", chunk * 4, P))
             st.stored = torch.cat(parts)[:st.numel]
             return
@@ -791,7 +791,7 @@ class KVStoreDist(KVStoreBase):
         # (ops.bsc_pull_compress + k_bsc_pull_pack) remains for star
         # topologies and is golden-tested against the reference
         # semantics on CPU and gfx950.
-        dist.broadcast(st.stored, src=owner_leader, group=group)
+        comm.broadcast(st.stored, src=owner_leader, group=group)
         self.wan.charge(cross_party_bytes("broadcast", st.numel * 4, P))
 
     def push_row_sparse(self, key, row_ids: torch.Tensor,
@@ -864,7 +864,7 @@ class KVStoreDist(KVStoreBase):
                       for _ in range(topo.num_workers)]
             cnt = torch.tensor([my_ids.numel()], dtype=torch.int64,
                                device=cdev)
-            dist.all_gather(counts, cnt, group=topo.party_group)
+            comm.all_gather(counts, cnt, group=topo.party_group)
             if topo.is_leader:
                 peer_ids = {}
                 for i, r in enumerate(topo.party_ranks):
@@ -872,10 +872,10 @@ class KVStoreDist(KVStoreBase):
                         continue
                     buf = torch.empty(int(counts[i].item()),
                                       dtype=torch.int64, device=cdev)
-                    dist.recv(buf, src=r)
+                    comm.recv(buf, src=r)
                     peer_ids[r] = buf
             else:
-                dist.send(my_ids, dst=topo.leader_rank)
+                comm.send(my_ids, dst=topo.leader_rank)
 
         # --- leader (WAN) tier: make the requested rows authoritative --
         if topo.is_leader:
@@ -896,11 +896,11 @@ class KVStoreDist(KVStoreBase):
                         continue
                     sel = stored2d[peer_ids[r].to(self._device)] \
                         .to(cdev).contiguous()
-                    dist.send(sel, dst=r)
+                    comm.send(sel, dst=r)
                 sel = stored2d[ids.to(self._device)]
             else:
                 rbuf = torch.empty(my_ids.numel(), width, device=cdev)
-                dist.recv(rbuf, src=topo.leader_rank)
+                comm.recv(rbuf, src=topo.leader_rank)
                 # scatter unique rows back to the (possibly repeated)
                 # requested order
                 pos = torch.searchsorted(my_ids.cpu(), ids.cpu())
@@ -942,20 +942,20 @@ class KVStoreDist(KVStoreBase):
                 if r == owner_leader:
                     continue
                 cnt = torch.zeros(1, dtype=torch.int64, device=cdev)
-                dist.recv(cnt, src=r)
+                comm.recv(cnt, src=r)
                 idx = torch.empty(int(cnt.item()), dtype=torch.int64,
                                   device=cdev)
-                dist.recv(idx, src=r)
+                comm.recv(idx, src=r)
                 sel = stored2d[idx.to(self._device)].to(cdev).contiguous()
-                dist.send(sel, dst=r)
+                comm.send(sel, dst=r)
                 self.wan.charge(sel.numel() * 4 + idx.numel() * 8)
         else:
             u = union.to(cdev)
-            dist.send(torch.tensor([u.numel()], dtype=torch.int64,
+            comm.send(torch.tensor([u.numel()], dtype=torch.int64,
                                    device=cdev), dst=owner_leader)
-            dist.send(u, dst=owner_leader)
+            comm.send(u, dst=owner_leader)
             rbuf = torch.empty(u.numel(), width, device=cdev)
-            dist.recv(rbuf, src=owner_leader)
+            comm.recv(rbuf, src=owner_leader)
             st.stored.reshape(rows, width)[union.to(self._device)] = \
                 rbuf.to(self._device)
 

@@ -46,6 +46,8 @@ from typing import Callable, List, Optional, Tuple
 import torch
 import torch.distributed as dist
 
+from .. import comm
+
 from .wan import TokenBucket
 
 MAX_GREED_RATE_TS = 0.9  # reference: ps-lite van.h max_greed_rate
@@ -127,7 +129,7 @@ class TSExchange:
             extra = self.link_model(self.me, dst_party, nbytes)
             if extra > 0:
                 time.sleep(extra)
-        dist.send(wire, dst=self.ranks[dst_party], group=self.group)
+        comm.send(wire, dst=self.ranks[dst_party], group=self.group)
         if wire.is_cuda:
             # NCCL send only enqueues on the comm stream; sync so the
             # measured interval covers the actual transfer (on the
@@ -142,7 +144,7 @@ class TSExchange:
     def _recv(self, numel: int, src_party: int, device,
               wire_dtype: torch.dtype) -> torch.Tensor:
         wire = torch.empty(numel, dtype=wire_dtype, device=device)
-        dist.recv(wire, src=self.ranks[src_party], group=self.group)
+        comm.recv(wire, src=self.ranks[src_party], group=self.group)
         return wire.float()
 
     def _sync_rows(self, device):
@@ -153,7 +155,7 @@ class TSExchange:
         row = torch.tensor(self._my_row, dtype=torch.float64,
                            device=device)
         rows = [torch.empty_like(row) for _ in range(self.P)]
-        dist.all_gather(rows, row, group=self.group)
+        comm.all_gather(rows, row, group=self.group)
         for i, r in enumerate(rows):
             for j, v in enumerate(r.tolist()):
                 if v >= 0:

@@ -39,7 +39,7 @@ from typing import Dict, List, Optional
 import torch
 import torch.distributed as dist
 
-from .. import ops
+from .. import comm, ops
 from ..config import Config
 from ..kvstore.optimizer import OptimizerSpec, ServerOptimizer
 from ..kvstore.wan import TokenBucket, cross_party_bytes
@@ -214,13 +214,13 @@ class GeoTrainer:
                 # half-traffic wire format (FP16-transmission analog);
                 # RCCL sums in bf16, result cast back for the fp32 update
                 h = b.flat.to(torch.bfloat16)
-                work = dist.all_reduce(h, async_op=True)
+                work = comm.all_reduce(h, async_op=True)
                 b.work = _Bf16Work(work, h, b.flat)
             else:
-                b.work = dist.all_reduce(b.flat, async_op=True)
+                b.work = comm.all_reduce(b.flat, async_op=True)
         else:
             b.flat.div_(self.topo.num_all_workers)
-            b.work = dist.all_reduce(b.flat, group=self.topo.party_group,
+            b.work = comm.all_reduce(b.flat, group=self.topo.party_group,
                                      async_op=True)
 
     # ------------------------------------------------------------------
@@ -308,7 +308,7 @@ class GeoTrainer:
                     apply_buf = b.wan_buf.float()
                 if topo.num_workers > 1:
                     # LAN fan-out of the arrived global gradient
-                    dist.broadcast(apply_buf, src=topo.leader_rank,
+                    comm.broadcast(apply_buf, src=topo.leader_rank,
                                    group=topo.party_group)
                 self.server_opt.update(("bucket", b.index), b.param_flat,
                                        apply_buf)
@@ -321,7 +321,7 @@ class GeoTrainer:
                     snap = b.flat.clone()
                     nbytes = snap.numel() * 4
                 b.wan_buf = snap
-                b.wan_work = dist.all_reduce(snap, group=topo.leader_group,
+                b.wan_work = comm.all_reduce(snap, group=topo.leader_group,
                                              async_op=True)
                 b.wan_ready = self.wan.charge_async(
                     cross_party_bytes("all_reduce", nbytes, P))
@@ -354,8 +354,8 @@ class GeoTrainer:
                                                  self.cfg.bsc_ratio)
                     vlist = [torch.empty_like(vals) for _ in range(P)]
                     ilist = [torch.empty_like(idx) for _ in range(P)]
-                    dist.all_gather(vlist, vals, group=topo.leader_group)
-                    dist.all_gather(ilist, idx, group=topo.leader_group)
+                    comm.all_gather(vlist, vals, group=topo.leader_group)
+                    comm.all_gather(ilist, idx, group=topo.leader_group)
                     self.wan.charge(cross_party_bytes(
                         "all_gather", vals.numel() * 8, P))
                     acc = torch.zeros_like(b.flat)
@@ -379,7 +379,7 @@ class GeoTrainer:
                         gathered = [[torch.empty_like(t) for _ in range(P)]
                                     for t in payload]
                         for lst, t in zip(gathered, payload):
-                            dist.all_gather(lst, t, group=topo.leader_group)
+                            comm.all_gather(lst, t, group=topo.leader_group)
                         self.wan.charge(cross_party_bytes(
                             "all_gather", b.dgt.wire_bytes(), P))
                         acc = torch.zeros_like(b.flat)
@@ -390,7 +390,7 @@ class GeoTrainer:
                     else:
                         lossy, wire = b.dgt.transform(b.flat)
                         b.flat.copy_(lossy)
-                        dist.all_reduce(b.flat, group=topo.leader_group)
+                        comm.all_reduce(b.flat, group=topo.leader_group)
                         self.wan.charge(cross_party_bytes(
                             "all_reduce", wire, P))
                 elif ctype == "2bit":
@@ -403,7 +403,7 @@ class GeoTrainer:
                     thr = self.cfg.threshold
                     packed = ops.quantize_2bit(b.flat, b.res2bit, thr)
                     plist = [torch.empty_like(packed) for _ in range(P)]
-                    dist.all_gather(plist, packed, group=topo.leader_group)
+                    comm.all_gather(plist, packed, group=topo.leader_group)
                     self.wan.charge(cross_party_bytes(
                         "all_gather", packed.numel() * 4, P))
                     acc = torch.zeros_like(b.flat)
@@ -419,18 +419,18 @@ class GeoTrainer:
                             b.flat, wire_dtype=torch.float16))
                     else:
                         h = b.flat.to(torch.float16)
-                        dist.all_reduce(h, group=topo.leader_group)
+                        comm.all_reduce(h, group=topo.leader_group)
                         self.wan.charge(cross_party_bytes(
                             "all_reduce", h.numel() * 2, P))
                         b.flat.copy_(h.float())
                 elif self._ts is not None:
                     b.flat.copy_(self._ts.allreduce_sum(b.flat))
                 else:
-                    dist.all_reduce(b.flat, group=topo.leader_group)
+                    comm.all_reduce(b.flat, group=topo.leader_group)
                     self.wan.charge(cross_party_bytes(
                         "all_reduce", b.flat.numel() * 4, P))
             if topo.num_workers > 1:
-                dist.broadcast(b.flat, src=topo.leader_rank,
+                comm.broadcast(b.flat, src=topo.leader_rank,
                                group=topo.party_group)
 
     @property
