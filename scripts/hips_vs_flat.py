@@ -63,6 +63,8 @@ def main():
                       "--bsc-ratio", str(args.bsc_ratio)]),
         ("hips mpq", ["--mode", "hips", "--compress", "mpq",
                       "--bsc-ratio", str(args.bsc_ratio)]),
+        ("hips 2bit", ["--mode", "hips", "--compress", "2bit"]),
+        ("hips dgt", ["--mode", "hips", "--compress", "dgt"]),
         ("hips async (stale-1)", ["--mode", "hips", "--sync-mode",
                                   "dist_async"]),
         ("hips async+fp16", ["--mode", "hips", "--sync-mode", "dist_async",
