@@ -32,7 +32,8 @@ def run_bench(nproc, extra, steps, warmup, bs, port):
            "--json-out", out] + extra
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=1800)
     if r.returncode != 0:
-        print(r.stdout[-2000:], r.stderr[-2000:], file=sys.stderr)
+        print(r.stdout[-4000:], file=sys.stderr)
+        print(r.stderr[-4000:], file=sys.stderr)
         raise RuntimeError(f"bench failed: {extra}")
     with open(out) as f:
         return json.loads(f.read())
@@ -49,12 +50,17 @@ def main():
     ap.add_argument("--image-size", type=int, default=28)
     ap.add_argument("--bsc-ratio", type=float, default=0.01)
     ap.add_argument("--port", type=int, default=29640)
+    ap.add_argument("--backend", type=str, default="auto",
+                    help="pass 'gloo' to run N ranks on one GPU "
+                         "(GPU compute, gloo wire; RCCL rejects "
+                         "duplicate devices)")
     ap.add_argument("--json-out", type=str, default=None)
     args = ap.parse_args()
 
     common = ["--image-size", str(args.image_size),
               "--parties", str(args.parties),
-              "--wan-gbps", str(args.wan_gbps)]
+              "--wan-gbps", str(args.wan_gbps),
+              "--backend", args.backend]
     configs = [
         ("flat (baseline)", ["--mode", "flat"]),
         ("hips dense", ["--mode", "hips"]),
