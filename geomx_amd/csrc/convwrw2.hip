@@ -1,0 +1,270 @@
+// Weight-gradient (wrw) v2 for 5x5 stride-1 NHWC conv, CI=16, on gfx950.
+//
+// dW[kh,kw,ci][o] = sum_{n,ho,wo} in[n,ho+kh,wo+kw,ci] * gout[n,ho,wo,o]
+//
+// v1 (convwrw.hip) stages TRANSPOSED images with per-element ds writes
+// and gathers the shifted A fragment with an alignbyte funnel; it is
+// LDS-gather bound (1.10 ms standalone on the conv2 shape vs MIOpen's
+// 1.60 in-step). v2 removes both costs:
+//
+//   * LDS images keep the NATURAL pixel-major layout ([pix][ci] for the
+//     input — identical to global NHWC — and per-o-tile [pix][16] for
+//     gout), so staging is pure LDS-DMA (__builtin_amdgcn_global_load_lds,
+//     16 B/lane): zero VALU work, zero ds_write instructions. Lanes whose
+//     pixel falls in the zero-padding read a device zero page, so padding
+//     is zero-filled in the same pass.
+//   * MFMA fragments come from ds_read_b64_tr_b16 (gfx950's LDS
+//     transpose read, cdna_hip_programming.md T10): each 16-lane group
+//     reads one 4(pix)x16(chan) row-major tile — lane g supplies
+//     tile_base + g*8 (8B-aligned) and receives CHANNEL COLUMN g across
+//     the 4 pixels. Both A and B fragments use the SAME k-slot->pixel
+//     permutation (slot q*8+j <-> pixel 4q+j / 16+4q+j), so the MFMA
+//     contraction pairs matched pixels. The kw shift of A is a whole-
+//     pixel base offset — no misaligned gathers, no funnel shifts.
+//
+// Work layout: grid-persistent workgroups own row-blocks of WRW2_R gout
+// rows; 4 waves split the 25 (kh,kw) tap tiles round-robin and each
+// wave computes BOTH o-tiles for its taps (A-fragment reuse), keeping
+// all accumulators in VGPRs until one final per-workgroup slab flush
+// (partials summed in torch, same contract as v1).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+typedef unsigned short bf16_t;
+typedef __attribute__((ext_vector_type(4))) short bf16x4;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define WRW2_THREADS 256  // 4 waves
+#define WRW2_R 2          // gout rows per block
+
+__device__ __attribute__((aligned(64))) bf16_t g_wrw2_zeros[1024];
+
+// ds_read_b64_tr_b16: the 16-lane group covers one 4x16 bf16 row-major
+// tile (lane g supplies tile_base + g*8); lane g receives column g
+// (4 bf16, one per tile row). `a` is this lane's LDS byte address.
+__device__ __forceinline__ bf16x4 tr16_b64(unsigned a) {
+  bf16x4 d;
+  asm volatile("ds_read_b64_tr_b16 %0, %1"
+               : "=v"(d)
+               : "v"(a)
+               : "memory");
+  return d;
+}
+
+__device__ __forceinline__ unsigned lds_addr(const void* p) {
+  return (unsigned)(uintptr_t)(const __attribute__((address_space(3)))
+                               void*)p;
+}
+
+template <int COT, int PIX>
+__global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw16_nhwc(
+    const bf16_t* __restrict__ in,    // [N][Hi][Wi][16]
+    const bf16_t* __restrict__ gout,  // [N][Ho][Wo][CO]
+    float* __restrict__ part,         // [nWG][T16][CO]
+    int Nn, int Hi, int Wi, int Ho, int Wo) {
+  constexpr int CI = 16;
+  constexpr int CO = COT * 16;
+  constexpr int NT = 25;                 // tap tiles = (kh,kw) pairs
+  constexpr int T16 = NT * 16;
+  // row region: one 16-wide subimage row, padded to whole 1 KiB LDS-DMA
+  // chunks so every glds writes full 64-lane spans (no partial EXEC)
+  constexpr int ROW_BYTES = PIX * CI * 2;
+  constexpr int NCH = (ROW_BYTES + 1023) / 1024;       // glds chunks/row
+  constexpr int RPB = NCH * 1024;                      // region bytes
+  constexpr int NROW_IN = WRW2_R + 4;
+  constexpr int NROW_GO = WRW2_R * COT;
+  __shared__ __attribute__((aligned(128))) char lds_all[(NROW_IN + NROW_GO) *
+                                                        RPB];
+
+  const int lane = threadIdx.x & 63;
+  const int q = lane >> 4;
+  const int m = lane & 15;
+  const int wid = threadIdx.x >> 6;
+
+  const int W = (Wo + 31) >> 5;          // 32-pixel K windows per row
+  const int blocks_h = (Ho + WRW2_R - 1) / WRW2_R;
+  const long long n_blocks = (long long)Nn * blocks_h;
+  const int wg = blockIdx.x;
+  const int n_wg = gridDim.x;
+
+  const unsigned lds0 = lds_addr(lds_all);
+
+  // wave-owned tap tiles: tt = wid + 4*j, accumulators for BOTH o-tiles
+  constexpr int MAXT = (NT + 3) / 4;     // 7
+  f32x4 acc[MAXT][COT];
+#pragma unroll
+  for (int j = 0; j < MAXT; ++j)
+#pragma unroll
+    for (int t = 0; t < COT; ++t) acc[j][t] = (f32x4)0.0f;
+
+  for (long long blk = wg; blk < n_blocks; blk += n_wg) {
+    const int bh = (int)(blk % blocks_h);
+    const long long n = blk / blocks_h;
+    const int ho0 = bh * WRW2_R;
+    const int nrows = (Ho - ho0) < WRW2_R ? (Ho - ho0) : WRW2_R;
+    const int irows = nrows + 4;
+
+    // ---- stage via LDS-DMA: chunks round-robin over the 4 waves.
+    // A rows: a global NHWC row IS the LDS [pix][16] row (straight copy).
+    {
+      const int nchunks_a = irows * NCH;
+      for (int t = wid; t < nchunks_a; t += 4) {
+        const int ir = t / NCH;
+        const int c = t - ir * NCH;
+        const int slot = c * 64 + lane;              // 16B granule index
+        const int pix = slot >> 1;
+        const bf16_t* src =
+            (pix < Wi) ? in + ((n * Hi + (ho0 + ir)) * (long long)Wi * CI +
+                               (long long)slot * 8)
+                       : g_wrw2_zeros;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)src,
+            (__attribute__((address_space(3))) void*)(lds_all + ir * RPB +
+                                                      c * 1024),
+            16, 0, 0);
+      }
+      // gout rows: global [pix][CO] -> per-o-tile LDS [pix][16]
+      const int nchunks_b = nrows * COT * NCH;
+      for (int t = wid; t < nchunks_b; t += 4) {
+        const int rr = t / (COT * NCH);
+        const int rem = t - rr * (COT * NCH);
+        const int ot = rem / NCH;
+        const int c = rem - ot * NCH;
+        const int slot = c * 64 + lane;
+        const int pix = slot >> 1;
+        const int half = slot & 1;
+        const bf16_t* src =
+            (pix < Wo)
+                ? gout + ((n * Ho + (ho0 + rr)) * (long long)Wo * CO +
+                          (long long)pix * CO + ot * 16 + half * 8)
+                : g_wrw2_zeros;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)src,
+            (__attribute__((address_space(3))) void*)(lds_all +
+                                                      (NROW_IN + rr * COT +
+                                                       ot) * RPB +
+                                                      c * 1024),
+            16, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    for (int r = 0; r < nrows; ++r) {
+      for (int w = 0; w < W; ++w) {
+        const int pix0 = w * 32;
+        // B fragments (shared across this wave's tap tiles): per ot,
+        // two tr reads = pixel tiles [pix0+4q, +4) and [pix0+16+4q, +4)
+        bf16x4 bfr[COT][2];
+#pragma unroll
+        for (int ot = 0; ot < COT; ++ot) {
+          const unsigned bb = lds0 + (NROW_IN + r * COT + ot) * RPB +
+                              (pix0 + 4 * q) * 32 + m * 8;
+          bfr[ot][0] = tr16_b64(bb);
+          bfr[ot][1] = tr16_b64(bb + 16 * 32);
+        }
+        // A fragments + MFMAs per owned tap tile
+        bf16x4 afr[MAXT][2];
+#pragma unroll
+        for (int j = 0; j < MAXT; ++j) {
+          const int tt = wid + 4 * j;
+          if (tt >= NT) continue;
+          const int kh = tt / 5;
+          const int kw = tt - kh * 5;
+          const unsigned ab = lds0 + (r + kh) * RPB +
+                              (pix0 + kw + 4 * q) * 32 + m * 8;
+          afr[j][0] = tr16_b64(ab);
+          afr[j][1] = tr16_b64(ab + 16 * 32);
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int j = 0; j < MAXT; ++j) {
+          const int tt = wid + 4 * j;
+          if (tt >= NT) continue;
+          union { struct { bf16x4 lo, hi; } p; bf16x8 v; } a;
+          a.p.lo = afr[j][0];
+          a.p.hi = afr[j][1];
+#pragma unroll
+          for (int ot = 0; ot < COT; ++ot) {
+            union { struct { bf16x4 lo, hi; } p; bf16x8 v; } b;
+            b.p.lo = bfr[ot][0];
+            b.p.hi = bfr[ot][1];
+            acc[j][ot] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a.v, b.v, acc[j][ot], 0, 0, 0);
+          }
+        }
+      }
+    }
+    __syncthreads();  // before the next block restages LDS
+  }
+
+  // ---- flush accumulators: part[wg][tap][o]
+  // C/D layout of mfma_f32_16x16x32_bf16: lane (q,m), reg i ->
+  // row = q*4 + i (the A row = tap-within-tile), col = m (the B col = o)
+  float* base = part + (long long)wg * T16 * CO;
+#pragma unroll
+  for (int j = 0; j < MAXT; ++j) {
+    const int tt = wid + 4 * j;
+    if (tt >= NT) continue;
+#pragma unroll
+    for (int ot = 0; ot < COT; ++ot) {
+      const int o = ot * 16 + m;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int tap = tt * 16 + q * 4 + i;
+        base[(long long)tap * CO + o] = acc[j][ot][i];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// probe: verify the tr16 lane mapping numerically (test_kernels_gpu).
+// in : 64 bf16 = one 4x16 row-major tile. out[lane][j] = element j of
+// lane `lane`'s tr read — expected out[l][j] == tile[j][l&15] for the
+// first 16-lane group (all groups read the same tile here).
+// ---------------------------------------------------------------------
+__global__ void k_tr16_probe(const bf16_t* __restrict__ in,
+                             bf16_t* __restrict__ out) {
+  __shared__ __attribute__((aligned(128))) bf16_t tile[64];
+  const int lane = threadIdx.x & 63;
+  tile[lane] = in[lane];
+  __syncthreads();
+  const unsigned a = lds_addr(tile) + (lane & 15) * 8;
+  bf16x4 d = tr16_b64(a);
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = d[j];
+}
+
+extern "C" {
+
+int geops_conv5_wrw16_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
+                           int Nn, int Hi, int Wi, int Ho, int Wo, int CI,
+                           int CO, int n_wg, hipStream_t s) {
+  if (CI != 16) return -1;
+  const int W = (Wo + 31) >> 5;
+  const int need = (W * 32 + 4) > Wi ? (W * 32 + 4) : Wi;
+#define W2LAUNCH(COT_, PIX_)                                              \
+  if (need <= PIX_) {                                                     \
+    hipLaunchKernelGGL((k_conv5_wrw16_nhwc<COT_, PIX_>), dim3(n_wg),      \
+                       dim3(WRW2_THREADS), 0, s, in, gout, part, Nn, Hi,  \
+                       Wi, Ho, Wo);                                       \
+    return n_wg;                                                          \
+  }
+  if (CO == 32) { W2LAUNCH(2, 136) W2LAUNCH(2, 232) }
+  if (CO == 16) { W2LAUNCH(1, 136) W2LAUNCH(1, 232) }
+#undef W2LAUNCH
+  return -1;
+}
+
+void geops_tr16_probe(const bf16_t* in, bf16_t* out, hipStream_t s) {
+  hipLaunchKernelGGL(k_tr16_probe, dim3(1), dim3(64), 0, s, in, out);
+}
+
+}  // extern "C"

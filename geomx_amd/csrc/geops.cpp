@@ -61,6 +61,10 @@ void geops_pad_ch3to4_nhwc(const void*, unsigned short*, long long, int,
 int geops_conv5_wrw_nhwc(const unsigned short*, const unsigned short*,
                          float*, int, int, int, int, int, int, int, int,
                          hipStream_t);
+int geops_conv5_wrw16_nhwc(const unsigned short*, const unsigned short*,
+                           float*, int, int, int, int, int, int, int, int,
+                           hipStream_t);
+void geops_tr16_probe(const unsigned short*, unsigned short*, hipStream_t);
 }
 
 namespace {
@@ -268,11 +272,25 @@ void conv5_wrw_nhwc(torch::Tensor in, torch::Tensor gout, torch::Tensor part,
   TORCH_CHECK(in.scalar_type() == torch::kBFloat16 &&
               gout.scalar_type() == torch::kBFloat16 &&
               part.scalar_type() == torch::kFloat32);
-  const int rc = geops_conv5_wrw_nhwc(
-      (const unsigned short*)in.data_ptr(),
-      (const unsigned short*)gout.data_ptr(), part.data_ptr<float>(),
-      (int)N, (int)Hi, (int)Wi, (int)Ho, (int)Wo, (int)CI, (int)CO,
-      (int)n_wg, cur_stream());
+  // CI=16 shapes go to the v2 kernel (LDS-DMA staging + tr16 reads);
+  // GEOPS_WRW_V1=1 forces the v1 path for A/B runs
+  static const bool force_v1 = [] {
+    const char* e = getenv("GEOPS_WRW_V1");
+    return e && e[0] == '1';
+  }();
+  int rc = -1;
+  if (CI == 16 && !force_v1)
+    rc = geops_conv5_wrw16_nhwc(
+        (const unsigned short*)in.data_ptr(),
+        (const unsigned short*)gout.data_ptr(), part.data_ptr<float>(),
+        (int)N, (int)Hi, (int)Wi, (int)Ho, (int)Wo, (int)CI, (int)CO,
+        (int)n_wg, cur_stream());
+  if (rc < 0)
+    rc = geops_conv5_wrw_nhwc(
+        (const unsigned short*)in.data_ptr(),
+        (const unsigned short*)gout.data_ptr(), part.data_ptr<float>(),
+        (int)N, (int)Hi, (int)Wi, (int)Ho, (int)Wo, (int)CI, (int)CO,
+        (int)n_wg, cur_stream());
   TORCH_CHECK(rc > 0, "conv5_wrw_nhwc: unsupported geometry CI=", CI,
               " CO=", CO, " Wo=", Wo);
   launch_check("conv5_wrw_nhwc");
@@ -387,6 +405,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv5_nhwc", &conv5_nhwc);
   m.def("pad_ch3to4_nhwc", &pad_ch3to4_nhwc);
   m.def("conv5_wrw_nhwc", &conv5_wrw_nhwc);
+  m.def("tr16_probe", [](torch::Tensor in, torch::Tensor out) {
+    TORCH_CHECK(in.is_cuda() && out.is_cuda());
+    TORCH_CHECK(in.numel() == 64 && out.numel() == 256);
+    geops_tr16_probe((const unsigned short*)in.data_ptr(),
+                     (unsigned short*)out.data_ptr(), cur_stream());
+    launch_check("tr16_probe");
+  });
   m.def("relu_maxpool2_fwd", &relu_maxpool2_fwd);
   m.def("relu_maxpool2_bwd", &relu_maxpool2_bwd);
   m.def("sgd_update", &sgd_update);

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         "geomx_amd/csrc/kernels.hip",
         "geomx_amd/csrc/conv.hip",
         "geomx_amd/csrc/convwrw.hip",
+        "geomx_amd/csrc/convwrw2.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

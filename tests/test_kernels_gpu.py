@@ -514,3 +514,47 @@ def test_bsc_boundary_matches_cpu_reference():
     b_gpu = _bsc_boundary_gpu(v.to(DEV), ratio, 42)
     b_cpu = ref.bsc_boundary(v, ratio, 42)
     assert abs(b_gpu - b_cpu) < 1e-6, (b_gpu, b_cpu)
+
+
+# ---------------------------------------------------------------------------
+# wrw v2 building blocks
+# ---------------------------------------------------------------------------
+
+def test_tr16_lane_mapping():
+    """ds_read_b64_tr_b16 probe: lane g of each 16-lane group must
+    receive column g of the 4x16 row-major bf16 tile (the fragment
+    contract the wrw v2 kernel builds on)."""
+    from geomx_amd import _geops
+    tile = torch.arange(64, dtype=torch.float32).reshape(4, 16) \
+        .to(torch.bfloat16)
+    out = torch.empty(256, dtype=torch.bfloat16, device=DEV)
+    _geops.tr16_probe(tile.reshape(-1).to(DEV), out)
+    got = out.cpu().float().reshape(64, 4)
+    for lane in range(64):
+        col = lane & 15
+        expect = tile.float()[:, col]
+        assert torch.equal(got[lane], expect), (lane, got[lane], expect)
+
+
+@pytest.mark.parametrize("ci,co,hw", [(16, 32, 70), (16, 16, 112),
+                                      (16, 32, 110)])
+def test_conv5_wrw_v2_matches_aten(ci, co, hw):
+    """v2 kernel (CI=16 default path) vs ATen on conv2-like geometries,
+    including multi-window rows (W>1) and odd Ho tails."""
+    from geomx_amd.ops import conv as C
+    torch.manual_seed(42)
+    N = 4
+    x = torch.randn(N, ci, hw, hw, device=DEV, dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    go = torch.randn(N, co, hw - 4, hw - 4, device=DEV,
+                     dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    idx, t16 = C.build_wrw_unpack_index((co, ci, 5, 5))
+    dw = C.wrw_via_kernel(x, go, idx.to(DEV), t16, (co, ci, 5, 5))
+    w = torch.zeros(co, ci, 5, 5, device=DEV, dtype=torch.bfloat16)
+    _, dw_ref, _ = torch.ops.aten.convolution_backward(
+        go, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+        [False, True, False])
+    err = (dw - dw_ref.float()).abs().max().item()
+    scale = dw_ref.float().abs().max().item()
+    assert err < 0.02 * scale + 2.0, (err, scale)
