@@ -24,6 +24,9 @@ void geops_bsc_pack(const float*, float*, float*, float*, int*, long long*,
                     float, long long, long long, float, bool, hipStream_t);
 void geops_bsc_pull_pack(const float*, float*, int*, long long*, long long,
                          long long, float, hipStream_t);
+int geops_bsc_fused(const float*, float*, float*, float*, int*, const float*,
+                    unsigned long long*, float, long long, long long, float,
+                    hipStream_t);
 void geops_bsc_unpack(const float*, const int*, float*, long long, bool,
                       hipStream_t);
 void geops_dgt_contribution(const float*, float*, long long, int, int,
@@ -141,6 +144,26 @@ void bsc_pack(torch::Tensor v, torch::Tensor u, torch::Tensor vals,
                  (float)boundary, v.numel(), vals.numel(),
                  (float)placeholder, /*zero_uv=*/true, cur_stream());
   launch_check("bsc_pack");
+}
+
+bool bsc_compress_fused(torch::Tensor g, torch::Tensor u, torch::Tensor v,
+                        torch::Tensor vals, torch::Tensor idx,
+                        torch::Tensor boundary, double mu,
+                        double placeholder) {
+  check_f32(g, "g"); check_f32(u, "u"); check_f32(v, "v");
+  check_f32(vals, "vals"); check_f32(boundary, "boundary");
+  TORCH_CHECK(idx.scalar_type() == torch::kInt32 && idx.is_contiguous());
+  TORCH_CHECK(vals.numel() == idx.numel());
+  const long long n = g.numel();
+  TORCH_CHECK(u.numel() == n && v.numel() == n && boundary.numel() == 1);
+  auto ws = make_workspace(g);
+  const int rc = geops_bsc_fused(
+      g.data_ptr<float>(), u.data_ptr<float>(), v.data_ptr<float>(),
+      vals.data_ptr<float>(), idx.data_ptr<int32_t>(),
+      boundary.data_ptr<float>(), (unsigned long long*)ws.data_ptr<int64_t>(),
+      (float)mu, n, vals.numel(), (float)placeholder, cur_stream());
+  launch_check("bsc_compress_fused");
+  return rc == 0;
 }
 
 void bsc_pull_pack(torch::Tensor x, torch::Tensor vals, torch::Tensor idx,
@@ -398,6 +421,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bsc_momentum", &bsc_momentum);
   m.def("bsc_pack", &bsc_pack);
   m.def("bsc_pull_pack", &bsc_pull_pack);
+  m.def("bsc_compress_fused", &bsc_compress_fused);
   m.def("bsc_unpack", &bsc_unpack);
   m.def("dgt_contribution", &dgt_contribution);
   m.def("quantize_4bit", &quantize_4bit);

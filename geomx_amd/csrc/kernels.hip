@@ -293,6 +293,170 @@ __global__ void k_bsc_pack_t(const float* __restrict__ v_in,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused single-pass Bi-Sparse compress: momentum + count + ordered pack in
+// ONE sweep of (g,u,v) with a decoupled-lookback prefix over blocks
+// (replaces the momentum + count + scan + pack 4-launch chain: the extra
+// 4n-byte count read disappears; the boundary arrives as a DEVICE scalar
+// so the host never synchronizes). Per block: phase 1 streams its
+// contiguous chunk (float4), applies u=mu*u+g, v+=u, records the
+// |v|>=boundary predicate as an LDS bitmask; phase 2 publishes the block
+// aggregate and resolves the exclusive prefix via single-8B agent-scope
+// atomics ({status,count} in one granule — the R2 untorn-granule form, no
+// fences needed); phase 3 walks the bitmask, wave-scans positions, and
+// writes the capacity-bounded ordered pack, zeroing u,v at sent slots.
+// Deadlock-free by construction: the grid is capped at 4 blocks/CU x 256
+// CUs, so every block is resident and publishes its aggregate with no
+// dependence on any other block.
+// ---------------------------------------------------------------------------
+
+#define BSC_FUSE_CHUNK_MAX (128 * 1024)  // elements per block (16 KiB mask)
+#define BSC_FUSE_MAX_BLOCKS 1024         // 4 blocks/CU x 256 CUs: all resident
+
+extern "C" __global__ __launch_bounds__(GEOPS_THREADS, 4) void k_bsc_fused(
+    const float* __restrict__ g, float* __restrict__ u,
+    float* __restrict__ v, float* __restrict__ vals, int* __restrict__ idx,
+    const float* __restrict__ boundary_p,
+    unsigned long long* __restrict__ ws,  // [nb+1] lookback slots
+    float mu, long long n, long long capacity, int nb) {
+  __shared__ unsigned char mask[BSC_FUSE_CHUNK_MAX / 8];
+  __shared__ long long wave_base[GEOPS_THREADS / 64];
+  __shared__ long long carry_s;
+  const int b = blockIdx.x;
+  // 8-aligned contiguous chunks (float4 x2 per thread-iteration)
+  const long long chunk = (((n + nb - 1) / nb) + 7) & ~7LL;
+  long long lo = (long long)b * chunk;
+  if (lo > n) lo = n;
+  long long hi = lo + chunk;
+  if (hi > n) hi = n;
+  const float boundary = *boundary_p;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+  const long long span = (long long)blockDim.x * 8;
+
+  // ---- phase 1: momentum + predicate bitmask + block count ----------
+  long long cnt = 0;
+  for (long long base = lo; base < hi; base += span) {
+    const long long i0 = base + (long long)threadIdx.x * 8;
+    unsigned byte = 0;
+    if (i0 + 8 <= hi) {
+      const float4 g0 = *reinterpret_cast<const float4*>(g + i0);
+      const float4 g1 = *reinterpret_cast<const float4*>(g + i0 + 4);
+      float4 u0 = *reinterpret_cast<float4*>(u + i0);
+      float4 u1 = *reinterpret_cast<float4*>(u + i0 + 4);
+      float4 v0 = *reinterpret_cast<float4*>(v + i0);
+      float4 v1 = *reinterpret_cast<float4*>(v + i0 + 4);
+      u0.x = u0.x * mu + g0.x; v0.x += u0.x;
+      u0.y = u0.y * mu + g0.y; v0.y += u0.y;
+      u0.z = u0.z * mu + g0.z; v0.z += u0.z;
+      u0.w = u0.w * mu + g0.w; v0.w += u0.w;
+      u1.x = u1.x * mu + g1.x; v1.x += u1.x;
+      u1.y = u1.y * mu + g1.y; v1.y += u1.y;
+      u1.z = u1.z * mu + g1.z; v1.z += u1.z;
+      u1.w = u1.w * mu + g1.w; v1.w += u1.w;
+      *reinterpret_cast<float4*>(u + i0) = u0;
+      *reinterpret_cast<float4*>(u + i0 + 4) = u1;
+      *reinterpret_cast<float4*>(v + i0) = v0;
+      *reinterpret_cast<float4*>(v + i0 + 4) = v1;
+      byte |= (fabsf(v0.x) >= boundary) << 0;
+      byte |= (fabsf(v0.y) >= boundary) << 1;
+      byte |= (fabsf(v0.z) >= boundary) << 2;
+      byte |= (fabsf(v0.w) >= boundary) << 3;
+      byte |= (fabsf(v1.x) >= boundary) << 4;
+      byte |= (fabsf(v1.y) >= boundary) << 5;
+      byte |= (fabsf(v1.z) >= boundary) << 6;
+      byte |= (fabsf(v1.w) >= boundary) << 7;
+    } else {
+      for (int j = 0; j < 8; ++j) {
+        const long long i = i0 + j;
+        if (i < hi) {
+          const float uu = u[i] * mu + g[i];
+          u[i] = uu;
+          const float vv = v[i] + uu;
+          v[i] = vv;
+          byte |= (fabsf(vv) >= boundary) << j;
+        }
+      }
+    }
+    if (i0 < hi) mask[(i0 - lo) >> 3] = (unsigned char)byte;
+    cnt += __popc(byte);
+  }
+  // block reduce
+  for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, 64);
+  if (lane == 0) wave_base[wid] = cnt;
+  __syncthreads();
+
+  // ---- phase 2: publish aggregate, decoupled lookback ----------------
+  if (threadIdx.x == 0) {
+    long long agg = 0;
+    for (int w = 0; w < nwaves; ++w) agg += wave_base[w];
+    // status 1 = aggregate ready (bit 62), 2 = inclusive prefix ready
+    __hip_atomic_store(&ws[b], (1ULL << 62) | (unsigned long long)agg,
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    long long excl = 0;
+    for (int p = b - 1; p >= 0;) {
+      const unsigned long long w64 = __hip_atomic_load(
+          &ws[p], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      const unsigned st = (unsigned)(w64 >> 62);
+      if (st == 0) {
+        __builtin_amdgcn_s_sleep(2);
+        continue;
+      }
+      excl += (long long)(w64 & ((1ULL << 62) - 1));
+      if (st >= 2) break;
+      --p;
+    }
+    __hip_atomic_store(&ws[b],
+                       (2ULL << 62) | (unsigned long long)(excl + agg),
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (b == nb - 1) ws[nb] = (unsigned long long)(excl + agg);  // total
+    carry_s = excl;
+  }
+  __syncthreads();
+
+  // ---- phase 3: bitmask walk, wave-scan, ordered capacity-bound pack -
+  for (long long base = lo; base < hi; base += span) {
+    const long long i0 = base + (long long)threadIdx.x * 8;
+    const unsigned byte = (i0 < hi) ? mask[(i0 - lo) >> 3] : 0u;
+    int c = __popc(byte);
+    int incl = c;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      const int y = __shfl_up(incl, off, 64);
+      if (lane >= off) incl += y;
+    }
+    const int excl = incl - c;
+    if (lane == 63) wave_base[wid] = incl;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      long long acc = carry_s;
+      for (int w = 0; w < nwaves; ++w) {
+        const long long t = wave_base[w];
+        wave_base[w] = acc;
+        acc += t;
+      }
+      carry_s = acc;
+    }
+    __syncthreads();
+    long long pos = wave_base[wid] + excl;
+    unsigned bits = byte;
+    while (bits) {
+      const int j = __ffs(bits) - 1;
+      bits &= bits - 1;
+      if (pos < capacity) {
+        const long long i = i0 + j;
+        vals[pos] = v[i];
+        idx[pos] = (int)i;
+        v[i] = 0.0f;
+        u[i] = 0.0f;
+      }
+      ++pos;
+    }
+    __syncthreads();
+  }
+}
+
 // placeholder fill for unused capacity (reference :262-266)
 extern "C" __global__ void k_bsc_fill_tail(float* __restrict__ vals,
                                            int* __restrict__ idx,
@@ -827,6 +991,27 @@ static int bsc_nblocks(long long n) {
   if (nb < 1) nb = 1;
   if (nb > 1024) nb = 1024;
   return nb;
+}
+
+
+// fused single-pass path; returns -1 when the shape needs the multi-pass
+// chain (chunk would exceed the LDS bitmask)
+int geops_bsc_fused(const float* g, float* u, float* v, float* vals,
+                    int* idx, const float* boundary_p,
+                    unsigned long long* ws, float mu, long long n,
+                    long long capacity, float placeholder, hipStream_t s) {
+  int nb = (int)((n + 16383) / 16384);
+  if (nb < 1) nb = 1;
+  if (nb > BSC_FUSE_MAX_BLOCKS) nb = BSC_FUSE_MAX_BLOCKS;
+  const long long chunk = (((n + nb - 1) / nb) + 7) & ~7LL;
+  if (chunk > BSC_FUSE_CHUNK_MAX) return -1;
+  hipMemsetAsync(ws, 0, (nb + 1) * sizeof(unsigned long long), s);
+  hipLaunchKernelGGL(k_bsc_fused, dim3(nb), dim3(GEOPS_THREADS), 0, s, g,
+                     u, v, vals, idx, boundary_p, ws, mu, n, capacity, nb);
+  hipLaunchKernelGGL(k_bsc_fill_tail, dim3(geops_blocks(capacity)),
+                     dim3(GEOPS_THREADS), 0, s, vals, idx, (long long*)ws,
+                     nb, capacity, placeholder);
+  return 0;
 }
 
 // workspace: (nb+1) int64 entries

@@ -93,17 +93,50 @@ def bsc_compress(grad, u, v, ratio, momentum=ref.BSC_MOMENTUM, seed=42
         k = ref.bsc_capacity(n, ratio)
         vals = torch.empty(k, dtype=torch.float32, device=grad.device)
         idx = torch.empty(k, dtype=torch.int32, device=grad.device)
-        # pass 1: fused momentum correction (u = mu*u + g ; v += u)
-        g.bsc_momentum(grad.reshape(-1), u, v, momentum)
-        # boundary: sampled top-k over |v| (small torch.topk on the sample)
+        gflat = grad.reshape(-1)
+        # boundary PREDICTED from the seeded sample of post-momentum v
+        # (v + mu*u + g at the sample positions) as a DEVICE scalar, so
+        # the fused kernel path never synchronizes with the host
+        bt = _bsc_boundary_pred_gpu(gflat, u, v, ratio, momentum, seed)
+        if g.bsc_compress_fused(gflat, u, v, vals, idx, bt, momentum,
+                                ref.BSC_PLACEHOLDER):
+            return vals, idx
+        # huge-n fallback: momentum + count + scan + pack chain
+        g.bsc_momentum(gflat, u, v, momentum)
         boundary = _bsc_boundary_gpu(v, ratio, seed)
-        # pass 2: capacity-bounded select + pack + zero u,v at selected
         g.bsc_pack(v, u, vals, idx, boundary, ref.BSC_PLACEHOLDER)
         return vals, idx
     return ref.bsc_compress(grad, u, v, ratio, momentum, seed)
 
 
 _bsc_idx_cache: dict = {}
+
+
+def _bsc_sample_idx(n: int, sample_size: int, seed: int, device):
+    ck = (n, sample_size, seed, device)
+    idx = _bsc_idx_cache.get(ck)
+    if idx is None:
+        if len(_bsc_idx_cache) > 64:
+            _bsc_idx_cache.clear()
+        idx = ref.bsc_sample_indices(n, sample_size, seed).to(device)
+        _bsc_idx_cache[ck] = idx
+    return idx
+
+
+def _bsc_boundary_pred_gpu(g: torch.Tensor, u: torch.Tensor,
+                           v: torch.Tensor, ratio: float, momentum: float,
+                           seed: int) -> torch.Tensor:
+    """Sampled top-k boundary over the POST-momentum |v|, computed from
+    (g,u,v) BEFORE the fused kernel mutates them: same value the CPU
+    golden model derives after its momentum pass (same seeded sample,
+    same op order mu*u + g then + v). Returns a device fp32 scalar."""
+    n = v.numel()
+    sample_size = min(ref.bsc_sample_size(n, ratio), n)
+    top_k = max(1, int(sample_size * ratio))
+    si = _bsc_sample_idx(n, sample_size, seed, v.device)
+    sample = (v[si] + (u[si] * momentum + g[si])).abs()
+    k = min(top_k, sample.numel())
+    return torch.topk(sample, k).values[k - 1:k].contiguous()
 
 
 def _bsc_boundary_gpu(v: torch.Tensor, ratio: float, seed: int) -> float:

@@ -558,3 +558,52 @@ def test_conv5_wrw_v2_matches_aten(ci, co, hw):
     err = (dw - dw_ref.float()).abs().max().item()
     scale = dw_ref.float().abs().max().item()
     assert err < 0.02 * scale + 2.0, (err, scale)
+
+
+# ---------------------------------------------------------------------------
+# fused single-pass BSC (momentum+count+pack, decoupled lookback)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("n", [10_000, 1 << 20, (1 << 22) + 13])
+def test_bsc_fused_matches_cpu_reference(n):
+    """The production GPU path (ops.bsc_compress -> k_bsc_fused) vs the
+    CPU golden model over a 3-step error-feedback chain: same boundary
+    (shared seeded sample), same ordered pack, same u/v mutation."""
+    torch.manual_seed(7)
+    ratio = 0.01
+    u_g = torch.zeros(n, device=DEV)
+    v_g = torch.zeros(n, device=DEV)
+    u_c = torch.zeros(n)
+    v_c = torch.zeros(n)
+    for step in range(3):
+        g = torch.randn(n) * (step + 1)
+        vals_g, idx_g = ops.bsc_compress(g.to(DEV), u_g, v_g, ratio)
+        vals_c, idx_c = ref.bsc_compress(g, u_c, v_c, ratio)
+        sel_g = int((idx_g >= 0).sum())
+        sel_c = int((idx_c >= 0).sum())
+        # boundary ties can differ by a few ulp-equal elements at most
+        assert abs(sel_g - sel_c) <= max(2, sel_c // 1000), (sel_g, sel_c)
+        k = min(sel_g, sel_c)
+        assert torch.equal(idx_g[:k].cpu(), idx_c[:k]), step
+        assert torch.allclose(vals_g[:k].cpu(), vals_c[:k], atol=1e-5)
+        assert torch.allclose(u_g.cpu(), u_c, atol=1e-5)
+        assert torch.allclose(v_g.cpu(), v_c, atol=1e-5)
+
+
+def test_bsc_fused_capacity_bound():
+    n = 100_000
+    u = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    g = torch.full((n,), 5.0, device=DEV)  # everything selected
+    vals, idx = ops.bsc_compress(g, u, v, 0.01)
+    k = ref.bsc_capacity(n, 0.01)
+    assert vals.numel() == k
+    assert int((idx >= 0).sum()) == k
+    # ordered: first k indices ascending
+    ii = idx.cpu().long()
+    assert torch.all(ii[1:] > ii[:-1])
+    # u,v zeroed ONLY at sent positions
+    sent = torch.zeros(n, dtype=torch.bool)
+    sent[ii] = True
+    assert torch.all(v.cpu()[sent] == 0)
+    assert torch.all(v.cpu()[~sent] != 0)
