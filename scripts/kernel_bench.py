@@ -84,6 +84,18 @@ def main():
     sec = timeit(lambda: geops.bsc_unpack(vals, idx, out, False), args.iters)
     report("bsc_unpack(1%)", n * 4 + k * 8, sec)
 
+    # --- fused single-pass compress (momentum+count+pack), the
+    # production bsc path: read g,u,v + write u,v in ONE sweep
+    bt = torch.full((1,), 2.3, device=dev)
+    sec = timeit(lambda: geops.bsc_compress_fused(g, u, v, vals, idx, bt,
+                                                  0.9, -65530.0),
+                 args.iters)
+    report("bsc_fused(1%)", n * 4 * 5, sec)
+
+    import geomx_amd.ops as _ops
+    sec = timeit(lambda: _ops.bsc_compress(g, u, v, 0.01), args.iters)
+    report("bsc_compress e2e", n * 4 * 5, sec)
+
     sec = timeit(lambda: geops.bsc_pull_pack(out, vals, idx, -65530.0),
                  args.iters)
     report("bsc_pull_pack", n * 4 * 2, sec)
