@@ -54,7 +54,7 @@ def parse_args():
     p.add_argument("--parties", type=int, default=0,
                    help="HiPS party count (default: world_size//2, min 2)")
     p.add_argument("--compress", type=str, default=None,
-                   choices=[None, "bsc", "fp16", "mpq", "2bit", "dgt"])
+                   choices=[None, "bsc", "fp16", "mpq", "2bit", "dgt", "bsc_dgt"])
     p.add_argument("--bsc-ratio", type=float, default=0.01)
     p.add_argument("--wan-gbps", type=float, default=0.0)
     p.add_argument("--party-wan-gbps", type=str, default=None,

@@ -181,7 +181,7 @@ class Config:
     def validate(self):
         if self.mode not in ("dist_sync", "dist_async", "local"):
             raise ValueError(f"mode must be dist_sync|dist_async|local, got {self.mode}")
-        if self.compression not in (None, "2bit", "bsc", "fp16", "mpq", "dgt"):
+        if self.compression not in (None, "2bit", "bsc", "fp16", "mpq", "dgt", "bsc_dgt"):
             raise ValueError(f"unknown compression {self.compression!r}")
         if not (0 < self.bsc_ratio < 1):
             raise ValueError("bsc_ratio must be in (0,1)")

@@ -345,7 +345,49 @@ class GeoTrainer:
         ctype = self.cfg.compression
         for b in self.buckets:
             if topo.is_leader:
-                if ctype == "bsc" or (
+                if ctype == "bsc_dgt":
+                    # BASELINE config 5 composition: Bi-Sparse selects the
+                    # content, DGT's 4-bit tier rides the packed VALUES
+                    # (the reference's DGT chunks whatever bytes a push
+                    # carries, including BSC-compressed pushes —
+                    # kv_app.h:917-995); indices stay exact.
+                    if b.bsc_u is None:
+                        b.bsc_u = torch.zeros_like(b.flat)
+                        b.bsc_v = torch.zeros_like(b.flat)
+                    vals, idx = ops.bsc_compress(b.flat, b.bsc_u, b.bsc_v,
+                                                 self.cfg.bsc_ratio)
+                    if b.dgt is None or b.dgt.numel != vals.numel():
+                        from ..kvstore.dgt import DGTState
+                        b.dgt = DGTState(vals.numel(), vals.device,
+                                         chunk_elems=max(
+                                             64, self.cfg.dgt_block_size // 4),
+                                         k=self.cfg.dgt_k,
+                                         alpha=self.cfg.dgt_alpha, mode=3)
+                    # placeholder slots (unused capacity) would poison a
+                    # mixed chunk's min/max codebook: ship them as zeros
+                    # (the receiver skips idx<0 slots anyway). Positions
+                    # in the packed buffer mean different elements every
+                    # step, so DGT's positional residual is meaningless
+                    # here — clear it.
+                    vals = vals.masked_fill(idx < 0, 0.0)
+                    b.dgt.residual.zero_()
+                    payload = b.dgt.compress(vals) + (idx,)
+                    gathered = [[torch.empty_like(t) for _ in range(P)]
+                                for t in payload]
+                    for lst, t in zip(gathered, payload):
+                        comm.all_gather(lst, t, group=topo.leader_group)
+                    self.wan.charge(cross_party_bytes(
+                        "all_gather",
+                        b.dgt.wire_bytes() + idx.numel() * 4, P))
+                    acc = torch.zeros_like(b.flat)
+                    for p_ in range(P):
+                        v_p = b.dgt.decompress(
+                            *(lst[p_] for lst in gathered[:4]))
+                        ops.bsc_decompress(v_p, gathered[4][p_],
+                                           b.flat.numel(), out=acc,
+                                           accumulate=True)
+                    b.flat.copy_(acc)
+                elif ctype == "bsc" or (
                         ctype == "mpq" and b.flat.numel() >= self.cfg.size_lower_bound):
                     if b.bsc_u is None:
                         b.bsc_u = torch.zeros_like(b.flat)
