@@ -136,7 +136,9 @@ def _bsc_boundary_pred_gpu(g: torch.Tensor, u: torch.Tensor,
     si = _bsc_sample_idx(n, sample_size, seed, v.device)
     sample = (v[si] + (u[si] * momentum + g[si])).abs()
     k = min(top_k, sample.numel())
-    return torch.topk(sample, k).values[k - 1:k].contiguous()
+    # clone: a slice view is 4B-aligned at best; the kernel's
+    # uniform scalar load wants its own 16B-aligned storage
+    return torch.topk(sample, k).values[k - 1:k].clone()
 
 
 def _bsc_boundary_gpu(v: torch.Tensor, ratio: float, seed: int) -> float:

@@ -1,0 +1,38 @@
+#!/usr/bin/env python3
+"""Error-structure dump for the wrw v2 kernel vs ATen."""
+import torch
+from geomx_amd.ops import conv as C
+
+DEV = "cuda:0"
+torch.manual_seed(1)
+for (ci, co, hw) in [(16, 32, 40), (16, 16, 40)]:
+    N = 2
+    x = torch.randn(N, ci, hw, hw, device=DEV, dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    go = torch.randn(N, co, hw - 4, hw - 4, device=DEV,
+                     dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    idx, t16 = C.build_wrw_unpack_index((co, ci, 5, 5))
+    dw = C.wrw_via_kernel(x, go, idx.to(DEV), t16, (co, ci, 5, 5))
+    w = torch.zeros(co, ci, 5, 5, device=DEV, dtype=torch.bfloat16)
+    _, dwr, _ = torch.ops.aten.convolution_backward(
+        go, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+        [False, True, False])
+    dwr = dwr.float()
+    err = (dw - dwr).abs()
+    print(f"=== ci={ci} co={co} hw={hw}  scale={dwr.abs().max():.1f} "
+          f"maxerr={err.max():.3f} relerr={(err.max()/dwr.abs().max()):.4f}")
+    # error by (kh, kw)
+    e_khkw = err.amax(dim=(0, 1))
+    print("err by (kh,kw):")
+    for kh in range(5):
+        print("  " + " ".join(f"{e_khkw[kh, kw]:8.3f}" for kw in range(5)))
+    # error by o (rows=o)
+    e_o = err.amax(dim=(1, 2, 3))
+    print("err by o tile0:", [f"{v:.2f}" for v in e_o[:16:4]],
+          "tile1:", [f"{v:.2f}" for v in e_o[16::4]] if co > 16 else "-")
+    e_ci = err.amax(dim=(0, 2, 3))
+    print("err by ci:", [f"{v:.2f}" for v in e_ci[::4]])
+    # ratio structure: is dw a permutation/shift of ref?
+    c0 = torch.corrcoef(torch.stack([dw.reshape(-1), dwr.reshape(-1)]))[0, 1]
+    print("corr(dw, ref) =", float(c0))
