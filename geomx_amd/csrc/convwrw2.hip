@@ -242,6 +242,71 @@ __global__ void k_tr16_probe(const bf16_t* __restrict__ in,
   for (int j = 0; j < 4; ++j) out[lane * 4 + j] = d[j];
 }
 
+// debug: run block 0's staging for (n=0, ho0=0) and dump the raw LDS
+// bytes so tests can check the glds images independently of the MFMAs.
+template <int COT, int PIX>
+__global__ void k_wrw2_dump(const bf16_t* __restrict__ in,
+                            const bf16_t* __restrict__ gout,
+                            bf16_t* __restrict__ dump,
+                            int Nn, int Hi, int Wi, int Ho, int Wo) {
+  constexpr int CI = 16;
+  constexpr int CO = COT * 16;
+  constexpr int ROW_BYTES = PIX * CI * 2;
+  constexpr int NCH = (ROW_BYTES + 1023) / 1024;
+  constexpr int RPB = NCH * 1024;
+  constexpr int NROW_IN = WRW2_R + 4;
+  constexpr int NROW_GO = WRW2_R * COT;
+  __shared__ __attribute__((aligned(128))) char lds_all[(NROW_IN + NROW_GO) *
+                                                        RPB];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nrows = Ho < WRW2_R ? Ho : WRW2_R;
+  const int irows = nrows + 4;
+  {
+    const int nchunks_a = irows * NCH;
+    for (int t = wid; t < nchunks_a; t += 4) {
+      const int ir = t / NCH;
+      const int c = t - ir * NCH;
+      const int slot = c * 64 + lane;
+      const int pix = slot >> 1;
+      const bf16_t* src = (pix < Wi)
+          ? in + ((long long)ir * Wi * CI + (long long)slot * 8)
+          : g_wrw2_zeros;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lds_all + ir * RPB +
+                                                    c * 1024),
+          16, 0, 0);
+    }
+    const int nchunks_b = nrows * COT * NCH;
+    for (int t = wid; t < nchunks_b; t += 4) {
+      const int rr = t / (COT * NCH);
+      const int rem = t - rr * (COT * NCH);
+      const int ot = rem / NCH;
+      const int c = rem - ot * NCH;
+      const int slot = c * 64 + lane;
+      const int pix = slot >> 1;
+      const int half = slot & 1;
+      const bf16_t* src = (pix < Wo)
+          ? gout + ((long long)rr * Wo * CO + (long long)pix * CO +
+                    ot * 16 + half * 8)
+          : g_wrw2_zeros;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lds_all +
+                                                    (NROW_IN + rr * COT +
+                                                     ot) * RPB + c * 1024),
+          16, 0, 0);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  const bf16_t* l16 = (const bf16_t*)lds_all;
+  for (int i = threadIdx.x; i < (NROW_IN + NROW_GO) * RPB / 2;
+       i += blockDim.x)
+    dump[i] = l16[i];
+}
+
 extern "C" {
 
 int geops_conv5_wrw16_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
@@ -265,6 +330,19 @@ int geops_conv5_wrw16_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
 
 void geops_tr16_probe(const bf16_t* in, bf16_t* out, hipStream_t s) {
   hipLaunchKernelGGL(k_tr16_probe, dim3(1), dim3(64), 0, s, in, out);
+}
+
+// dump size in bf16 elements for the (COT,PIX) variant, or -1
+int geops_wrw2_dump(const bf16_t* in, const bf16_t* gout, bf16_t* dump,
+                    int Nn, int Hi, int Wi, int Ho, int Wo, int CO,
+                    hipStream_t s) {
+  if (CO == 32) {
+    constexpr int RPB = ((136 * 16 * 2 + 1023) / 1024) * 1024;
+    hipLaunchKernelGGL((k_wrw2_dump<2, 136>), dim3(1), dim3(WRW2_THREADS),
+                       0, s, in, gout, dump, Nn, Hi, Wi, Ho, Wo);
+    return (6 + 4) * RPB / 2;
+  }
+  return -1;
 }
 
 }  // extern "C"

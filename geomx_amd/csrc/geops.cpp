@@ -68,6 +68,9 @@ int geops_conv5_wrw16_nhwc(const unsigned short*, const unsigned short*,
                            float*, int, int, int, int, int, int, int, int,
                            hipStream_t);
 void geops_tr16_probe(const unsigned short*, unsigned short*, hipStream_t);
+int geops_wrw2_dump(const unsigned short*, const unsigned short*,
+                    unsigned short*, int, int, int, int, int, int,
+                    hipStream_t);
 }
 
 namespace {
@@ -429,6 +432,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv5_nhwc", &conv5_nhwc);
   m.def("pad_ch3to4_nhwc", &pad_ch3to4_nhwc);
   m.def("conv5_wrw_nhwc", &conv5_wrw_nhwc);
+  m.def("wrw2_dump", [](torch::Tensor in, torch::Tensor gout,
+                        torch::Tensor dump, int64_t N, int64_t Hi,
+                        int64_t Wi, int64_t Ho, int64_t Wo, int64_t CO) {
+    const int sz = geops_wrw2_dump(
+        (const unsigned short*)in.data_ptr(),
+        (const unsigned short*)gout.data_ptr(),
+        (unsigned short*)dump.data_ptr(), (int)N, (int)Hi, (int)Wi,
+        (int)Ho, (int)Wo, (int)CO, cur_stream());
+    launch_check("wrw2_dump");
+    return (int64_t)sz;
+  });
   m.def("tr16_probe", [](torch::Tensor in, torch::Tensor out) {
     TORCH_CHECK(in.is_cuda() && out.is_cuda());
     TORCH_CHECK(in.numel() == 64 && out.numel() == 256);

@@ -36,3 +36,47 @@ for (ci, co, hw) in [(16, 32, 40), (16, 16, 40)]:
     # ratio structure: is dw a permutation/shift of ref?
     c0 = torch.corrcoef(torch.stack([dw.reshape(-1), dwr.reshape(-1)]))[0, 1]
     print("corr(dw, ref) =", float(c0))
+
+# ---- staging dump check ----------------------------------------------
+from geomx_amd import _geops
+ci, co, hw = 16, 32, 40
+Wi = Hi = hw; Wo = Ho = hw - 4
+x = torch.randn(1, ci, Hi, Wi, device=DEV, dtype=torch.bfloat16) \
+    .to(memory_format=torch.channels_last)
+go = torch.randn(1, co, Ho, Wo, device=DEV, dtype=torch.bfloat16) \
+    .to(memory_format=torch.channels_last)
+RPB = ((136 * 16 * 2 + 1023) // 1024) * 1024  # 5120 bytes
+dump = torch.zeros((6 + 4) * RPB // 2, dtype=torch.bfloat16, device=DEV)
+sz = _geops.wrw2_dump(x, go, dump, 1, Hi, Wi, Ho, Wo, co)
+d = dump.cpu().float()
+xl = x.permute(0, 2, 3, 1).cpu().float()   # [n][h][w][ci]
+gl = go.permute(0, 2, 3, 1).cpu().float()  # [n][h][w][o]
+RP = RPB // 2  # elements per region
+ok_a = True
+for ir in range(6):
+    img = d[ir * RP:(ir + 1) * RP]
+    for pix in range(0, Wi, 7):
+        got = img[pix * 16:(pix + 1) * 16]
+        exp = xl[0, ir, pix]
+        if not torch.allclose(got, exp):
+            print("A-image mismatch ir", ir, "pix", pix, got[:4], exp[:4])
+            ok_a = False
+            break
+    pad = img[Wi * 16: (RPB // 32) * 16]
+    if pad.abs().max() > 0:
+        print("A pad nonzero at ir", ir, float(pad.abs().max()))
+        ok_a = False
+print("A image ok:", ok_a)
+ok_b = True
+for rr in range(2):
+    for ot in range(2):
+        img = d[(6 + rr * 2 + ot) * RP:(6 + rr * 2 + ot + 1) * RP]
+        for pix in range(0, Wo, 5):
+            got = img[pix * 16:(pix + 1) * 16]
+            exp = gl[0, rr, pix, ot * 16:(ot + 1) * 16]
+            if not torch.allclose(got, exp):
+                print("B mismatch rr", rr, "ot", ot, "pix", pix,
+                      got[:4], exp[:4])
+                ok_b = False
+                break
+print("B image ok:", ok_b)
