@@ -1,5 +1,7 @@
 #!/usr/bin/env python3
 """Error-structure dump for the wrw v2 kernel vs ATen."""
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from geomx_amd.ops import conv as C
 
