@@ -154,32 +154,66 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw16_nhwc(
     __syncthreads();
 
     for (int r = 0; r < nrows; ++r) {
+      // A-row base addresses for this wave's 7 tap tiles (tt clamped so
+      // the unused 7th tile of waves 1-3 still reads in-bounds LDS; its
+      // MFMA is skipped)
+      unsigned abase[MAXT];
+#pragma unroll
+      for (int j = 0; j < MAXT; ++j) {
+        int tt = wid + 4 * j;
+        if (tt >= NT) tt = NT - 1;
+        const int kh = tt / 5;
+        const int kw = tt - kh * 5;
+        abase[j] = lds0 + (r + kh) * RPB + (kw + 4 * q) * 32 + m * 8;
+      }
+      const unsigned bbase0 = lds0 + (NROW_IN + r * COT + 0) * RPB +
+                              (4 * q) * 32 + m * 8;
+      const unsigned bbase1 = lds0 + (NROW_IN + r * COT + (COT - 1)) * RPB +
+                              (4 * q) * 32 + m * 8;
       for (int w = 0; w < W; ++w) {
-        const int pix0 = w * 32;
-        // B fragments (shared across this wave's tap tiles): per ot,
-        // two tr reads = pixel tiles [pix0+4q, +4) and [pix0+16+4q, +4)
-        bf16x4 bfr[COT][2];
-#pragma unroll
-        for (int ot = 0; ot < COT; ++ot) {
-          const unsigned bb = lds0 + (NROW_IN + r * COT + ot) * RPB +
-                              (pix0 + 4 * q) * 32 + m * 8;
-          bfr[ot][0] = tr16_b64(bb);
-          bfr[ot][1] = tr16_b64(bb + 16 * 32);
-        }
-        // A fragments + MFMAs per owned tap tile
+        const unsigned poff = (unsigned)(w * 32 * 32);  // pix0 bytes
+        // ONE asm block issues every tr read for this (r,w) and ends
+        // with the lgkmcnt drain: the compiler materialises the
+        // bf16x4->bf16x8 repack VALU at an asm OUTPUT's def site, so
+        // any multi-asm structure lets it consume un-landed tr
+        // destinations BEFORE a separate wait (observed: v_lshrrev/
+        // v_perm ahead of s_waitcnt -> corrupt fragments). Outputs are
+        // early-clobber so no read's destination aliases a later-used
+        // address register.
+        bf16x4 bfr[2][2];
         bf16x4 afr[MAXT][2];
-#pragma unroll
-        for (int j = 0; j < MAXT; ++j) {
-          const int tt = wid + 4 * j;
-          if (tt >= NT) continue;
-          const int kh = tt / 5;
-          const int kw = tt - kh * 5;
-          const unsigned ab = lds0 + (r + kh) * RPB +
-                              (pix0 + kw + 4 * q) * 32 + m * 8;
-          afr[j][0] = tr16_b64(ab);
-          afr[j][1] = tr16_b64(ab + 16 * 32);
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %18\n\t"
+            "ds_read_b64_tr_b16 %1, %18 offset:512\n\t"
+            "ds_read_b64_tr_b16 %2, %19\n\t"
+            "ds_read_b64_tr_b16 %3, %19 offset:512\n\t"
+            "ds_read_b64_tr_b16 %4, %20\n\t"
+            "ds_read_b64_tr_b16 %5, %20 offset:512\n\t"
+            "ds_read_b64_tr_b16 %6, %21\n\t"
+            "ds_read_b64_tr_b16 %7, %21 offset:512\n\t"
+            "ds_read_b64_tr_b16 %8, %22\n\t"
+            "ds_read_b64_tr_b16 %9, %22 offset:512\n\t"
+            "ds_read_b64_tr_b16 %10, %23\n\t"
+            "ds_read_b64_tr_b16 %11, %23 offset:512\n\t"
+            "ds_read_b64_tr_b16 %12, %24\n\t"
+            "ds_read_b64_tr_b16 %13, %24 offset:512\n\t"
+            "ds_read_b64_tr_b16 %14, %25\n\t"
+            "ds_read_b64_tr_b16 %15, %25 offset:512\n\t"
+            "ds_read_b64_tr_b16 %16, %26\n\t"
+            "ds_read_b64_tr_b16 %17, %26 offset:512\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(bfr[0][0]), "=&v"(bfr[0][1]), "=&v"(bfr[1][0]),
+              "=&v"(bfr[1][1]), "=&v"(afr[0][0]), "=&v"(afr[0][1]),
+              "=&v"(afr[1][0]), "=&v"(afr[1][1]), "=&v"(afr[2][0]),
+              "=&v"(afr[2][1]), "=&v"(afr[3][0]), "=&v"(afr[3][1]),
+              "=&v"(afr[4][0]), "=&v"(afr[4][1]), "=&v"(afr[5][0]),
+              "=&v"(afr[5][1]), "=&v"(afr[6][0]), "=&v"(afr[6][1])
+            : "v"(bbase0 + poff), "v"(bbase1 + poff),
+              "v"(abase[0] + poff), "v"(abase[1] + poff),
+              "v"(abase[2] + poff), "v"(abase[3] + poff),
+              "v"(abase[4] + poff), "v"(abase[5] + poff),
+              "v"(abase[6] + poff)
+            : "memory");
         __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
         for (int j = 0; j < MAXT; ++j) {
@@ -235,9 +269,9 @@ __global__ void k_tr16_probe(const bf16_t* __restrict__ in,
   tile[lane] = in[lane];
   __syncthreads();
   const unsigned a = lds_addr(tile) + (lane & 15) * 8;
-  bf16x4 d = tr16_b64(a);
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_sched_barrier(0);
+  bf16x4 d;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=&v"(d) : "v"(a) : "memory");
 #pragma unroll
   for (int j = 0; j < 4; ++j) out[lane * 4 + j] = d[j];
 }
