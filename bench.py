@@ -23,10 +23,10 @@ import os
 import sys
 import time
 
-# fast MIOpen find: measured identical step perf to the exhaustive
-# default (75.0k vs 74.3k samples/s) with far cheaper warmup — matters
-# when 8 ranks run find concurrently on the same node
-os.environ.setdefault("MIOPEN_FIND_MODE", "3")
+# MIOpen find: NORMAL find (1). The fast mode (3) is cheaper at warmup
+# but non-robust: in r02 it picked a 2.5x slower conv1 wrw igemm
+# (1.61 vs 0.64 ms/step). Override via env for quick smokes.
+os.environ.setdefault("MIOPEN_FIND_MODE", "1")
 
 import torch
 import torch.distributed as distmod

@@ -303,13 +303,11 @@ class GeoConv5(torch.nn.Conv2d):
                 and self.kernel_size == (5, 5) and self.groups == 1)
 
     # split-backward ATen path on non-custom shapes: dgrad by ATen,
-    # weight grad by the custom wrw kernel. Off by default: the custom
-    # wrw wins standalone (1.10 vs 1.62 ms) but the split costs more
-    # elsewhere in the full step (bench 69.9k vs 74.4k samples/s) —
-    # see ROADMAP item 1.
-    # opt-in via env for A/B runs (GEOPS_SPLIT_BWD=1)
+    # weight grad by the custom wrw v2 kernel (0.208 ms vs MIOpen's
+    # 1.61 ms igemm on conv2, r02 steady profile). DEFAULT ON since
+    # wrw v2; GEOPS_SPLIT_BWD=0 reverts to the fused ATen backward
     SPLIT_BACKWARD = __import__("os").environ.get(
-        "GEOPS_SPLIT_BWD", "0") == "1"
+        "GEOPS_SPLIT_BWD", "1") == "1"
 
     def forward(self, x):
         if not self._eligible(x):
