@@ -276,6 +276,176 @@ __global__ void k_tr16_probe(const bf16_t* __restrict__ in,
   for (int j = 0; j < 4; ++j) out[lane * 4 + j] = d[j];
 }
 
+// ---------------------------------------------------------------------
+// CI=4 variant (conv1: 4 padded channels -> 16 outputs).
+//
+// The CI=16 scheme needs 16-wide image rows so that a tr-read's j step
+// is ONE k-pixel on both operands. With 4 channels the natural image
+// rows are 4-wide, so instead we stage a KH-INTERLEAVED image
+//     I_p[pix][kho][ci] = in[row p+kho][pix][ci]   (kho = 0..3)
+// whose 16-wide rows make lane m = (kho*4+ci) the channel column and
+// keep the 1-pixel j step. Tap tiles become t = kw*2 + khg
+// (kw = 0..4, khg = 0..1, kh = khg*4+kho): 10 tiles of 16 slots hold
+// the 100 real taps (kh>4 slots are dead weight, 38%; their values are
+// whatever the clamped staging rows hold and are dropped at unpack).
+// Four interleaved images (p = r + 4*khg for r in 0..1, khg in 0..1)
+// are filled by 4-byte LDS-DMA granules (dest byte 4l maps to
+// pix = l>>3, kho = (l>>1)&3, ci-pair = l&1 — per-lane global source
+// does the interleave, dest stays lane-linear).
+// ---------------------------------------------------------------------
+
+template <int PIX>
+__global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
+    const bf16_t* __restrict__ in,    // [N][Hi][Wi][4]
+    const bf16_t* __restrict__ gout,  // [N][Ho][Wo][16]
+    float* __restrict__ part,         // [nWG][160][16]
+    int Nn, int Hi, int Wi, int Ho, int Wo) {
+  constexpr int CI = 4;
+  constexpr int CO = 16;
+  constexpr int NT = 10;                  // (kw 0..4) x (khg 0..1)
+  constexpr int T16 = NT * 16;
+  constexpr int ROW_BYTES = PIX * 16 * 2;  // one 16-wide image
+  constexpr int NCH4 = ROW_BYTES / 256;    // 4B-granule chunks per image
+  constexpr int NCHB = (ROW_BYTES + 1023) / 1024;
+  constexpr int RPB = NCHB * 1024;
+  constexpr int NIMG = 4;                 // I_{r+4*khg}: p in {0,1,4,5}
+  __shared__ __attribute__((aligned(128))) char lds_all[(NIMG + WRW2_R) *
+                                                        RPB];
+
+  const int lane = threadIdx.x & 63;
+  const int q = lane >> 4;
+  const int m = lane & 15;
+  const int wid = threadIdx.x >> 6;
+
+  const int W = (Wo + 31) >> 5;
+  const int blocks_h = (Ho + WRW2_R - 1) / WRW2_R;
+  const long long n_blocks = (long long)Nn * blocks_h;
+  const int wg = blockIdx.x;
+  const int n_wg = gridDim.x;
+  const unsigned lds0 = lds_addr(lds_all);
+
+  constexpr int MAXT = 3;                 // 10 tiles over 4 waves
+  f32x4 acc[MAXT];
+#pragma unroll
+  for (int j = 0; j < MAXT; ++j) acc[j] = (f32x4)0.0f;
+
+  static constexpr int P_OF[NIMG] = {0, 1, 4, 5};
+
+  for (long long blk = wg; blk < n_blocks; blk += n_wg) {
+    const int bh = (int)(blk % blocks_h);
+    const long long n = blk / blocks_h;
+    const int ho0 = bh * WRW2_R;
+    const int nrows = (Ho - ho0) < WRW2_R ? (Ho - ho0) : WRW2_R;
+
+    // ---- stage the 4 interleaved input images (4B LDS-DMA granules)
+    {
+      const int nunits = NIMG * NCH4;   // 256B chunks
+      for (int t = wid; t < nunits; t += 4) {
+        const int im = t / NCH4;
+        const int c = t - im * NCH4;
+        const int b4 = c * 256 + lane * 4;          // dest byte
+        const int el = b4 >> 1;                     // dest element
+        const int pix = el >> 4;
+        const int kho = (el >> 2) & 3;
+        const int ci2 = el & 3;                     // 0 or 2
+        int row = ho0 + P_OF[im] + kho;
+        if (row >= Hi) row = Hi - 1;                // dead-tap rows only
+        const bf16_t* src =
+            (pix < Wi) ? in + ((n * Hi + row) * (long long)Wi * CI +
+                               (long long)pix * CI + ci2)
+                       : g_wrw2_zeros;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)src,
+            (__attribute__((address_space(3))) void*)(lds_all + im * RPB +
+                                                      c * 256),
+            4, 0, 0);
+      }
+      // gout rows: [pix][16], straight 16B granules
+      const int nchunks_b = nrows * NCHB;
+      for (int t = wid; t < nchunks_b; t += 4) {
+        const int rr = t / NCHB;
+        const int c = t - rr * NCHB;
+        const int slot = c * 64 + lane;
+        const int pix = slot >> 1;
+        const bf16_t* src =
+            (pix < Wo) ? gout + ((n * Ho + (ho0 + rr)) * (long long)Wo * CO +
+                                 (long long)slot * 8)
+                       : g_wrw2_zeros;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)src,
+            (__attribute__((address_space(3))) void*)(lds_all +
+                                                      (NIMG + rr) * RPB +
+                                                      c * 1024),
+            16, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    for (int r = 0; r < nrows; ++r) {
+      unsigned abase[MAXT];
+#pragma unroll
+      for (int j = 0; j < MAXT; ++j) {
+        int t = wid + 4 * j;
+        if (t >= NT) t = NT - 1;
+        const int kw = t / 2;
+        const int khg = t - kw * 2;
+        const int im = r + 2 * khg;      // index into P_OF ({0,1,4,5})
+        abase[j] = lds0 + im * RPB + (kw + 4 * q) * 32 + m * 8;
+      }
+      const unsigned bbase = lds0 + (NIMG + r) * RPB + (4 * q) * 32 + m * 8;
+      for (int w = 0; w < W; ++w) {
+        const unsigned poff = (unsigned)(w * 32 * 32);
+        bf16x4 bfr[2];
+        bf16x4 afr[MAXT][2];
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %8\n\t"
+            "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+            "ds_read_b64_tr_b16 %2, %9\n\t"
+            "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"
+            "ds_read_b64_tr_b16 %4, %10\n\t"
+            "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"
+            "ds_read_b64_tr_b16 %6, %11\n\t"
+            "ds_read_b64_tr_b16 %7, %11 offset:512\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(bfr[0]), "=&v"(bfr[1]), "=&v"(afr[0][0]),
+              "=&v"(afr[0][1]), "=&v"(afr[1][0]), "=&v"(afr[1][1]),
+              "=&v"(afr[2][0]), "=&v"(afr[2][1])
+            : "v"(bbase + poff), "v"(abase[0] + poff),
+              "v"(abase[1] + poff), "v"(abase[2] + poff)
+            : "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        union { struct { bf16x4 lo, hi; } p; bf16x8 v; } b;
+        b.p.lo = bfr[0];
+        b.p.hi = bfr[1];
+#pragma unroll
+        for (int j = 0; j < MAXT; ++j) {
+          if (wid + 4 * j >= NT) continue;
+          union { struct { bf16x4 lo, hi; } p; bf16x8 v; } a;
+          a.p.lo = afr[j][0];
+          a.p.hi = afr[j][1];
+          acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v,
+                                                           acc[j], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- flush: part[wg][t*16 + m][o], D row = q*4+i = slot m, col = o
+  float* base = part + (long long)wg * T16 * CO;
+#pragma unroll
+  for (int j = 0; j < MAXT; ++j) {
+    const int t = wid + 4 * j;
+    if (t >= NT) continue;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int slot = t * 16 + q * 4 + i;
+      base[(long long)slot * CO + m] = acc[j][i];
+    }
+  }
+}
+
 // debug: run block 0's staging for (n=0, ho0=0) and dump the raw LDS
 // bytes so tests can check the glds images independently of the MFMAs.
 template <int COT, int PIX>
@@ -359,6 +529,24 @@ int geops_conv5_wrw16_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
   if (CO == 32) { W2LAUNCH(2, 136) W2LAUNCH(2, 232) }
   if (CO == 16) { W2LAUNCH(1, 136) W2LAUNCH(1, 232) }
 #undef W2LAUNCH
+  return -1;
+}
+
+int geops_conv5_wrw4_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
+                          int Nn, int Hi, int Wi, int Ho, int Wo, int CI,
+                          int CO, int n_wg, hipStream_t s) {
+  if (CI != 4 || CO != 16) return -1;
+  const int W = (Wo + 31) >> 5;
+  const int need = (W * 32 + 4) > Wi ? (W * 32 + 4) : Wi;
+#define W4LAUNCH(PIX_)                                                    \
+  if (need <= PIX_) {                                                     \
+    hipLaunchKernelGGL((k_conv5_wrw4_nhwc<PIX_>), dim3(n_wg),             \
+                       dim3(WRW2_THREADS), 0, s, in, gout, part, Nn, Hi,  \
+                       Wi, Ho, Wo);                                       \
+    return n_wg;                                                          \
+  }
+  W4LAUNCH(136) W4LAUNCH(232)
+#undef W4LAUNCH
   return -1;
 }
 

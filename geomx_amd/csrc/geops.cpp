@@ -67,6 +67,9 @@ int geops_conv5_wrw_nhwc(const unsigned short*, const unsigned short*,
 int geops_conv5_wrw16_nhwc(const unsigned short*, const unsigned short*,
                            float*, int, int, int, int, int, int, int, int,
                            hipStream_t);
+int geops_conv5_wrw4_nhwc(const unsigned short*, const unsigned short*,
+                          float*, int, int, int, int, int, int, int, int,
+                          hipStream_t);
 void geops_tr16_probe(const unsigned short*, unsigned short*, hipStream_t);
 int geops_wrw2_dump(const unsigned short*, const unsigned short*,
                     unsigned short*, int, int, int, int, int, int,
@@ -305,6 +308,12 @@ void conv5_wrw_nhwc(torch::Tensor in, torch::Tensor gout, torch::Tensor part,
     return e && e[0] == '1';
   }();
   int rc = -1;
+  if (CI == 4 && CO == 16 && !force_v1)
+    rc = geops_conv5_wrw4_nhwc(
+        (const unsigned short*)in.data_ptr(),
+        (const unsigned short*)gout.data_ptr(), part.data_ptr<float>(),
+        (int)N, (int)Hi, (int)Wi, (int)Ho, (int)Wo, (int)CI, (int)CO,
+        (int)n_wg, cur_stream());
   if (CI == 16 && !force_v1)
     rc = geops_conv5_wrw16_nhwc(
         (const unsigned short*)in.data_ptr(),

@@ -99,17 +99,33 @@ _WRW_NWG = 768  # partial slabs (3 workgroups per CU)
 # opt-in via env for A/B runs (GEOPS_WRW=1); module default stays off
 # because v1 loses net in-step (see docs/kernels.md wrw v2 notes)
 import os as _os
-WRW_ENABLED = _os.environ.get("GEOPS_WRW", "0") == "1"
+# r02: default ON — conv1's wrw runs the CI=4 kh-interleaved v2 kernel
+# (MIOpen's igemm took 1.61 ms/step for it); GEOPS_WRW=0 reverts to ATen
+WRW_ENABLED = _os.environ.get("GEOPS_WRW", "1") == "1"
 
 
 def build_wrw_unpack_index(weight_shape) -> torch.Tensor:
-    """Gather index from the wrw kernel's [T16][CO] tap-major layout
-    (tap = (kh*5+kw)*CI + ci, CI padded to mult of 4) back to OIHW."""
+    """Gather index from the wrw kernel slab [T16][CO] back to OIHW.
+
+    CI=16 (v2): tap = (kh*5+kw)*16 + ci.
+    CI=4  (v2-4, kh-interleaved): slot = (kw*2 + kh//4)*16 + (kh%4)*4+ci
+    (38% of the 160 slots are dead kh>4 padding, dropped here).
+    CI=8.. other: v1 layout tap = (kh*5+kw)*CI + ci.
+    """
     CO, CIr, KH, KW = weight_shape
     CI = (CIr + 3) & ~3
+    idx = torch.empty(CO, CIr, KH, KW, dtype=torch.int64)
+    if CI == 4 and CO == 16:
+        T16 = 160
+        for o in range(CO):
+            for ci in range(CIr):
+                for kh in range(KH):
+                    for kw in range(KW):
+                        slot = (kw * 2 + kh // 4) * 16 + (kh % 4) * 4 + ci
+                        idx[o, ci, kh, kw] = slot * CO + o
+        return idx.reshape(-1), T16
     NT = (25 * CI + 15) // 16
     T16 = NT * 16
-    idx = torch.empty(CO, CIr, KH, KW, dtype=torch.int64)
     for o in range(CO):
         for ci in range(CIr):
             for kh in range(KH):
