@@ -222,7 +222,11 @@ class _Conv5Fn(torch.autograd.Function):
                 grad_w = wrw_via_kernel(xb_pad, go, unpack_idx, T16,
                                         weight.shape).to(weight.dtype)
                 if ctx.has_bias:
-                    grad_b = go.float().sum(dim=(0, 2, 3)).to(weight.dtype)
+                    # dtype= fuses the fp32 accumulation into the
+                    # reduce (no materialized fp32 copy of go)
+                    grad_b = go.sum(dim=(0, 2, 3),
+                                    dtype=torch.float32) \
+                        .to(weight.dtype)
             else:
                 gi, gw, gb = torch.ops.aten.convolution_backward(
                     go, x, weight.to(torch.bfloat16),
