@@ -394,40 +394,53 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
         abase[j] = lds0 + im * RPB + (kw + 4 * q) * 32 + m * 8;
       }
       const unsigned bbase = lds0 + (NIMG + r) * RPB + (4 * q) * 32 + m * 8;
+#define WRW4_ISSUE(FR, POFF)                                              \
+        asm volatile(                                                     \
+            "ds_read_b64_tr_b16 %0, %8\n\t"                               \
+            "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"                    \
+            "ds_read_b64_tr_b16 %2, %9\n\t"                               \
+            "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"                    \
+            "ds_read_b64_tr_b16 %4, %10\n\t"                              \
+            "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"                   \
+            "ds_read_b64_tr_b16 %6, %11\n\t"                              \
+            "ds_read_b64_tr_b16 %7, %11 offset:512"                       \
+            : "=&v"(FR[0]), "=&v"(FR[1]), "=&v"(FR[2]), "=&v"(FR[3]),     \
+              "=&v"(FR[4]), "=&v"(FR[5]), "=&v"(FR[6]), "=&v"(FR[7])      \
+            : "v"(bbase + (POFF)), "v"(abase[0] + (POFF)),                \
+              "v"(abase[1] + (POFF)), "v"(abase[2] + (POFF))              \
+            : "memory")
+#define WRW4_LAND(FR, CNT)                                                \
+        asm volatile("s_waitcnt lgkmcnt(" #CNT ")"                        \
+            : "+v"(FR[0]), "+v"(FR[1]), "+v"(FR[2]), "+v"(FR[3]),         \
+              "+v"(FR[4]), "+v"(FR[5]), "+v"(FR[6]), "+v"(FR[7])          \
+            :: "memory")
+      bf16x4 frA[8], frB[8];
+      WRW4_ISSUE(frA, 0u);
       for (int w = 0; w < W; ++w) {
-        const unsigned poff = (unsigned)(w * 32 * 32);
-        bf16x4 bfr[2];
-        bf16x4 afr[MAXT][2];
-        asm volatile(
-            "ds_read_b64_tr_b16 %0, %8\n\t"
-            "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
-            "ds_read_b64_tr_b16 %2, %9\n\t"
-            "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"
-            "ds_read_b64_tr_b16 %4, %10\n\t"
-            "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"
-            "ds_read_b64_tr_b16 %6, %11\n\t"
-            "ds_read_b64_tr_b16 %7, %11 offset:512\n\t"
-            "s_waitcnt lgkmcnt(0)"
-            : "=&v"(bfr[0]), "=&v"(bfr[1]), "=&v"(afr[0][0]),
-              "=&v"(afr[0][1]), "=&v"(afr[1][0]), "=&v"(afr[1][1]),
-              "=&v"(afr[2][0]), "=&v"(afr[2][1])
-            : "v"(bbase + poff), "v"(abase[0] + poff),
-              "v"(abase[1] + poff), "v"(abase[2] + poff)
-            : "memory");
+        bf16x4* cur = (w & 1) ? frB : frA;
+        bf16x4* nxt = (w & 1) ? frA : frB;
+        if (w + 1 < W) {
+          WRW4_ISSUE(nxt, (unsigned)((w + 1) * 32 * 32));
+          WRW4_LAND(cur, 8);   // 8 reads of w+1 still in flight
+        } else {
+          WRW4_LAND(cur, 0);
+        }
         __builtin_amdgcn_sched_barrier(0);
         union { struct { bf16x4 lo, hi; } p; bf16x8 v; } b;
-        b.p.lo = bfr[0];
-        b.p.hi = bfr[1];
+        b.p.lo = cur[0];
+        b.p.hi = cur[1];
 #pragma unroll
         for (int j = 0; j < MAXT; ++j) {
           if (wid + 4 * j >= NT) continue;
           union { struct { bf16x4 lo, hi; } p; bf16x8 v; } a;
-          a.p.lo = afr[j][0];
-          a.p.hi = afr[j][1];
+          a.p.lo = cur[2 + 2 * j];
+          a.p.hi = cur[3 + 2 * j];
           acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v,
                                                            acc[j], 0, 0, 0);
         }
       }
+#undef WRW4_ISSUE
+#undef WRW4_LAND
     }
     __syncthreads();
   }
