@@ -68,7 +68,10 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw16_nhwc(
   constexpr int CI = 16;
   constexpr int CO = COT * 16;
   constexpr int NT = 25;                 // tap tiles = (kh,kw) pairs
-  constexpr int T16 = NT * 16;
+  // +1 slab tile: the spare (wid 1, j 6) slot computes the BIAS grad
+  // with an all-ones A row (D[0][o] = sum_k B[k][o] accumulated over
+  // every pixel window) — kills the separate at::native bias reduce
+  constexpr int T16 = (NT + 1) * 16;
   // row region: one 16-wide subimage row, padded to whole 1 KiB LDS-DMA
   // chunks so every glds writes full 64-lane spans (no partial EXEC)
   constexpr int ROW_BYTES = PIX * CI * 2;
@@ -218,10 +221,16 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw16_nhwc(
 #pragma unroll
         for (int j = 0; j < MAXT; ++j) {
           const int tt = wid + 4 * j;
-          if (tt >= NT) continue;
+          const bool bias_tile = (tt == NT);   // spare slot: wid 1, j 6
+          if (tt >= NT && !bias_tile) continue;
           union { struct { bf16x4 lo, hi; } p; bf16x8 v; } a;
-          a.p.lo = afr[j][0];
-          a.p.hi = afr[j][1];
+          if (bias_tile) {
+            const bf16x8 ones = (bf16x8)(short)0x3F80;  // bf16 1.0 splat
+            a.v = (m == 0) ? ones : (bf16x8)(short)0;
+          } else {
+            a.p.lo = afr[j][0];
+            a.p.hi = afr[j][1];
+          }
 #pragma unroll
           for (int ot = 0; ot < COT; ++ot) {
             union { struct { bf16x4 lo, hi; } p; bf16x8 v; } b;
@@ -243,7 +252,7 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw16_nhwc(
 #pragma unroll
   for (int j = 0; j < MAXT; ++j) {
     const int tt = wid + 4 * j;
-    if (tt >= NT) continue;
+    if (tt > NT) continue;               // NT itself = the bias tile
 #pragma unroll
     for (int ot = 0; ot < COT; ++ot) {
       const int o = ot * 16 + m;
@@ -303,7 +312,7 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
   constexpr int CI = 4;
   constexpr int CO = 16;
   constexpr int NT = 10;                  // (kw 0..4) x (khg 0..1)
-  constexpr int T16 = NT * 16;
+  constexpr int T16 = (NT + 1) * 16;      // +bias tile (wid 2, j 2)
   constexpr int ROW_BYTES = PIX * 16 * 2;  // one 16-wide image
   constexpr int NCH4 = ROW_BYTES / 256;    // 4B-granule chunks per image
   constexpr int NCHB = (ROW_BYTES + 1023) / 1024;
@@ -431,10 +440,17 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
         b.p.hi = cur[1];
 #pragma unroll
         for (int j = 0; j < MAXT; ++j) {
-          if (wid + 4 * j >= NT) continue;
+          const int t = wid + 4 * j;
+          const bool bias_tile = (t == NT);  // spare slot: wid 2, j 2
+          if (t >= NT && !bias_tile) continue;
           union { struct { bf16x4 lo, hi; } p; bf16x8 v; } a;
-          a.p.lo = cur[2 + 2 * j];
-          a.p.hi = cur[3 + 2 * j];
+          if (bias_tile) {
+            const bf16x8 ones = (bf16x8)(short)0x3F80;
+            a.v = (m == 0) ? ones : (bf16x8)(short)0;
+          } else {
+            a.p.lo = cur[2 + 2 * j];
+            a.p.hi = cur[3 + 2 * j];
+          }
           acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v,
                                                            acc[j], 0, 0, 0);
         }
@@ -450,7 +466,7 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
 #pragma unroll
   for (int j = 0; j < MAXT; ++j) {
     const int t = wid + 4 * j;
-    if (t >= NT) continue;
+    if (t > NT) continue;                // NT itself = the bias tile
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int slot = t * 16 + q * 4 + i;

@@ -116,7 +116,7 @@ def build_wrw_unpack_index(weight_shape) -> torch.Tensor:
     CI = (CIr + 3) & ~3
     idx = torch.empty(CO, CIr, KH, KW, dtype=torch.int64)
     if CI == 4 and CO == 16:
-        T16 = 160
+        T16 = 176  # 10 tap tiles + the fused-bias tile
         for o in range(CO):
             for ci in range(CIr):
                 for kh in range(KH):
@@ -125,7 +125,7 @@ def build_wrw_unpack_index(weight_shape) -> torch.Tensor:
                         idx[o, ci, kh, kw] = slot * CO + o
         return idx.reshape(-1), T16
     NT = (25 * CI + 15) // 16
-    T16 = NT * 16
+    T16 = NT * 16 + (16 if CI == 16 else 0)  # v2: +fused-bias tile
     for o in range(CO):
         for ci in range(CIr):
             for kh in range(KH):
@@ -136,20 +136,26 @@ def build_wrw_unpack_index(weight_shape) -> torch.Tensor:
 
 
 def wrw_via_kernel(xb_padded: torch.Tensor, go: torch.Tensor,
-                   unpack_idx: torch.Tensor, T16: int, weight_shape):
-    """Run the custom wrw kernel; returns dW in OIHW fp32."""
+                   unpack_idx: torch.Tensor, T16: int, weight_shape,
+                   want_bias: bool = False):
+    """Run the custom wrw kernel; returns dW in OIHW fp32 (and, with
+    want_bias, the bias gradient the kernel fused via its all-ones tap
+    tile — slab row T16-16)."""
     from geomx_amd import _geops
     N, CI, Hi, Wi = xb_padded.shape
     CO = go.shape[1]
     Ho, Wo = go.shape[2], go.shape[3]
     n_blocks = N * ((Ho + 3) // 4)
     n_wg = min(_WRW_NWG, n_blocks)
-    part = torch.empty(n_wg, T16, CO, dtype=torch.float32,
+    part = torch.zeros(n_wg, T16, CO, dtype=torch.float32,
                        device=go.device)
     _geops.conv5_wrw_nhwc(xb_padded, go, part, N, Hi, Wi, Ho, Wo, CI, CO,
                           n_wg)
-    full = part.sum(dim=0).reshape(-1)
-    return full[unpack_idx].reshape(weight_shape)
+    full = part.sum(dim=0)
+    dw = full.reshape(-1)[unpack_idx].reshape(weight_shape)
+    if want_bias:
+        return dw, full[T16 - 16]
+    return dw
 
 
 class _Conv5Fn(torch.autograd.Function):
@@ -219,14 +225,11 @@ class _Conv5Fn(torch.autograd.Function):
             if ctx.wrw_pack is not None:
                 xb_pad = x  # saved padded (see forward)
                 unpack_idx, T16 = ctx.wrw_pack
-                grad_w = wrw_via_kernel(xb_pad, go, unpack_idx, T16,
-                                        weight.shape).to(weight.dtype)
+                grad_w, gb = wrw_via_kernel(xb_pad, go, unpack_idx, T16,
+                                            weight.shape, want_bias=True)
+                grad_w = grad_w.to(weight.dtype)
                 if ctx.has_bias:
-                    # dtype= fuses the fp32 accumulation into the
-                    # reduce (no materialized fp32 copy of go)
-                    grad_b = go.sum(dim=(0, 2, 3),
-                                    dtype=torch.float32) \
-                        .to(weight.dtype)
+                    grad_b = gb.to(weight.dtype)  # fused in the kernel
             else:
                 gi, gw, gb = torch.ops.aten.convolution_backward(
                     go, x, weight.to(torch.bfloat16),
@@ -284,11 +287,10 @@ class _AtenSplitConvFn(torch.autograd.Function):
             if native_available() and (CIr, CO) in _WRW_SUPPORTED \
                     and CIr % 4 == 0:
                 idx, t16 = _wrw_cached_index(w.shape, go.device)
-                grad_w = wrw_via_kernel(
+                grad_w, gb = wrw_via_kernel(
                     x.contiguous(memory_format=torch.channels_last), go,
-                    idx, t16, w.shape)
-                grad_b = go.sum(dim=(0, 2, 3), dtype=torch.float32) \
-                    if ctx.has_bias else None
+                    idx, t16, w.shape, want_bias=True)
+                grad_b = gb if ctx.has_bias else None
             else:
                 _, gw, gb = torch.ops.aten.convolution_backward(
                     go, x, w, [CO] if ctx.has_bias else None, [1, 1],

@@ -607,3 +607,23 @@ def test_bsc_fused_capacity_bound():
     sent[ii] = True
     assert torch.all(v.cpu()[sent] == 0)
     assert torch.all(v.cpu()[~sent] != 0)
+
+
+@pytest.mark.parametrize("ci,co,hw", [(16, 32, 70), (4, 16, 70)])
+def test_wrw_fused_bias_matches_sum(ci, co, hw):
+    """The all-ones bias tile fused into the wrw kernels must equal the
+    plain fp32 reduce of grad_out over (N,H,W)."""
+    from geomx_amd.ops import conv as C
+    torch.manual_seed(61)
+    N = 3
+    x = torch.randn(N, ci, hw, hw, device=DEV, dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    go = torch.randn(N, co, hw - 4, hw - 4, device=DEV,
+                     dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    idx, t16 = C.build_wrw_unpack_index((co, ci, 5, 5))
+    _, bias = C.wrw_via_kernel(x, go, idx.to(DEV), t16, (co, ci, 5, 5),
+                               want_bias=True)
+    ref_b = go.sum(dim=(0, 2, 3), dtype=torch.float32)
+    assert torch.allclose(bias, ref_b, rtol=1e-3, atol=0.5), \
+        (bias - ref_b).abs().max()
