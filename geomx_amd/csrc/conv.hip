@@ -203,6 +203,137 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_nhwc(
   }
 }
 
+
+// ---------------------------------------------------------------------
+// Fused conv(5x5, CI<=4 padded, CO=16) + bias + ReLU + 2x2 maxpool for
+// the flagship conv1 stage: computes TWO conv rows per wave, pools them
+// in-register and writes only the pooled row + the argmax-quadrant mask
+// (the same uint8 code format k_relu_maxpool2_bwd consumes: 0..3 =
+// quadrant (dy,dx), 255 = relu-clamped). Eliminates the full-resolution
+// conv output round trip (1.55 GB write + read + 0.19 GB mask at the
+// bench shape) and the separate pool kernel.
+// ---------------------------------------------------------------------
+
+template <int CI>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv5_pool_nhwc(
+    const bf16_t* __restrict__ in,       // [N][Hi][Wi][CI]
+    const bf16_t* __restrict__ w_frags,  // [1][nK][64][8]
+    const float* __restrict__ bias,      // [16] or nullptr
+    bf16_t* __restrict__ out,            // [N][Hop][Wop][16]
+    uint8_t* __restrict__ mask,          // [N*Hop*Wop*16]
+    int Nn, int Hi, int Wi, int Ho, int Wo) {
+  constexpr int S = 5 * CI;
+  constexpr int Sp = (S + 7) & ~7;
+  constexpr int K = 5 * Sp;
+  constexpr int nK = (K + 31) / 32;
+  constexpr int CO = 16;
+  static_assert(nK < 8, "register-B variant only");
+
+  const int lane = threadIdx.x & 63;
+  const int p = lane & 15;
+  const int q = lane >> 4;
+  const int m = lane & 15;
+
+  bf16x8 breg[nK];
+#pragma unroll
+  for (int i = 0; i < nK; ++i)
+    breg[i] = *reinterpret_cast<const bf16x8*>(w_frags + (i * 64 + lane) * 8);
+  const float bias_v = bias ? bias[m] : 0.0f;
+
+  const int Hop = Ho >> 1, Wop = Wo >> 1;
+  const int tiles_w = (Wo + 15) >> 4;
+  const long long n_rows = (long long)Nn * Hop;
+  const long long wave_id =
+      ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const long long n_waves = ((long long)gridDim.x * blockDim.x) >> 6;
+
+  for (long long row = wave_id; row < n_rows; row += n_waves) {
+    const int hp = (int)(row % Hop);
+    const long long n = row / Hop;
+    const int ho0 = hp * 2;
+    const long long in_n = (long long)n * Hi * Wi * CI;
+    const long long out_row =
+        ((long long)n * Hop + hp) * (long long)Wop * CO;
+
+    auto load_row = [&](int tw, int r, bf16x8(&dst)[nK]) {
+      const int wo_raw = (tw << 4) + p;
+      const int wo = wo_raw < Wo ? wo_raw : (Wo - 1);
+      const long long e_base =
+          in_n + (long long)wo * CI + (long long)(ho0 + r) * (Wi * CI);
+#pragma unroll
+      for (int km = 0; km < nK; ++km) {
+        const int k0 = km * 32 + q * 8;
+        const int kh = k0 / Sp;
+        const int j0 = k0 % Sp;
+        dst[km] = (bf16x8)0;
+        if (k0 < K && j0 < S) {
+          const long long e = e_base + (long long)kh * (Wi * CI) + j0;
+          if (S - j0 >= 8) {
+            dst[km] = *reinterpret_cast<const bf16x8*>(in + e);
+          } else {
+            const uint2 v = *reinterpret_cast<const uint2*>(in + e);
+            union { uint4 u; bf16x8 h; } cv;
+            cv.u = make_uint4(v.x, v.y, 0u, 0u);
+            dst[km] = cv.h;
+          }
+        }
+      }
+    };
+
+    bf16x8 af0[nK], af1[nK];
+    load_row(0, 0, af0);
+    load_row(0, 1, af1);
+    for (int tw = 0; tw < tiles_w; ++tw) {
+      f32x4 a0 = (f32x4)0.0f, a1 = (f32x4)0.0f;
+      bf16x8 nf0[nK], nf1[nK];
+      if (tw + 1 < tiles_w) {
+        load_row(tw + 1, 0, nf0);
+        load_row(tw + 1, 1, nf1);
+      }
+#pragma unroll
+      for (int km = 0; km < nK; ++km) {
+        a0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0[km], breg[km], a0,
+                                                     0, 0, 0);
+        a1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1[km], breg[km], a1,
+                                                     0, 0, 0);
+      }
+#pragma unroll
+      for (int km = 0; km < nK; ++km) { af0[km] = nf0[km]; af1[km] = nf1[km]; }
+
+      // pool: conv pixel = tw*16 + q*4 + i; horizontal pairs live in
+      // (i, i+1) of the SAME lane, vertical in (a0, a1)
+      const int wo0 = tw << 4;
+#pragma unroll
+      for (int pi = 0; pi < 2; ++pi) {
+        const int i0 = pi * 2;
+        const int wp = (wo0 + q * 4 + i0) >> 1;
+        if (wp >= Wop) continue;
+        const float q0 = a0[i0] + bias_v;      // (dy,dx)=(0,0)
+        const float q1 = a0[i0 + 1] + bias_v;  // (0,1)
+        const float q2 = a1[i0] + bias_v;      // (1,0)
+        const float q3 = a1[i0 + 1] + bias_v;  // (1,1)
+        float mx = q0;
+        int arg = 0;
+        if (q1 > mx) { mx = q1; arg = 1; }
+        if (q2 > mx) { mx = q2; arg = 2; }
+        if (q3 > mx) { mx = q3; arg = 3; }
+        bf16_t ov;
+        uint8_t code;
+        if (mx <= 0.0f) {
+          ov = (bf16_t)0;
+          code = 255;
+        } else {
+          ov = cf2bf(mx);
+          code = (uint8_t)arg;
+        }
+        const long long o_off = out_row + (long long)wp * CO + m;
+        out[o_off] = ov;
+        mask[o_off] = code;
+      }
+    }
+  }
+}
+
 // NHWC channel pad: [N,H,W,3] (fp32 or bf16) -> [N,H,W,4] bf16 with a
 // zero 4th channel. One thread per OUTPUT pixel: reads 3 elems, writes
 // one 8-byte bf16x4.
@@ -258,6 +389,22 @@ int geops_conv5_nhwc(const bf16_t* in, const bf16_t* w_frags,
     if (CI == 32 && CO == 32) { LAUNCH(32, 2, 0) }
   }
 #undef LAUNCH
+  return -1;
+}
+
+
+int geops_conv5_pool_nhwc(const bf16_t* in, const bf16_t* w_frags,
+                          const float* bias, bf16_t* out, uint8_t* mask,
+                          int Nn, int Hi, int Wi, int Ho, int Wo, int CI,
+                          int CO, hipStream_t s) {
+  if (CO != 16 || (Ho & 1) || (Wo & 1)) return -1;
+  const long long rows = (long long)Nn * (Ho >> 1);
+  const dim3 grid(conv_blocks(rows)), block(CONV_THREADS);
+  if (CI == 4) {
+    hipLaunchKernelGGL((k_conv5_pool_nhwc<4>), grid, block, 0, s, in,
+                       w_frags, bias, out, mask, Nn, Hi, Wi, Ho, Wo);
+    return 0;
+  }
   return -1;
 }
 

@@ -59,6 +59,9 @@ void geops_relu_maxpool2_bwd(const unsigned short*, const uint8_t*,
 int geops_conv5_nhwc(const unsigned short*, const unsigned short*,
                      const float*, unsigned short*, int, int, int, int, int,
                      int, int, int, hipStream_t);
+int geops_conv5_pool_nhwc(const unsigned short*, const unsigned short*,
+                          const float*, unsigned short*, uint8_t*, int, int,
+                          int, int, int, int, int, hipStream_t);
 void geops_pad_ch3to4_nhwc(const void*, unsigned short*, long long, int,
                            hipStream_t);
 int geops_conv5_wrw_nhwc(const unsigned short*, const unsigned short*,
@@ -294,6 +297,27 @@ void conv5_nhwc(torch::Tensor in, torch::Tensor w_frags, torch::Tensor bias,
   launch_check("conv5_nhwc");
 }
 
+void conv5_pool_nhwc(torch::Tensor in, torch::Tensor w_frags,
+                     torch::Tensor bias, torch::Tensor out,
+                     torch::Tensor mask, int64_t N, int64_t Hi, int64_t Wi,
+                     int64_t Ho, int64_t Wo, int64_t CI, int64_t CO) {
+  TORCH_CHECK(in.is_cuda() && w_frags.is_cuda() && out.is_cuda() &&
+              mask.is_cuda());
+  TORCH_CHECK(in.scalar_type() == torch::kBFloat16 &&
+              out.scalar_type() == torch::kBFloat16 &&
+              mask.scalar_type() == torch::kUInt8);
+  const bool has_bias = bias.numel() > 0;
+  const int rc = geops_conv5_pool_nhwc(
+      (const unsigned short*)in.data_ptr(),
+      (const unsigned short*)w_frags.data_ptr(),
+      has_bias ? bias.data_ptr<float>() : nullptr,
+      (unsigned short*)out.data_ptr(), mask.data_ptr<uint8_t>(), (int)N,
+      (int)Hi, (int)Wi, (int)Ho, (int)Wo, (int)CI, (int)CO, cur_stream());
+  TORCH_CHECK(rc == 0, "conv5_pool_nhwc: unsupported geometry CI=", CI,
+              " CO=", CO);
+  launch_check("conv5_pool_nhwc");
+}
+
 void conv5_wrw_nhwc(torch::Tensor in, torch::Tensor gout, torch::Tensor part,
                     int64_t N, int64_t Hi, int64_t Wi, int64_t Ho,
                     int64_t Wo, int64_t CI, int64_t CO, int64_t n_wg) {
@@ -441,6 +465,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv5_nhwc", &conv5_nhwc);
   m.def("pad_ch3to4_nhwc", &pad_ch3to4_nhwc);
   m.def("conv5_wrw_nhwc", &conv5_wrw_nhwc);
+  m.def("conv5_pool_nhwc", &conv5_pool_nhwc);
   m.def("wrw2_dump", [](torch::Tensor in, torch::Tensor gout,
                         torch::Tensor dump, int64_t N, int64_t Hi,
                         int64_t Wi, int64_t Ho, int64_t Wo, int64_t CO) {

@@ -19,11 +19,12 @@ class GeoCNN(nn.Module):
     def __init__(self, in_channels: int = 3, image_size: int = 224,
                  num_classes: int = 10):
         super().__init__()
-        from ..ops.conv import GeoConv5
+        from ..ops.conv import GeoConv5, GeoConv5Pool
         from ..ops.fused import FusedReLUPool2
         self.features = nn.Sequential(
-            GeoConv5(in_channels, 16),   # MFMA direct conv on GPU
-            FusedReLUPool2(),   # relu+maxpool in one gfx950 kernel on GPU
+            # conv1 stage: conv+bias+relu+maxpool in ONE gfx950 kernel
+            # (the full-res conv output never touches HBM)
+            GeoConv5Pool(in_channels, 16),
             GeoConv5(16, 32),
             FusedReLUPool2(),
         )

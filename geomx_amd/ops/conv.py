@@ -344,3 +344,108 @@ class GeoConv5(torch.nn.Conv2d):
             self._wrw_t16 = t16
         return _Conv5Fn.apply(x, self.weight, self.bias, self._fwd_idx,
                               self._dgrad_idx, self._wrw_idx, self._wrw_t16)
+
+
+class _Conv5PoolFn(torch.autograd.Function):
+    """Fused conv1 stage: conv5 + bias + ReLU + 2x2 maxpool in ONE
+    kernel (k_conv5_pool_nhwc) — the full-resolution conv output
+    (1.55 GB at the bench shape) never exists in memory. Backward:
+    the recorded argmax-quadrant mask drives the existing
+    relu_maxpool2_bwd scatter, then the custom dgrad conv and the
+    wrw kernel (fused bias tile) as in _Conv5Fn."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, fwd_idx, dgrad_idx, wrw_idx, wrw_t16):
+        from geomx_amd import _geops
+        N, CIr, Hi, Wi = x.shape
+        CO = weight.shape[0]
+        CI = (CIr + 3) & ~3
+        Ho, Wo = Hi - 4, Wi - 4
+        if CI != CIr:
+            xc = x.contiguous(memory_format=torch.channels_last)
+            xb = torch.empty(N, CI, Hi, Wi, dtype=torch.bfloat16,
+                             device=x.device,
+                             memory_format=torch.channels_last)
+            _geops.pad_ch3to4_nhwc(xc, xb, N * Hi * Wi)
+        else:
+            xb = x.to(torch.bfloat16) \
+                .contiguous(memory_format=torch.channels_last)
+        wb = weight.detach().to(torch.bfloat16).reshape(-1)
+        wz = torch.cat([wb, wb.new_zeros(1)])
+        w_frags = wz[fwd_idx].contiguous()
+        Hop, Wop = Ho // 2, Wo // 2
+        out = torch.empty(N, CO, Hop, Wop, dtype=torch.bfloat16,
+                          device=x.device,
+                          memory_format=torch.channels_last)
+        mask = torch.empty(N * Hop * Wop * CO, dtype=torch.uint8,
+                           device=x.device)
+        b = bias.detach().float() if bias is not None else torch.Tensor()
+        _geops.conv5_pool_nhwc(xb, w_frags, b, out, mask, N, Hi, Wi, Ho,
+                               Wo, CI, CO)
+        ctx.save_for_backward(xb, weight, dgrad_idx, mask)
+        ctx.wrw_pack = (wrw_idx, wrw_t16)
+        ctx.has_bias = bias is not None
+        ctx.dims = (N, CIr, CI, CO, Hi, Wi, Ho, Wo)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_pooled):
+        from geomx_amd import _geops
+        import torch.nn.functional as F
+        xb, weight, dgrad_idx, mask = ctx.saved_tensors
+        N, CIr, CI, CO, Hi, Wi, Ho, Wo = ctx.dims
+        gp = grad_pooled.to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        go = torch.empty(N, CO, Ho, Wo, dtype=torch.bfloat16,
+                         device=gp.device,
+                         memory_format=torch.channels_last)
+        _geops.relu_maxpool2_bwd(gp, mask, go, N, CO, Ho, Wo)
+        grad_x = None
+        if ctx.needs_input_grad[0]:
+            COp = (CIr + 15) & ~15
+            wb = weight.detach().to(torch.bfloat16).reshape(-1)
+            wz = torch.cat([wb, wb.new_zeros(1)])
+            w_frags = wz[dgrad_idx].contiguous()
+            gop = F.pad(go, (4, 4, 4, 4)) \
+                .contiguous(memory_format=torch.channels_last)
+            gx = torch.empty(N, COp, Hi, Wi, dtype=torch.bfloat16,
+                             device=xb.device,
+                             memory_format=torch.channels_last)
+            _geops.conv5_nhwc(gop, w_frags, torch.Tensor(), gx, N, Ho + 8,
+                              Wo + 8, Hi, Wi, CO, COp, 0)
+            grad_x = gx[:, :CIr] if COp != CIr else gx
+        grad_w = grad_b = None
+        if ctx.needs_input_grad[1] or (ctx.has_bias and
+                                       ctx.needs_input_grad[2]):
+            unpack_idx, T16 = ctx.wrw_pack
+            grad_w, gb = wrw_via_kernel(xb, go, unpack_idx, T16,
+                                        weight.shape, want_bias=True)
+            grad_w = grad_w.to(weight.dtype)
+            grad_b = gb.to(weight.dtype) if ctx.has_bias else None
+        return grad_x, grad_w, grad_b, None, None, None, None
+
+
+class GeoConv5Pool(GeoConv5):
+    """Drop-in for GeoConv5 -> ReLU -> MaxPool2d(2,2): one fused kernel
+    on the GPU path, eager fallback elsewhere. Parameters live on this
+    module exactly as on GeoConv5 (state-dict compatible)."""
+
+    def _pool_eligible(self, x) -> bool:
+        Ho, Wo = x.shape[2] - 4, x.shape[3] - 4
+        return (self._eligible(x) and self.out_channels == 16
+                and Ho % 2 == 0 and Wo % 2 == 0)
+
+    def forward(self, x):
+        if not self._pool_eligible(x):
+            y = super().forward(x)
+            return torch.nn.functional.max_pool2d(
+                torch.nn.functional.relu(y), 2, 2)
+        if self._fwd_idx is None or self._fwd_idx.device != x.device:
+            self._fwd_idx = build_fwd_index(self.weight.shape).to(x.device)
+            self._dgrad_idx = build_dgrad_index(self.weight.shape).to(x.device)
+            wi, t16 = build_wrw_unpack_index(self.weight.shape)
+            self._wrw_idx = wi.to(x.device)
+            self._wrw_t16 = t16
+        return _Conv5PoolFn.apply(x, self.weight, self.bias, self._fwd_idx,
+                                  self._dgrad_idx, self._wrw_idx,
+                                  self._wrw_t16)

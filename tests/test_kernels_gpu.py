@@ -627,3 +627,70 @@ def test_wrw_fused_bias_matches_sum(ci, co, hw):
     ref_b = go.sum(dim=(0, 2, 3), dtype=torch.float32)
     assert torch.allclose(bias, ref_b, rtol=1e-3, atol=0.5), \
         (bias - ref_b).abs().max()
+
+
+def test_conv5_pool_fused_matches_eager():
+    """GeoConv5Pool (conv+bias+relu+maxpool in one kernel) vs the eager
+    chain, forward AND all three grads."""
+    from geomx_amd.ops.conv import GeoConv5Pool
+    torch.manual_seed(71)
+    m = GeoConv5Pool(3, 16).to(DEV)
+    x = torch.randn(4, 3, 68, 68, device=DEV, dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    y = m(x)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    m2 = torch.nn.Conv2d(3, 16, 5).to(DEV)
+    with torch.no_grad():
+        m2.weight.copy_(m.weight); m2.bias.copy_(m.bias)
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = torch.nn.functional.max_pool2d(torch.nn.functional.relu(
+        torch.nn.functional.conv2d(x2, m2.weight.to(torch.bfloat16),
+                                   m2.bias.to(torch.bfloat16))), 2, 2)
+    y2.backward(g)
+    assert y.shape == y2.shape
+    assert torch.allclose(y.float(), y2.float(), atol=0.1, rtol=0.05), \
+        (y.float() - y2.float()).abs().max()
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=0.1,
+                          rtol=0.05)
+    assert torch.allclose(m.weight.grad, m2.weight.grad.float(), atol=1.0,
+                          rtol=0.05), \
+        (m.weight.grad - m2.weight.grad.float()).abs().max()
+    assert torch.allclose(m.bias.grad, m2.bias.grad.float(), atol=0.5,
+                          rtol=0.05)
+
+
+def test_geo_cnn_gpu_full_model_grads():
+    """Whole flagship model (fused conv1 stage + split conv2) vs a plain
+    torch model: one step, matching loss and grads within bf16."""
+    from geomx_amd.models import create_model
+    torch.manual_seed(72)
+    m = create_model("geomx_cnn", image_size=68).to(DEV) \
+        .to(memory_format=torch.channels_last)
+    ref = torch.nn.Sequential(
+        torch.nn.Conv2d(3, 16, 5), torch.nn.ReLU(),
+        torch.nn.MaxPool2d(2, 2),
+        torch.nn.Conv2d(16, 32, 5), torch.nn.ReLU(),
+        torch.nn.MaxPool2d(2, 2), torch.nn.Flatten(),
+    ).to(DEV)
+    with torch.no_grad():
+        ref[0].weight.copy_(m.features[0].weight)
+        ref[0].bias.copy_(m.features[0].bias)
+        ref[3].weight.copy_(m.features[1].weight)
+        ref[3].bias.copy_(m.features[1].bias)
+    x = torch.randn(4, 3, 68, 68, device=DEV)
+    with torch.autocast("cuda", torch.bfloat16):
+        feat = m.features(x.to(memory_format=torch.channels_last))
+        feat2 = ref(x)
+    loss = feat.float().square().mean()
+    loss2 = feat2.float().square().mean()
+    loss.backward(); loss2.backward()
+    assert torch.allclose(feat.reshape(4, -1).float(), feat2.float(),
+                          atol=0.1, rtol=0.05)
+    for a, b in [(m.features[0].weight, ref[0].weight),
+                 (m.features[1].weight, ref[3].weight)]:
+        scale = b.grad.abs().max().item()
+        assert torch.allclose(a.grad.float(), b.grad.float(),
+                              atol=0.05 * scale + 1e-5, rtol=0.05), \
+            (a.grad.float() - b.grad.float()).abs().max()
