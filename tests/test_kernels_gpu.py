@@ -630,8 +630,11 @@ def test_wrw_fused_bias_matches_sum(ci, co, hw):
 
 
 def test_conv5_pool_fused_matches_eager():
-    """GeoConv5Pool (conv+bias+relu+maxpool in one kernel) vs the eager
-    chain, forward AND all three grads."""
+    """GeoConv5Pool (conv+bias+relu+maxpool in one kernel) vs an fp32-
+    accumulation eager chain — the kernel argmaxes the fp32 MFMA
+    accumulators, so the reference must pool the fp32 conv (pooling the
+    bf16-rounded conv creates ties whose subgradient routing then fans
+    out x25 through the dgrad, a semantic difference, not an error)."""
     from geomx_amd.ops.conv import GeoConv5Pool
     torch.manual_seed(71)
     m = GeoConv5Pool(3, 16).to(DEV)
@@ -641,31 +644,30 @@ def test_conv5_pool_fused_matches_eager():
     g = torch.randn_like(y)
     y.backward(g)
 
-    m2 = torch.nn.Conv2d(3, 16, 5).to(DEV)
-    with torch.no_grad():
-        m2.weight.copy_(m.weight); m2.bias.copy_(m.bias)
-    x2 = x.detach().clone().requires_grad_(True)
-    y2 = torch.nn.functional.max_pool2d(torch.nn.functional.relu(
-        torch.nn.functional.conv2d(x2, m2.weight.to(torch.bfloat16),
-                                   m2.bias.to(torch.bfloat16))), 2, 2)
-    y2.backward(g)
+    x2 = x.detach().float().requires_grad_(True)
+    w32 = m.weight.detach().to(torch.bfloat16).float()
+    b32 = m.bias.detach().to(torch.bfloat16).float()
+    conv32 = torch.nn.functional.conv2d(x2, w32, b32)
+    y2 = torch.nn.functional.max_pool2d(
+        torch.nn.functional.relu(conv32), 2, 2)
+    y2.backward(g.float())
     assert y.shape == y2.shape
-    assert torch.allclose(y.float(), y2.float(), atol=0.1, rtol=0.05), \
-        (y.float() - y2.float()).abs().max()
-    # x.grad can legitimately differ at POOL TIES: the fused kernel
-    # argmaxes the fp32 accumulators while eager argmaxes bf16-rounded
-    # conv outputs — both are valid subgradients, so bound the FRACTION
-    # of rerouted elements instead of demanding pointwise equality
-    diff = (x.grad.float() - x2.grad.float()).abs()
-    frac = (diff > 0.05).float().mean().item()
-    assert frac < 0.01, frac
-    assert torch.allclose(x.grad.float().sum(), x2.grad.float().sum(),
-                          rtol=0.05, atol=1.0)
-    assert torch.allclose(m.weight.grad, m2.weight.grad.float(), atol=1.0,
+    assert torch.allclose(y.float(), y2, atol=0.1, rtol=0.05), \
+        (y.float() - y2).abs().max()
+    assert torch.allclose(x.grad.float(), x2.grad, atol=0.1, rtol=0.05), \
+        (x.grad.float() - x2.grad).abs().max()
+    # weight/bias grads flow through our wrw kernel on the mask-routed go
+    scale = m.weight.grad.abs().max().item() + 1e-6
+    # reference weight grad via autograd on the fp32 chain
+    w_ref = w32.detach().requires_grad_(True)
+    b_ref = b32.detach().requires_grad_(True)
+    yr = torch.nn.functional.max_pool2d(torch.nn.functional.relu(
+        torch.nn.functional.conv2d(x.detach().float(), w_ref, b_ref)), 2, 2)
+    yr.backward(g.float())
+    assert torch.allclose(m.weight.grad, w_ref.grad, atol=0.03 * scale + 0.5,
                           rtol=0.05), \
-        (m.weight.grad - m2.weight.grad.float()).abs().max()
-    assert torch.allclose(m.bias.grad, m2.bias.grad.float(), atol=0.5,
-                          rtol=0.05)
+        (m.weight.grad - w_ref.grad).abs().max()
+    assert torch.allclose(m.bias.grad, b_ref.grad, atol=0.5, rtol=0.05)
 
 
 def test_geo_cnn_gpu_full_model_grads():
