@@ -193,7 +193,9 @@ class _Conv5Fn(torch.autograd.Function):
         ctx.wrw_pack = (wrw_idx, wrw_t16) if use_wrw else None
         ctx.has_bias = bias is not None
         ctx.dims = (N, CIr, CI, CO, Hi, Wi, Ho, Wo)
-        return out
+        # outside autocast (fp32 graphs) the downstream layers expect the
+        # input dtype back
+        return out if x.dtype == torch.bfloat16 else out.to(x.dtype)
 
     @staticmethod
     def backward(ctx, grad_out):
@@ -268,7 +270,7 @@ class _AtenSplitConvFn(torch.autograd.Function):
         out = torch.nn.functional.conv2d(xb, wb, bb)
         ctx.save_for_backward(xb, wb)
         ctx.has_bias = bias is not None
-        return out
+        return out if x.dtype == torch.bfloat16 else out.to(x.dtype)
 
     @staticmethod
     def backward(ctx, grad_out):
@@ -386,7 +388,7 @@ class _Conv5PoolFn(torch.autograd.Function):
         ctx.wrw_pack = (wrw_idx, wrw_t16)
         ctx.has_bias = bias is not None
         ctx.dims = (N, CIr, CI, CO, Hi, Wi, Ho, Wo)
-        return out
+        return out if x.dtype == torch.bfloat16 else out.to(x.dtype)
 
     @staticmethod
     def backward(ctx, grad_pooled):
