@@ -403,41 +403,28 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
         abase[j] = lds0 + im * RPB + (kw + 4 * q) * 32 + m * 8;
       }
       const unsigned bbase = lds0 + (NIMG + r) * RPB + (4 * q) * 32 + m * 8;
-#define WRW4_ISSUE(FR, POFF)                                              \
-        asm volatile(                                                     \
-            "ds_read_b64_tr_b16 %0, %8\n\t"                               \
-            "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"                    \
-            "ds_read_b64_tr_b16 %2, %9\n\t"                               \
-            "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"                    \
-            "ds_read_b64_tr_b16 %4, %10\n\t"                              \
-            "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"                   \
-            "ds_read_b64_tr_b16 %6, %11\n\t"                              \
-            "ds_read_b64_tr_b16 %7, %11 offset:512"                       \
-            : "=&v"(FR[0]), "=&v"(FR[1]), "=&v"(FR[2]), "=&v"(FR[3]),     \
-              "=&v"(FR[4]), "=&v"(FR[5]), "=&v"(FR[6]), "=&v"(FR[7])      \
-            : "v"(bbase + (POFF)), "v"(abase[0] + (POFF)),                \
-              "v"(abase[1] + (POFF)), "v"(abase[2] + (POFF))              \
-            : "memory")
-#define WRW4_LAND(FR, CNT)                                                \
-        asm volatile("s_waitcnt lgkmcnt(" #CNT ")"                        \
-            : "+v"(FR[0]), "+v"(FR[1]), "+v"(FR[2]), "+v"(FR[3]),         \
-              "+v"(FR[4]), "+v"(FR[5]), "+v"(FR[6]), "+v"(FR[7])          \
-            :: "memory")
-      bf16x4 frA[8], frB[8];
-      WRW4_ISSUE(frA, 0u);
-      for (int w = 0; w < W; ++w) {
-        bf16x4* cur = (w & 1) ? frB : frA;
-        bf16x4* nxt = (w & 1) ? frA : frB;
-        if (w + 1 < W) {
-          WRW4_ISSUE(nxt, (unsigned)((w + 1) * 32 * 32));
-          WRW4_LAND(cur, 8);   // 8 reads of w+1 still in flight
-        } else {
-          WRW4_LAND(cur, 0);
-        }
+for (int w = 0; w < W; ++w) {
+        const unsigned poff = (unsigned)(w * 32 * 32);
+        bf16x4 fr[8];
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %8\n\t"
+            "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+            "ds_read_b64_tr_b16 %2, %9\n\t"
+            "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"
+            "ds_read_b64_tr_b16 %4, %10\n\t"
+            "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"
+            "ds_read_b64_tr_b16 %6, %11\n\t"
+            "ds_read_b64_tr_b16 %7, %11 offset:512\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(fr[0]), "=&v"(fr[1]), "=&v"(fr[2]), "=&v"(fr[3]),
+              "=&v"(fr[4]), "=&v"(fr[5]), "=&v"(fr[6]), "=&v"(fr[7])
+            : "v"(bbase + poff), "v"(abase[0] + poff),
+              "v"(abase[1] + poff), "v"(abase[2] + poff)
+            : "memory");
         __builtin_amdgcn_sched_barrier(0);
         union { struct { bf16x4 lo, hi; } p; bf16x8 v; } b;
-        b.p.lo = cur[0];
-        b.p.hi = cur[1];
+        b.p.lo = fr[0];
+        b.p.hi = fr[1];
 #pragma unroll
         for (int j = 0; j < MAXT; ++j) {
           const int t = wid + 4 * j;
@@ -448,15 +435,13 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
             const bf16x8 ones = (bf16x8)(short)0x3F80;
             a.v = (m == 0) ? ones : (bf16x8)(short)0;
           } else {
-            a.p.lo = cur[2 + 2 * j];
-            a.p.hi = cur[3 + 2 * j];
+            a.p.lo = fr[2 + 2 * j];
+            a.p.hi = fr[3 + 2 * j];
           }
           acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v,
                                                            acc[j], 0, 0, 0);
         }
       }
-#undef WRW4_ISSUE
-#undef WRW4_LAND
     }
     __syncthreads();
   }
