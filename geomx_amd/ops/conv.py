@@ -193,9 +193,11 @@ class _Conv5Fn(torch.autograd.Function):
         ctx.wrw_pack = (wrw_idx, wrw_t16) if use_wrw else None
         ctx.has_bias = bias is not None
         ctx.dims = (N, CIr, CI, CO, Hi, Wi, Ho, Wo)
-        # outside autocast (fp32 graphs) the downstream layers expect the
-        # input dtype back
-        return out if x.dtype == torch.bfloat16 else out.to(x.dtype)
+        # outside autocast (fp32 graphs) the downstream layers expect
+        # the input dtype back; under autocast bf16 IS the compute dtype
+        if x.dtype == torch.bfloat16 or torch.is_autocast_enabled():
+            return out
+        return out.to(x.dtype)
 
     @staticmethod
     def backward(ctx, grad_out):
@@ -270,7 +272,9 @@ class _AtenSplitConvFn(torch.autograd.Function):
         out = torch.nn.functional.conv2d(xb, wb, bb)
         ctx.save_for_backward(xb, wb)
         ctx.has_bias = bias is not None
-        return out if x.dtype == torch.bfloat16 else out.to(x.dtype)
+        if x.dtype == torch.bfloat16 or torch.is_autocast_enabled():
+            return out
+        return out.to(x.dtype)
 
     @staticmethod
     def backward(ctx, grad_out):
@@ -388,7 +392,9 @@ class _Conv5PoolFn(torch.autograd.Function):
         ctx.wrw_pack = (wrw_idx, wrw_t16)
         ctx.has_bias = bias is not None
         ctx.dims = (N, CIr, CI, CO, Hi, Wi, Ho, Wo)
-        return out if x.dtype == torch.bfloat16 else out.to(x.dtype)
+        if x.dtype == torch.bfloat16 or torch.is_autocast_enabled():
+            return out
+        return out.to(x.dtype)
 
     @staticmethod
     def backward(ctx, grad_pooled):
