@@ -307,19 +307,30 @@ template <int PIX>
 __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
     const bf16_t* __restrict__ in,    // [N][Hi][Wi][4]
     const bf16_t* __restrict__ gout,  // [N][Ho][Wo][16]
-    float* __restrict__ part,         // [nWG][160][16]
+    float* __restrict__ part,         // [nWG][176][16]
     int Nn, int Hi, int Wi, int Ho, int Wo) {
+  // v3: the transpose read gathers PER-LANE 8B quarters (verified by
+  // scripts/tr16_scatter_probe.py: out[m][j] = mem[addr[4j+(m>>2)] +
+  // (m&3) elems]), so the A fragment (lane m = kho*4+ci) reads STRAIGHT
+  // from the 6 raw [pix][4] input rows: lane l points at
+  // row(r + khg*4 + (l&3)), pixel (pix0 + kw + 4q + (l>>2)&3).
+  // No kh-interleaved images, no 4x staging duplication: per block the
+  // A staging is 6 rows x PIX*8B once. Row regions are skewed by 64 B
+  // so the 4 rows a group touches land on distinct LDS banks.
   constexpr int CI = 4;
   constexpr int CO = 16;
   constexpr int NT = 10;                  // (kw 0..4) x (khg 0..1)
   constexpr int T16 = (NT + 1) * 16;      // +bias tile (wid 2, j 2)
-  constexpr int ROW_BYTES = PIX * 16 * 2;  // one 16-wide image
-  constexpr int NCH4 = ROW_BYTES / 256;    // 4B-granule chunks per image
-  constexpr int NCHB = (ROW_BYTES + 1023) / 1024;
-  constexpr int RPB = NCHB * 1024;
-  constexpr int NIMG = 4;                 // I_{r+4*khg}: p in {0,1,4,5}
-  __shared__ __attribute__((aligned(128))) char lds_all[(NIMG + WRW2_R) *
-                                                        RPB];
+  constexpr int AROW_BYTES = PIX * CI * 2;
+  constexpr int NCHA = (AROW_BYTES + 1023) / 1024;
+  constexpr int ARPB = NCHA * 1024 + 64;  // 64B skew: distinct banks/row
+  constexpr int BROW_BYTES = PIX * 16 * 2;
+  constexpr int NCHB = (BROW_BYTES + 1023) / 1024;
+  constexpr int BRPB = NCHB * 1024;
+  constexpr int NROW_A = 6;               // rows ho0 .. ho0+5
+  __shared__ __attribute__((aligned(128))) char lds_all[NROW_A * ARPB +
+                                                        WRW2_R * BRPB];
+  constexpr int BOFF = NROW_A * ARPB;
 
   const int lane = threadIdx.x & 63;
   const int q = lane >> 4;
@@ -333,12 +344,10 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
   const int n_wg = gridDim.x;
   const unsigned lds0 = lds_addr(lds_all);
 
-  constexpr int MAXT = 3;                 // 10 tiles over 4 waves
+  constexpr int MAXT = 3;
   f32x4 acc[MAXT];
 #pragma unroll
   for (int j = 0; j < MAXT; ++j) acc[j] = (f32x4)0.0f;
-
-  static constexpr int P_OF[NIMG] = {0, 1, 4, 5};
 
   for (long long blk = wg; blk < n_blocks; blk += n_wg) {
     const int bh = (int)(blk % blocks_h);
@@ -346,30 +355,26 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
     const int ho0 = bh * WRW2_R;
     const int nrows = (Ho - ho0) < WRW2_R ? (Ho - ho0) : WRW2_R;
 
-    // ---- stage the 4 interleaved input images (4B LDS-DMA granules)
+    // ---- stage 6 raw input rows + gout rows (16B LDS-DMA granules)
     {
-      const int nunits = NIMG * NCH4;   // 256B chunks
-      for (int t = wid; t < nunits; t += 4) {
-        const int im = t / NCH4;
-        const int c = t - im * NCH4;
-        const int b4 = c * 256 + lane * 4;          // dest byte
-        const int el = b4 >> 1;                     // dest element
-        const int pix = el >> 4;
-        const int kho = (el >> 2) & 3;
-        const int ci2 = el & 3;                     // 0 or 2
-        int row = ho0 + P_OF[im] + kho;
-        if (row >= Hi) row = Hi - 1;                // dead-tap rows only
+      const int nchunks_a = NROW_A * NCHA;
+      for (int t = wid; t < nchunks_a; t += 4) {
+        const int ir = t / NCHA;
+        const int c = t - ir * NCHA;
+        const int slot = c * 64 + lane;              // 16B = 2 pixels
+        const int pix = slot * 2;
+        int row = ho0 + ir;
+        if (row >= Hi) row = Hi - 1;                 // tail blocks only
         const bf16_t* src =
-            (pix < Wi) ? in + ((n * Hi + row) * (long long)Wi * CI +
-                               (long long)pix * CI + ci2)
-                       : g_wrw2_zeros;
+            (pix + 1 < Wi) ? in + ((n * Hi + row) * (long long)Wi * CI +
+                                   (long long)pix * CI)
+                           : g_wrw2_zeros;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)src,
-            (__attribute__((address_space(3))) void*)(lds_all + im * RPB +
-                                                      c * 256),
-            4, 0, 0);
+            (__attribute__((address_space(3))) void*)(lds_all + ir * ARPB +
+                                                      c * 1024),
+            16, 0, 0);
       }
-      // gout rows: [pix][16], straight 16B granules
       const int nchunks_b = nrows * NCHB;
       for (int t = wid; t < nchunks_b; t += 4) {
         const int rr = t / NCHB;
@@ -382,9 +387,8 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
                        : g_wrw2_zeros;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)src,
-            (__attribute__((address_space(3))) void*)(lds_all +
-                                                      (NIMG + rr) * RPB +
-                                                      c * 1024),
+            (__attribute__((address_space(3))) void*)(lds_all + BOFF +
+                                                      rr * BRPB + c * 1024),
             16, 0, 0);
       }
     }
@@ -392,34 +396,39 @@ __global__ __launch_bounds__(WRW2_THREADS) void k_conv5_wrw4_nhwc(
     __syncthreads();
 
     for (int r = 0; r < nrows; ++r) {
+      // per-lane A bases: lane quarter l&15 supplies
+      // row(r + khg*4 + (lane&3)), pixel (kw + 4q + ((lane&15)>>2))
       unsigned abase[MAXT];
 #pragma unroll
       for (int j = 0; j < MAXT; ++j) {
         int t = wid + 4 * j;
         if (t >= NT) t = NT - 1;
-        const int kw = t / 2;
-        const int khg = t - kw * 2;
-        const int im = r + 2 * khg;      // index into P_OF ({0,1,4,5})
-        abase[j] = lds0 + im * RPB + (kw + 4 * q) * 32 + m * 8;
+        const int kw = t >> 1;
+        const int khg = t & 1;
+        int rowsel = r + khg * 4 + (lane & 3);
+        if (rowsel > NROW_A - 1) rowsel = NROW_A - 1;  // dead taps only
+        abase[j] = lds0 + rowsel * ARPB +
+                   (unsigned)((kw + 4 * q + ((lane & 15) >> 2)) * 8);
       }
-      const unsigned bbase = lds0 + (NIMG + r) * RPB + (4 * q) * 32 + m * 8;
-for (int w = 0; w < W; ++w) {
-        const unsigned poff = (unsigned)(w * 32 * 32);
+      const unsigned bbase = lds0 + BOFF + r * BRPB + (4 * q) * 32 + m * 8;
+      for (int w = 0; w < W; ++w) {
+        const unsigned poffB = (unsigned)(w * 32 * 32);
+        const unsigned poffA = (unsigned)(w * 32 * 8);
         bf16x4 fr[8];
         asm volatile(
             "ds_read_b64_tr_b16 %0, %8\n\t"
             "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
             "ds_read_b64_tr_b16 %2, %9\n\t"
-            "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"
+            "ds_read_b64_tr_b16 %3, %9 offset:128\n\t"
             "ds_read_b64_tr_b16 %4, %10\n\t"
-            "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"
+            "ds_read_b64_tr_b16 %5, %10 offset:128\n\t"
             "ds_read_b64_tr_b16 %6, %11\n\t"
-            "ds_read_b64_tr_b16 %7, %11 offset:512\n\t"
+            "ds_read_b64_tr_b16 %7, %11 offset:128\n\t"
             "s_waitcnt lgkmcnt(0)"
             : "=&v"(fr[0]), "=&v"(fr[1]), "=&v"(fr[2]), "=&v"(fr[3]),
               "=&v"(fr[4]), "=&v"(fr[5]), "=&v"(fr[6]), "=&v"(fr[7])
-            : "v"(bbase + poff), "v"(abase[0] + poff),
-              "v"(abase[1] + poff), "v"(abase[2] + poff)
+            : "v"(bbase + poffB), "v"(abase[0] + poffA),
+              "v"(abase[1] + poffA), "v"(abase[2] + poffA)
             : "memory");
         __builtin_amdgcn_sched_barrier(0);
         union { struct { bf16x4 lo, hi; } p; bf16x8 v; } b;
@@ -525,6 +534,26 @@ __global__ void k_wrw2_dump(const bf16_t* __restrict__ in,
     dump[i] = l16[i];
 }
 
+
+// probe 2: SCATTERED per-lane addresses. If the transpose read gathers
+// each lane's own 8B quarter (out[g][j] = mem[addr[4j+(g>>2)] + 2*(g&3)])
+// the wrw4 staging can drop its interleaved-image duplication. in: 512
+// bf16 of content; addr_off[l] (elements) programs lane l's address.
+__global__ void k_tr16_probe2(const bf16_t* __restrict__ in,
+                              const int* __restrict__ addr_off,
+                              bf16_t* __restrict__ out) {
+  __shared__ __attribute__((aligned(128))) bf16_t buf[512];
+  const int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 512; i += blockDim.x) buf[i] = in[i];
+  __syncthreads();
+  const unsigned a = lds_addr(buf) + (unsigned)addr_off[lane] * 2;
+  bf16x4 d;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=&v"(d) : "v"(a) : "memory");
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = d[j];
+}
+
 extern "C" {
 
 int geops_conv5_wrw16_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
@@ -549,7 +578,8 @@ int geops_conv5_wrw16_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
 int geops_conv5_wrw4_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
                           int Nn, int Hi, int Wi, int Ho, int Wo, int CI,
                           int CO, int n_wg, hipStream_t s) {
-  if (CI != 4 || CO != 16) return -1;
+  // Wi must be even: the A staging copies 2-pixel (16B) granules
+  if (CI != 4 || CO != 16 || (Wi & 1)) return -1;
   const int W = (Wo + 31) >> 5;
   const int need = (W * 32 + 4) > Wi ? (W * 32 + 4) : Wi;
 #define W4LAUNCH(PIX_)                                                    \
@@ -566,6 +596,12 @@ int geops_conv5_wrw4_nhwc(const bf16_t* in, const bf16_t* gout, float* part,
 
 void geops_tr16_probe(const bf16_t* in, bf16_t* out, hipStream_t s) {
   hipLaunchKernelGGL(k_tr16_probe, dim3(1), dim3(64), 0, s, in, out);
+}
+
+void geops_tr16_probe2(const bf16_t* in, const int* addr_off, bf16_t* out,
+                       hipStream_t s) {
+  hipLaunchKernelGGL(k_tr16_probe2, dim3(1), dim3(64), 0, s, in, addr_off,
+                     out);
 }
 
 // dump size in bf16 elements for the (COT,PIX) variant, or -1

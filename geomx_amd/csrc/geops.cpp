@@ -74,6 +74,8 @@ int geops_conv5_wrw4_nhwc(const unsigned short*, const unsigned short*,
                           float*, int, int, int, int, int, int, int, int,
                           hipStream_t);
 void geops_tr16_probe(const unsigned short*, unsigned short*, hipStream_t);
+void geops_tr16_probe2(const unsigned short*, const int*, unsigned short*,
+                       hipStream_t);
 int geops_wrw2_dump(const unsigned short*, const unsigned short*,
                     unsigned short*, int, int, int, int, int, int,
                     hipStream_t);
@@ -476,6 +478,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         (int)Ho, (int)Wo, (int)CO, cur_stream());
     launch_check("wrw2_dump");
     return (int64_t)sz;
+  });
+  m.def("tr16_probe2", [](torch::Tensor in, torch::Tensor addr,
+                          torch::Tensor out) {
+    TORCH_CHECK(in.is_cuda() && addr.is_cuda() && out.is_cuda());
+    TORCH_CHECK(in.numel() == 512 && addr.numel() == 64 &&
+                out.numel() == 256);
+    TORCH_CHECK(addr.scalar_type() == torch::kInt32);
+    geops_tr16_probe2((const unsigned short*)in.data_ptr(),
+                      addr.data_ptr<int32_t>(),
+                      (unsigned short*)out.data_ptr(), cur_stream());
+    launch_check("tr16_probe2");
   });
   m.def("tr16_probe", [](torch::Tensor in, torch::Tensor out) {
     TORCH_CHECK(in.is_cuda() && out.is_cuda());
