@@ -334,6 +334,146 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_pool_nhwc(
   }
 }
 
+
+// v2 of the fused conv1 stage: the block stages its 6 input rows ONCE
+// into LDS (16B LDS-DMA granules, straight [pix][4] copy) and the 4
+// waves split the width tiles, reading A fragments with ds_read_b128
+// from LDS instead of re-issuing overlapping global loads per lane
+// (the v1 kernel is global-load latency bound: every lane re-reads the
+// 5x5 window of its pixel from L1/L2).
+__device__ __attribute__((aligned(64))) bf16_t g_convp_zeros[1024];
+
+template <int PIX>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv5_pool2_nhwc(
+    const bf16_t* __restrict__ in,       // [N][Hi][Wi][4]
+    const bf16_t* __restrict__ w_frags,  // [1][nK][64][8]
+    const float* __restrict__ bias,      // [16] or nullptr
+    bf16_t* __restrict__ out,            // [N][Hop][Wop][16]
+    uint8_t* __restrict__ mask,          // [N*Hop*Wop*16]
+    int Nn, int Hi, int Wi, int Ho, int Wo) {
+  constexpr int CI = 4;
+  constexpr int S = 5 * CI;             // 20
+  constexpr int Sp = (S + 7) & ~7;      // 24
+  constexpr int K = 5 * Sp;             // 120
+  constexpr int nK = (K + 31) / 32;     // 4
+  constexpr int CO = 16;
+  constexpr int AROW_BYTES = PIX * CI * 2;
+  constexpr int NCHA = (AROW_BYTES + 1023) / 1024;
+  constexpr int ARPB = NCHA * 1024 + 64;  // bank skew per row
+  constexpr int NROW = 6;                 // 2 conv rows + 4 halo
+  __shared__ __attribute__((aligned(128))) char lds_rows[NROW * ARPB];
+
+  const int lane = threadIdx.x & 63;
+  const int p = lane & 15;
+  const int q = lane >> 4;
+  const int m = lane & 15;
+  const int wid = threadIdx.x >> 6;
+
+  bf16x8 breg[nK];
+#pragma unroll
+  for (int i = 0; i < nK; ++i)
+    breg[i] = *reinterpret_cast<const bf16x8*>(w_frags + (i * 64 + lane) * 8);
+  const float bias_v = bias ? bias[m] : 0.0f;
+
+  const int Hop = Ho >> 1, Wop = Wo >> 1;
+  const int tiles_w = (Wo + 15) >> 4;
+  const long long n_blocks = (long long)Nn * Hop;
+  const int wg = blockIdx.x;
+  const int n_wg = gridDim.x;
+  const unsigned lds0 =
+      (unsigned)(uintptr_t)(__attribute__((address_space(3))) char*)lds_rows;
+
+  for (long long blk = wg; blk < n_blocks; blk += n_wg) {
+    const int hp = (int)(blk % Hop);
+    const long long n = blk / Hop;
+    const int ho0 = hp * 2;
+    // ---- stage the 6 input rows
+    for (int t = wid; t < NROW * NCHA; t += 4) {
+      const int ir = t / NCHA;
+      const int c = t - ir * NCHA;
+      const int slot = c * 64 + lane;      // 16B = 2 pixels
+      const int pix = slot * 2;
+      const bf16_t* src =
+          (pix + 1 < Wi) ? in + ((n * Hi + (ho0 + ir)) * (long long)Wi * CI +
+                                 (long long)pix * CI)
+                         : g_convp_zeros;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lds_rows + ir * ARPB +
+                                                    c * 1024),
+          16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    const long long out_row =
+        ((long long)n * Hop + hp) * (long long)Wop * CO;
+    for (int tw = wid; tw < tiles_w; tw += 4) {
+      const int wo_raw = (tw << 4) + p;
+      const int wo = wo_raw < Wo ? wo_raw : (Wo - 1);
+      f32x4 a0 = (f32x4)0.0f, a1 = (f32x4)0.0f;
+#pragma unroll
+      for (int km = 0; km < nK; ++km) {
+        const int k0 = km * 32 + q * 8;
+        const int kh = k0 / Sp;
+        const int j0 = k0 % Sp;
+        bf16x8 f0 = (bf16x8)0, f1 = (bf16x8)0;
+        if (k0 < K && j0 < S) {
+          // contiguous within the row span; slots past S hit the next
+          // pixel's data, zeroed by the Sp padding in w_frags. Odd wo
+          // makes the address 8B-aligned only: two b64 reads.
+          const unsigned off = (unsigned)(wo * (CI * 2) + j0 * 2);
+          typedef __attribute__((ext_vector_type(2))) unsigned uu2;
+          union { struct { uu2 lo, hi; } p; bf16x8 h; } c0, c1;
+          c0.p.lo = *(const __attribute__((address_space(3)))
+                      uu2*)(uintptr_t)(lds0 + kh * ARPB + off);
+          c0.p.hi = *(const __attribute__((address_space(3)))
+                      uu2*)(uintptr_t)(lds0 + kh * ARPB + off + 8);
+          c1.p.lo = *(const __attribute__((address_space(3)))
+                      uu2*)(uintptr_t)(lds0 + (kh + 1) * ARPB + off);
+          c1.p.hi = *(const __attribute__((address_space(3)))
+                      uu2*)(uintptr_t)(lds0 + (kh + 1) * ARPB + off + 8);
+          f0 = c0.h;
+          f1 = c1.h;
+        }
+        a0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f0, breg[km], a0,
+                                                     0, 0, 0);
+        a1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f1, breg[km], a1,
+                                                     0, 0, 0);
+      }
+      const int wo0 = tw << 4;
+#pragma unroll
+      for (int pi = 0; pi < 2; ++pi) {
+        const int i0 = pi * 2;
+        const int wp = (wo0 + q * 4 + i0) >> 1;
+        if (wp >= Wop) continue;
+        const float q0 = a0[i0] + bias_v;
+        const float q1 = a0[i0 + 1] + bias_v;
+        const float q2 = a1[i0] + bias_v;
+        const float q3 = a1[i0 + 1] + bias_v;
+        float mx = q0;
+        int arg = 0;
+        if (q1 > mx) { mx = q1; arg = 1; }
+        if (q2 > mx) { mx = q2; arg = 2; }
+        if (q3 > mx) { mx = q3; arg = 3; }
+        bf16_t ov;
+        uint8_t code;
+        if (mx <= 0.0f) {
+          ov = (bf16_t)0;
+          code = 255;
+        } else {
+          ov = cf2bf(mx);
+          code = (uint8_t)arg;
+        }
+        const long long o_off = out_row + (long long)wp * CO + m;
+        out[o_off] = ov;
+        mask[o_off] = code;
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // NHWC channel pad: [N,H,W,3] (fp32 or bf16) -> [N,H,W,4] bf16 with a
 // zero 4th channel. One thread per OUTPUT pixel: reads 3 elems, writes
 // one 8-byte bf16x4.
@@ -400,6 +540,16 @@ int geops_conv5_pool_nhwc(const bf16_t* in, const bf16_t* w_frags,
   if (CO != 16 || (Ho & 1) || (Wo & 1)) return -1;
   const long long rows = (long long)Nn * (Ho >> 1);
   const dim3 grid(conv_blocks(rows)), block(CONV_THREADS);
+  if (CI == 4 && !(Wi & 1)) {
+    const int W16 = (Wo + 15) >> 4;
+    const int need = W16 * 16 + 4 > Wi ? W16 * 16 + 4 : Wi;
+    int n_wg = (int)((rows < 2048) ? rows : 2048);
+    if (need <= 232) {
+      hipLaunchKernelGGL((k_conv5_pool2_nhwc<232>), dim3(n_wg), block, 0, s,
+                         in, w_frags, bias, out, mask, Nn, Hi, Wi, Ho, Wo);
+      return 0;
+    }
+  }
   if (CI == 4) {
     hipLaunchKernelGGL((k_conv5_pool_nhwc<4>), grid, block, 0, s, in,
                        w_frags, bias, out, mask, Nn, Hi, Wi, Ho, Wo);
