@@ -474,6 +474,152 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_pool2_nhwc(
   }
 }
 
+
+// CI=16 fused conv+bias+relu+pool (conv2 stage, CO = COT*16): like the
+// CI=4 v2 but the weight fragments live in LDS (26.6 KB for 16->32),
+// staged ONCE before the persistent block loop (w_frags are
+// block-invariant), and A-fragment reads are 16B-aligned b128.
+template <int COT, int PIX>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv5_pool16_nhwc(
+    const bf16_t* __restrict__ in,       // [N][Hi][Wi][16]
+    const bf16_t* __restrict__ w_frags,  // [COT][nK][64][8]
+    const float* __restrict__ bias,      // [CO] or nullptr
+    bf16_t* __restrict__ out,            // [N][Hop][Wop][CO]
+    uint8_t* __restrict__ mask,          // [N*Hop*Wop*CO]
+    int Nn, int Hi, int Wi, int Ho, int Wo) {
+  constexpr int CI = 16;
+  constexpr int S = 5 * CI;             // 80
+  constexpr int Sp = S;                 // already a multiple of 8
+  constexpr int K = 5 * Sp;             // 400
+  constexpr int nK = (K + 31) / 32;     // 13
+  constexpr int CO = COT * 16;
+  constexpr int AROW_BYTES = PIX * CI * 2;
+  constexpr int NCHA = (AROW_BYTES + 1023) / 1024;
+  constexpr int ARPB = NCHA * 1024 + 64;
+  constexpr int NROW = 6;
+  constexpr int BBYTES = COT * nK * 64 * 16;
+  __shared__ __attribute__((aligned(128))) char lds_all[NROW * ARPB +
+                                                        BBYTES];
+  constexpr int BOFF = NROW * ARPB;
+
+  const int lane = threadIdx.x & 63;
+  const int p = lane & 15;
+  const int q = lane >> 4;
+  const int m = lane & 15;
+  const int wid = threadIdx.x >> 6;
+
+  // stage the weight panel once (block-invariant across the loop)
+  for (int t = threadIdx.x; t < COT * nK * 64; t += blockDim.x) {
+    reinterpret_cast<uint4*>(lds_all + BOFF)[t] =
+        reinterpret_cast<const uint4*>(w_frags)[t];
+  }
+  float bias_v[COT];
+#pragma unroll
+  for (int ct = 0; ct < COT; ++ct)
+    bias_v[ct] = bias ? bias[ct * 16 + m] : 0.0f;
+
+  const int Hop = Ho >> 1, Wop = Wo >> 1;
+  const int tiles_w = (Wo + 15) >> 4;
+  const long long n_blocks = (long long)Nn * Hop;
+  const int wg = blockIdx.x;
+  const int n_wg = gridDim.x;
+  const unsigned lds0 =
+      (unsigned)(uintptr_t)(__attribute__((address_space(3))) char*)lds_all;
+  const bf16x8* lds_bv =
+      reinterpret_cast<const bf16x8*>(lds_all + BOFF) + lane;
+
+  for (long long blk = wg; blk < n_blocks; blk += n_wg) {
+    const int hp = (int)(blk % Hop);
+    const long long n = blk / Hop;
+    const int ho0 = hp * 2;
+    for (int t = wid; t < NROW * NCHA; t += 4) {
+      const int ir = t / NCHA;
+      const int c = t - ir * NCHA;
+      const int slot = c * 64 + lane;      // 16B = half a pixel
+      const int pix = slot >> 1;
+      const bf16_t* src =
+          (pix < Wi) ? in + ((n * Hi + (ho0 + ir)) * (long long)Wi * CI +
+                             (long long)slot * 8)
+                     : g_convp_zeros;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lds_all + ir * ARPB +
+                                                    c * 1024),
+          16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    const long long out_row =
+        ((long long)n * Hop + hp) * (long long)Wop * CO;
+    for (int tw = wid; tw < tiles_w; tw += 4) {
+      const int wo_raw = (tw << 4) + p;
+      const int wo = wo_raw < Wo ? wo_raw : (Wo - 1);
+      f32x4 a0[COT], a1[COT];
+#pragma unroll
+      for (int ct = 0; ct < COT; ++ct) {
+        a0[ct] = (f32x4)0.0f;
+        a1[ct] = (f32x4)0.0f;
+      }
+#pragma unroll 1
+      for (int km = 0; km < nK; ++km) {
+        const int k0 = km * 32 + q * 8;
+        const int kh = k0 / Sp;
+        const int j0 = k0 % Sp;
+        bf16x8 f0 = (bf16x8)0, f1 = (bf16x8)0;
+        if (k0 < K) {
+          const unsigned off = (unsigned)(wo * (CI * 2) + j0 * 2);
+          f0 = *(const __attribute__((address_space(3)))
+                 bf16x8*)(uintptr_t)(lds0 + kh * ARPB + off);
+          f1 = *(const __attribute__((address_space(3)))
+                 bf16x8*)(uintptr_t)(lds0 + (kh + 1) * ARPB + off);
+        }
+#pragma unroll
+        for (int ct = 0; ct < COT; ++ct) {
+          const bf16x8 bf = lds_bv[(ct * nK + km) * 64];
+          a0[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f0, bf, a0[ct],
+                                                           0, 0, 0);
+          a1[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f1, bf, a1[ct],
+                                                           0, 0, 0);
+        }
+      }
+      const int wo0 = tw << 4;
+#pragma unroll
+      for (int ct = 0; ct < COT; ++ct) {
+#pragma unroll
+        for (int pi = 0; pi < 2; ++pi) {
+          const int i0 = pi * 2;
+          const int wp = (wo0 + q * 4 + i0) >> 1;
+          if (wp >= Wop) continue;
+          const float q0 = a0[ct][i0] + bias_v[ct];
+          const float q1 = a0[ct][i0 + 1] + bias_v[ct];
+          const float q2 = a1[ct][i0] + bias_v[ct];
+          const float q3 = a1[ct][i0 + 1] + bias_v[ct];
+          float mx = q0;
+          int arg = 0;
+          if (q1 > mx) { mx = q1; arg = 1; }
+          if (q2 > mx) { mx = q2; arg = 2; }
+          if (q3 > mx) { mx = q3; arg = 3; }
+          bf16_t ov;
+          uint8_t code;
+          if (mx <= 0.0f) {
+            ov = (bf16_t)0;
+            code = 255;
+          } else {
+            ov = cf2bf(mx);
+            code = (uint8_t)arg;
+          }
+          const long long o_off =
+              out_row + (long long)wp * CO + ct * 16 + m;
+          out[o_off] = ov;
+          mask[o_off] = code;
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // NHWC channel pad: [N,H,W,3] (fp32 or bf16) -> [N,H,W,4] bf16 with a
 // zero 4th channel. One thread per OUTPUT pixel: reads 3 elems, writes
 // one 8-byte bf16x4.
@@ -540,6 +686,24 @@ int geops_conv5_pool_nhwc(const bf16_t* in, const bf16_t* w_frags,
   if (CO != 16 || (Ho & 1) || (Wo & 1)) return -1;
   const long long rows = (long long)Nn * (Ho >> 1);
   const dim3 grid(conv_blocks(rows)), block(CONV_THREADS);
+  if (CI == 16) {
+    const int W16 = (Wo + 15) >> 4;
+    const int need = W16 * 16 + 4 > Wi ? W16 * 16 + 4 : Wi;
+    int n_wg = (int)((rows < 2048) ? rows : 2048);
+    if (CO == 32 && need <= 120) {
+      hipLaunchKernelGGL((k_conv5_pool16_nhwc<2, 120>), dim3(n_wg), block,
+                         0, s, in, w_frags, bias, out, mask, Nn, Hi, Wi,
+                         Ho, Wo);
+      return 0;
+    }
+    if (CO == 16 && need <= 120) {
+      hipLaunchKernelGGL((k_conv5_pool16_nhwc<1, 120>), dim3(n_wg), block,
+                         0, s, in, w_frags, bias, out, mask, Nn, Hi, Wi,
+                         Ho, Wo);
+      return 0;
+    }
+    return -1;
+  }
   if (CI == 4 && !(Wi & 1)) {
     const int W16 = (Wo + 15) >> 4;
     const int need = W16 * 16 + 4 > Wi ? W16 * 16 + 4 : Wi;

@@ -19,14 +19,12 @@ class GeoCNN(nn.Module):
     def __init__(self, in_channels: int = 3, image_size: int = 224,
                  num_classes: int = 10):
         super().__init__()
-        from ..ops.conv import GeoConv5, GeoConv5Pool
-        from ..ops.fused import FusedReLUPool2
+        from ..ops.conv import GeoConv5Pool
         self.features = nn.Sequential(
-            # conv1 stage: conv+bias+relu+maxpool in ONE gfx950 kernel
-            # (the full-res conv output never touches HBM)
+            # both conv stages: conv+bias+relu+maxpool in ONE gfx950
+            # kernel each (no full-resolution conv output in HBM)
             GeoConv5Pool(in_channels, 16),
-            GeoConv5(16, 32),
-            FusedReLUPool2(),
+            GeoConv5Pool(16, 32),
         )
         with torch.no_grad():
             probe = torch.zeros(1, in_channels, image_size, image_size)

@@ -410,18 +410,27 @@ class _Conv5PoolFn(torch.autograd.Function):
         _geops.relu_maxpool2_bwd(gp, mask, go, N, CO, Ho, Wo)
         grad_x = None
         if ctx.needs_input_grad[0]:
-            COp = (CIr + 15) & ~15
-            wb = weight.detach().to(torch.bfloat16).reshape(-1)
-            wz = torch.cat([wb, wb.new_zeros(1)])
-            w_frags = wz[dgrad_idx].contiguous()
-            gop = F.pad(go, (4, 4, 4, 4)) \
-                .contiguous(memory_format=torch.channels_last)
-            gx = torch.empty(N, COp, Hi, Wi, dtype=torch.bfloat16,
-                             device=xb.device,
-                             memory_format=torch.channels_last)
-            _geops.conv5_nhwc(gop, w_frags, torch.Tensor(), gx, N, Ho + 8,
-                              Wo + 8, Hi, Wi, CO, COp, 0)
-            grad_x = gx[:, :CIr] if COp != CIr else gx
+            if CI == 4:
+                COp = (CIr + 15) & ~15
+                wb = weight.detach().to(torch.bfloat16).reshape(-1)
+                wz = torch.cat([wb, wb.new_zeros(1)])
+                w_frags = wz[dgrad_idx].contiguous()
+                gop = F.pad(go, (4, 4, 4, 4)) \
+                    .contiguous(memory_format=torch.channels_last)
+                gx = torch.empty(N, COp, Hi, Wi, dtype=torch.bfloat16,
+                                 device=xb.device,
+                                 memory_format=torch.channels_last)
+                _geops.conv5_nhwc(gop, w_frags, torch.Tensor(), gx, N,
+                                  Ho + 8, Wo + 8, Hi, Wi, CO, COp, 0)
+                grad_x = gx[:, :CIr] if COp != CIr else gx
+            else:
+                # conv2-class dgrad: MIOpen's igemm beats the custom
+                # direct kernel here (0.66 vs 1.45 ms, scripts/dgrad_ab)
+                gx, _, _ = torch.ops.aten.convolution_backward(
+                    go, xb, weight.detach().to(torch.bfloat16), None,
+                    [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+                    [True, False, False])
+                grad_x = gx
         grad_w = grad_b = None
         if ctx.needs_input_grad[1] or (ctx.has_bias and
                                        ctx.needs_input_grad[2]):
@@ -438,10 +447,16 @@ class GeoConv5Pool(GeoConv5):
     on the GPU path, eager fallback elsewhere. Parameters live on this
     module exactly as on GeoConv5 (state-dict compatible)."""
 
+    _POOL_SHAPES = {(4, 16), (16, 32), (16, 16)}
+
     def _pool_eligible(self, x) -> bool:
+        CI = (self.in_channels + 3) & ~3
         Ho, Wo = x.shape[2] - 4, x.shape[3] - 4
-        return (self._eligible(x) and self.out_channels == 16
-                and Ho % 2 == 0 and Wo % 2 == 0)
+        return (x.is_cuda and native_available()
+                and (CI, self.out_channels) in self._POOL_SHAPES
+                and self.stride == (1, 1) and self.padding == (0, 0)
+                and self.kernel_size == (5, 5) and self.groups == 1
+                and Ho % 2 == 0 and Wo % 2 == 0 and x.shape[3] % 2 == 0)
 
     def forward(self, x):
         if not self._pool_eligible(x):
