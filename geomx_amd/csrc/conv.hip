@@ -683,7 +683,7 @@ int geops_conv5_pool_nhwc(const bf16_t* in, const bf16_t* w_frags,
                           const float* bias, bf16_t* out, uint8_t* mask,
                           int Nn, int Hi, int Wi, int Ho, int Wo, int CI,
                           int CO, hipStream_t s) {
-  if (CO != 16 || (Ho & 1) || (Wo & 1)) return -1;
+  if ((CO != 16 && CO != 32) || (Ho & 1) || (Wo & 1)) return -1;
   const long long rows = (long long)Nn * (Ho >> 1);
   const dim3 grid(conv_blocks(rows)), block(CONV_THREADS);
   if (CI == 16) {
