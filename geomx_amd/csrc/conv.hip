@@ -620,6 +620,142 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_pool16_nhwc(
   }
 }
 
+
+// LDS-staged DENSE conv for the conv2 data-gradient: CI=32 (the padded
+// grad-out), CO=16 (the dgrad output channels). Same structure as the
+// fused pool16 kernel — block stages its 5 input rows + the weight
+// panel lives in LDS — but writes the full-resolution output (no pool).
+template <int CIT, int COT, int PIX>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
+    const bf16_t* __restrict__ in,       // [N][Hi][Wi][CI]
+    const bf16_t* __restrict__ w_frags,  // [COT][nK][64][8]
+    const float* __restrict__ bias,      // [CO] or nullptr
+    bf16_t* __restrict__ out,            // [N][Ho][Wo][CO]
+    int Nn, int Hi, int Wi, int Ho, int Wo) {
+  constexpr int CI = CIT;
+  constexpr int S = 5 * CI;
+  constexpr int Sp = (S + 7) & ~7;
+  constexpr int K = 5 * Sp;
+  constexpr int nK = (K + 31) / 32;
+  constexpr int CO = COT * 16;
+  constexpr int AROW_BYTES = PIX * CI * 2;
+  constexpr int NCHA = (AROW_BYTES + 1023) / 1024;
+  constexpr int ARPB = NCHA * 1024 + 64;
+  constexpr int NROW = 5;                // one output row
+  constexpr int BBYTES = COT * nK * 64 * 16;
+  __shared__ __attribute__((aligned(128))) char lds_all[NROW * ARPB +
+                                                        BBYTES];
+  constexpr int BOFF = NROW * ARPB;
+
+  const int lane = threadIdx.x & 63;
+  const int p = lane & 15;
+  const int q = lane >> 4;
+  const int m = lane & 15;
+  const int wid = threadIdx.x >> 6;
+
+  for (int t = threadIdx.x; t < COT * nK * 64; t += blockDim.x) {
+    reinterpret_cast<uint4*>(lds_all + BOFF)[t] =
+        reinterpret_cast<const uint4*>(w_frags)[t];
+  }
+  float bias_v[COT];
+#pragma unroll
+  for (int ct = 0; ct < COT; ++ct)
+    bias_v[ct] = bias ? bias[ct * 16 + m] : 0.0f;
+
+  const int tiles_w = (Wo + 15) >> 4;
+  const long long n_blocks = (long long)Nn * Ho;
+  const int wg = blockIdx.x;
+  const int n_wg = gridDim.x;
+  const unsigned lds0 =
+      (unsigned)(uintptr_t)(__attribute__((address_space(3))) char*)lds_all;
+  const bf16x8* lds_bv =
+      reinterpret_cast<const bf16x8*>(lds_all + BOFF) + lane;
+
+  for (long long blk = wg; blk < n_blocks; blk += n_wg) {
+    const int ho = (int)(blk % Ho);
+    const long long n = blk / Ho;
+    for (int t = wid; t < NROW * NCHA; t += 4) {
+      const int ir = t / NCHA;
+      const int c = t - ir * NCHA;
+      const int slot = c * 64 + lane;
+      const int pix = (slot * 16) / (CI * 2);      // 16B granule -> pixel
+      // CI=32: a pixel is exactly one 64-dword bank row, so un-swizzled
+      // fragment reads are 4-way bank conflicted; XOR the granule index
+      // with (pix>>2)&3 on the SOURCE (dest stays lane-linear, guide
+      // rule 21) and apply the same XOR at the read
+      long long sslot = slot;
+      if (CI == 32) {
+        const int sub = slot & 3;
+        sslot = (slot & ~3LL) | (sub ^ ((pix >> 2) & 3));
+      }
+      const bf16_t* src =
+          (pix < Wi) ? in + ((n * Hi + (ho + ir)) * (long long)Wi * CI +
+                             sslot * 8)
+                     : g_convp_zeros;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lds_all + ir * ARPB +
+                                                    c * 1024),
+          16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    const long long out_row = ((long long)n * Ho + ho) * (long long)Wo * CO;
+    for (int tw = wid; tw < tiles_w; tw += 4) {
+      const int wo_raw = (tw << 4) + p;
+      const int wo = wo_raw < Wo ? wo_raw : (Wo - 1);
+      f32x4 acc[COT];
+#pragma unroll
+      for (int ct = 0; ct < COT; ++ct) acc[ct] = (f32x4)0.0f;
+#pragma unroll 1
+      for (int km = 0; km < nK; ++km) {
+        const int k0 = km * 32 + q * 8;
+        const int kh = k0 / Sp;
+        const int j0 = k0 % Sp;
+        bf16x8 f = (bf16x8)0;
+        if (k0 < K && j0 < S) {
+          unsigned off;
+          if (CI == 32) {
+            // granule g = j0>>3 relative to wo; the swizzle XOR uses the
+            // granule's OWN pixel P' = wo + g/4 (matching the staging's
+            // per-absolute-pixel XOR)
+            const int g = j0 >> 3;
+            const int Pp = wo + (g >> 2);
+            off = (unsigned)(((wo * 4 + (g & ~3)) +
+                              ((g & 3) ^ ((Pp >> 2) & 3))) * 16);
+          } else {
+            off = (unsigned)(wo * (CI * 2) + j0 * 2);
+          }
+          if (S - j0 >= 8) {
+            f = *(const __attribute__((address_space(3)))
+                  bf16x8*)(uintptr_t)(lds0 + kh * ARPB + off);
+          }  // Sp>S tail impossible for CI%4==0 here (S%8==0)
+        }
+#pragma unroll
+        for (int ct = 0; ct < COT; ++ct) {
+          acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              f, lds_bv[(ct * nK + km) * 64], acc[ct], 0, 0, 0);
+        }
+      }
+      const int wo0 = tw << 4;
+      const bool full = (wo0 + 16 <= Wo);
+      const long long s_base = out_row + (long long)wo0 * CO + m;
+#pragma unroll
+      for (int ct = 0; ct < COT; ++ct) {
+        const long long sb = s_base + ct * 16;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int prow = q * 4 + i;
+          if (full || wo0 + prow < Wo)
+            out[sb + (long long)prow * CO] = cf2bf(acc[ct][i] + bias_v[ct]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // NHWC channel pad: [N,H,W,3] (fp32 or bf16) -> [N,H,W,4] bf16 with a
 // zero 4th channel. One thread per OUTPUT pixel: reads 3 elems, writes
 // one 8-byte bf16x4.
@@ -667,6 +803,24 @@ int geops_conv5_nhwc(const bf16_t* in, const bf16_t* w_frags,
   return 0;
   // pad handling is done by the caller (physical zero-padding of the
   // data-grad input); every variant is the interior-only PAD=0 kernel
+  if (pad == 0 && CI == 32 && CO == 16) {
+    const int W16 = (Wo + 15) >> 4;
+    const int need = W16 * 16 + 4 > Wi ? W16 * 16 + 4 : Wi;
+    long long prows = (long long)Nn * Ho;
+    int n_wg = (int)((prows < 4096) ? prows : 4096);
+    if (need <= 120) {
+      hipLaunchKernelGGL((k_conv5_lds_nhwc<32, 1, 120>), dim3(n_wg),
+                         dim3(CONV_THREADS), 0, s, in, w_frags, bias, out,
+                         Nn, Hi, Wi, Ho, Wo);
+      return 0;
+    }
+    if (need <= 232) {
+      hipLaunchKernelGGL((k_conv5_lds_nhwc<32, 1, 232>), dim3(n_wg),
+                         dim3(CONV_THREADS), 0, s, in, w_frags, bias, out,
+                         Nn, Hi, Wi, Ho, Wo);
+      return 0;
+    }
+  }
   if (pad == 0) {
     if (CI == 4 && CO == 16) { LAUNCH(4, 1, 0) }
     if (CI == 16 && CO == 32) { LAUNCH(16, 2, 0) }
