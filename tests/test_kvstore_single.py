@@ -446,3 +446,19 @@ def test_async_push_priority_order():
     assert order.count("a") == 2 and len(kv._pending) == 1
     kv.pull("a", out)
     assert torch.allclose(out, -3 * torch.ones(4))
+
+
+def test_geoconv5pool_cpu_fallback_equivalence():
+    """GeoConv5Pool's eager fallback must equal conv+relu+maxpool
+    exactly (the fused GPU path is covered by test_kernels_gpu)."""
+    from geomx_amd.ops.conv import GeoConv5Pool
+    torch.manual_seed(12)
+    m = GeoConv5Pool(3, 16)
+    x = torch.randn(2, 3, 36, 36)
+    y = m(x)
+    ref = torch.nn.functional.max_pool2d(
+        torch.nn.functional.relu(
+            torch.nn.functional.conv2d(x, m.weight, m.bias)), 2, 2)
+    assert torch.allclose(y, ref, atol=1e-6)
+    y.sum().backward()
+    assert m.weight.grad is not None and m.weight.grad.abs().sum() > 0
