@@ -641,7 +641,8 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
   constexpr int AROW_BYTES = PIX * CI * 2;
   constexpr int NCHA = (AROW_BYTES + 1023) / 1024;
   constexpr int ARPB = NCHA * 1024 + 64;
-  constexpr int NROW = 5;                // one output row
+  constexpr int ROWS = 2;                // output rows per block
+  constexpr int NROW = ROWS + 4;         // staged input rows
   constexpr int BBYTES = COT * nK * 64 * 16;
   __shared__ __attribute__((aligned(128))) char lds_all[NROW * ARPB +
                                                         BBYTES];
@@ -663,7 +664,8 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
     bias_v[ct] = bias ? bias[ct * 16 + m] : 0.0f;
 
   const int tiles_w = (Wo + 15) >> 4;
-  const long long n_blocks = (long long)Nn * Ho;
+  const int hblocks = (Ho + ROWS - 1) / ROWS;
+  const long long n_blocks = (long long)Nn * hblocks;
   const int wg = blockIdx.x;
   const int n_wg = gridDim.x;
   const unsigned lds0 =
@@ -672,9 +674,11 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
       reinterpret_cast<const bf16x8*>(lds_all + BOFF) + lane;
 
   for (long long blk = wg; blk < n_blocks; blk += n_wg) {
-    const int ho = (int)(blk % Ho);
-    const long long n = blk / Ho;
-    for (int t = wid; t < NROW * NCHA; t += 4) {
+    const int hb = (int)(blk % hblocks);
+    const long long n = blk / hblocks;
+    const int ho0 = hb * ROWS;
+    const int nrows = (Ho - ho0) < ROWS ? (Ho - ho0) : ROWS;
+    for (int t = wid; t < (nrows + 4) * NCHA; t += 4) {
       const int ir = t / NCHA;
       const int c = t - ir * NCHA;
       const int slot = c * 64 + lane;
@@ -689,7 +693,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
         sslot = (slot & ~3LL) | (sub ^ ((pix >> 2) & 3));
       }
       const bf16_t* src =
-          (pix < Wi) ? in + ((n * Hi + (ho + ir)) * (long long)Wi * CI +
+          (pix < Wi) ? in + ((n * Hi + (ho0 + ir)) * (long long)Wi * CI +
                              sslot * 8)
                      : g_convp_zeros;
       __builtin_amdgcn_global_load_lds(
@@ -701,7 +705,9 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
-    const long long out_row = ((long long)n * Ho + ho) * (long long)Wo * CO;
+    for (int r = 0; r < nrows; ++r) {
+    const long long out_row =
+        ((long long)n * Ho + (ho0 + r)) * (long long)Wo * CO;
     for (int tw = wid; tw < tiles_w; tw += 4) {
       const int wo_raw = (tw << 4) + p;
       const int wo = wo_raw < Wo ? wo_raw : (Wo - 1);
@@ -729,7 +735,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
           }
           if (S - j0 >= 8) {
             f = *(const __attribute__((address_space(3)))
-                  bf16x8*)(uintptr_t)(lds0 + kh * ARPB + off);
+                  bf16x8*)(uintptr_t)(lds0 + (r + kh) * ARPB + off);
           }  // Sp>S tail impossible for CI%4==0 here (S%8==0)
         }
 #pragma unroll
@@ -751,6 +757,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
             out[sb + (long long)prow * CO] = cf2bf(acc[ct][i] + bias_v[ct]);
         }
       }
+    }
     }
     __syncthreads();
   }
@@ -806,7 +813,7 @@ int geops_conv5_nhwc(const bf16_t* in, const bf16_t* w_frags,
   if (pad == 0 && CI == 32 && CO == 16) {
     const int W16 = (Wo + 15) >> 4;
     const int need = W16 * 16 + 4 > Wi ? W16 * 16 + 4 : Wi;
-    long long prows = (long long)Nn * Ho;
+    long long prows = (long long)Nn * ((Ho + 1) / 2);
     int n_wg = (int)((prows < 4096) ? prows : 4096);
     if (need <= 120) {
       hipLaunchKernelGGL((k_conv5_lds_nhwc<32, 1, 120>), dim3(n_wg),
