@@ -705,27 +705,26 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
-    for (int r = 0; r < nrows; ++r) {
-    const long long out_row =
-        ((long long)n * Ho + (ho0 + r)) * (long long)Wo * CO;
+    const long long out_row0 =
+        ((long long)n * Ho + ho0) * (long long)Wo * CO;
     for (int tw = wid; tw < tiles_w; tw += 4) {
       const int wo_raw = (tw << 4) + p;
       const int wo = wo_raw < Wo ? wo_raw : (Wo - 1);
-      f32x4 acc[COT];
+      f32x4 a0[COT], a1[COT];
 #pragma unroll
-      for (int ct = 0; ct < COT; ++ct) acc[ct] = (f32x4)0.0f;
+      for (int ct = 0; ct < COT; ++ct) {
+        a0[ct] = (f32x4)0.0f;
+        a1[ct] = (f32x4)0.0f;
+      }
 #pragma unroll 1
       for (int km = 0; km < nK; ++km) {
         const int k0 = km * 32 + q * 8;
         const int kh = k0 / Sp;
         const int j0 = k0 % Sp;
-        bf16x8 f = (bf16x8)0;
+        bf16x8 f0 = (bf16x8)0, f1 = (bf16x8)0;
         if (k0 < K && j0 < S) {
           unsigned off;
           if (CI == 32) {
-            // granule g = j0>>3 relative to wo; the swizzle XOR uses the
-            // granule's OWN pixel P' = wo + g/4 (matching the staging's
-            // per-absolute-pixel XOR)
             const int g = j0 >> 3;
             const int Pp = wo + (g >> 2);
             off = (unsigned)(((wo * 4 + (g & ~3)) +
@@ -734,30 +733,38 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
             off = (unsigned)(wo * (CI * 2) + j0 * 2);
           }
           if (S - j0 >= 8) {
-            f = *(const __attribute__((address_space(3)))
-                  bf16x8*)(uintptr_t)(lds0 + (r + kh) * ARPB + off);
-          }  // Sp>S tail impossible for CI%4==0 here (S%8==0)
+            f0 = *(const __attribute__((address_space(3)))
+                   bf16x8*)(uintptr_t)(lds0 + kh * ARPB + off);
+            f1 = *(const __attribute__((address_space(3)))
+                   bf16x8*)(uintptr_t)(lds0 + (kh + 1) * ARPB + off);
+          }
         }
 #pragma unroll
         for (int ct = 0; ct < COT; ++ct) {
-          acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              f, lds_bv[(ct * nK + km) * 64], acc[ct], 0, 0, 0);
+          const bf16x8 bf = lds_bv[(ct * nK + km) * 64];
+          a0[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f0, bf, a0[ct],
+                                                           0, 0, 0);
+          a1[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f1, bf, a1[ct],
+                                                           0, 0, 0);
         }
       }
       const int wo0 = tw << 4;
       const bool full = (wo0 + 16 <= Wo);
-      const long long s_base = out_row + (long long)wo0 * CO + m;
+      const long long s_base = out_row0 + (long long)wo0 * CO + m;
 #pragma unroll
       for (int ct = 0; ct < COT; ++ct) {
         const long long sb = s_base + ct * 16;
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int prow = q * 4 + i;
-          if (full || wo0 + prow < Wo)
-            out[sb + (long long)prow * CO] = cf2bf(acc[ct][i] + bias_v[ct]);
+          if (full || wo0 + prow < Wo) {
+            out[sb + (long long)prow * CO] = cf2bf(a0[ct][i] + bias_v[ct]);
+            if (nrows > 1)
+              out[sb + (long long)Wo * CO + (long long)prow * CO] =
+                  cf2bf(a1[ct][i] + bias_v[ct]);
+          }
         }
       }
-    }
     }
     __syncthreads();
   }
