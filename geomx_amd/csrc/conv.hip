@@ -625,13 +625,16 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_pool16_nhwc(
 // grad-out), CO=16 (the dgrad output channels). Same structure as the
 // fused pool16 kernel — block stages its 5 input rows + the weight
 // panel lives in LDS — but writes the full-resolution output (no pool).
-template <int CIT, int COT, int PIX>
+template <int CIT, int COT, int PIX, int PAD = 0>
 __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
-    const bf16_t* __restrict__ in,       // [N][Hi][Wi][CI]
+    const bf16_t* __restrict__ in,   // [N][Hi-2P][Wi-2P][CI] (P virtual)
     const bf16_t* __restrict__ w_frags,  // [COT][nK][64][8]
     const float* __restrict__ bias,      // [CO] or nullptr
     bf16_t* __restrict__ out,            // [N][Ho][Wo][CO]
     int Nn, int Hi, int Wi, int Ho, int Wo) {
+  // PAD>0: `in` is the UNPADDED tensor; the staging pass materialises
+  // the zero border directly in LDS (granules map within one pixel for
+  // CI>=8, so the border test is per-granule)
   constexpr int CI = CIT;
   constexpr int S = 5 * CI;
   constexpr int Sp = (S + 7) & ~7;
@@ -692,10 +695,22 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv5_lds_nhwc(
         const int sub = slot & 3;
         sslot = (slot & ~3LL) | (sub ^ ((pix >> 2) & 3));
       }
-      const bf16_t* src =
-          (pix < Wi) ? in + ((n * Hi + (ho0 + ir)) * (long long)Wi * CI +
-                             sslot * 8)
-                     : g_convp_zeros;
+      const bf16_t* src;
+      if (PAD == 0) {
+        src = (pix < Wi) ? in + ((n * Hi + (ho0 + ir)) * (long long)Wi * CI +
+                               sslot * 8)
+                         : g_convp_zeros;
+      } else {
+        const int rrow = ho0 + ir - PAD;
+        const int rpix = pix - PAD;
+        const int rWi = Wi - 2 * PAD;
+        const bool ok = rrow >= 0 && rrow < (Hi - 2 * PAD) && rpix >= 0 &&
+                        rpix < rWi;
+        // sslot relative to the real pixel: subtract the border columns
+        src = ok ? in + ((n * (Hi - 2 * PAD) + rrow) * (long long)rWi * CI +
+                         (sslot - (long long)PAD * (CI / 8)) * 8)
+                 : g_convp_zeros;
+      }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(lds_all + ir * ARPB +
@@ -817,11 +832,23 @@ int geops_conv5_nhwc(const bf16_t* in, const bf16_t* w_frags,
   return 0;
   // pad handling is done by the caller (physical zero-padding of the
   // data-grad input); every variant is the interior-only PAD=0 kernel
-  if (pad == 0 && CI == 32 && CO == 16) {
+  if ((pad == 0 || pad == 4) && CI == 32 && CO == 16) {
     const int W16 = (Wo + 15) >> 4;
     const int need = W16 * 16 + 4 > Wi ? W16 * 16 + 4 : Wi;
     long long prows = (long long)Nn * ((Ho + 1) / 2);
     int n_wg = (int)((prows < 4096) ? prows : 4096);
+    if (pad == 4 && need <= 120) {
+      hipLaunchKernelGGL((k_conv5_lds_nhwc<32, 1, 120, 4>), dim3(n_wg),
+                         dim3(CONV_THREADS), 0, s, in, w_frags, bias, out,
+                         Nn, Hi, Wi, Ho, Wo);
+      return 0;
+    }
+    if (pad == 4 && need <= 232) {
+      hipLaunchKernelGGL((k_conv5_lds_nhwc<32, 1, 232, 4>), dim3(n_wg),
+                         dim3(CONV_THREADS), 0, s, in, w_frags, bias, out,
+                         Nn, Hi, Wi, Ho, Wo);
+      return 0;
+    }
     if (need <= 120) {
       hipLaunchKernelGGL((k_conv5_lds_nhwc<32, 1, 120>), dim3(n_wg),
                          dim3(CONV_THREADS), 0, s, in, w_frags, bias, out,
