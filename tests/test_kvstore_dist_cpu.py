@@ -457,3 +457,39 @@ def _row_sparse_sharded_dense(rank, world):
 
 def test_row_sparse_pull_sharded_ws2():
     run_dist(2, _row_sparse_sharded_dense)
+
+
+# ---------------------------------------------------------------------------
+# Async push: deferred WAN tier must flush in the SAME order on every
+# rank (collective matching), even when ranks pull in different orders
+# ---------------------------------------------------------------------------
+
+def _async_push_flush_order(rank, world):
+    kv = _mk(num_parties=2)
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    torch.manual_seed(0)
+    keys = ["a", "b", "c", "d", "e"]
+    for k in keys:
+        kv.init(k, torch.full((8,), 1.0))
+    # same push order everywhere (SPMD contract), varied priorities
+    for i, k in enumerate(keys):
+        kv.push(k, torch.full((8,), float(rank + 1)), priority=-i)
+    assert len(kv._pending) == len(keys)
+    # each rank pulls in a DIFFERENT order: the first pull flushes ALL
+    # pending keys in deterministic (priority, seq) order, so the
+    # deferred collectives still match across ranks
+    order = keys[rank:] + keys[:rank]
+    outs = {}
+    for k in order:
+        o = torch.empty(8)
+        kv.pull(k, o)
+        outs[k] = o
+    total = sum(r + 1 for r in range(world))
+    for k in keys:
+        expect = 1.0 - 0.1 * total
+        assert torch.allclose(outs[k], torch.full((8,), expect),
+                              atol=1e-5), (k, outs[k][0])
+
+
+def test_async_push_flush_order_ws4():
+    run_dist(4, _async_push_flush_order)
