@@ -37,6 +37,17 @@ Synchronization modes:
   * HFA — leaders store party aggregates locally and only every K2-th
     push syncs globally, transmitting the milestone DELTA
     (stored - milestone)/P (kvstore_dist_server.h:959-972,1324-1343).
+
+SPMD ordering contract: the reference is message-passing, so workers
+may push/pull keys in any order; this design rides collectives, so
+every rank must issue the SAME sequence of key operations. push() is
+exempt in effect — it defers its WAN tier, and the flush applies every
+pending key in deterministic (priority, seq) order at the next
+state-reading call, so ranks may interleave OTHER work freely between
+pushes. pull() order must match across ranks whenever the pull has a
+wire (sharded dense mode); in "replicated" mode (and for compressed
+exchanges, where every leader already holds the result) pulls are
+wire-free and order-independent (test_async_push_flush_order_ws2).
 """
 
 from __future__ import annotations

@@ -465,20 +465,21 @@ def test_row_sparse_pull_sharded_ws2():
 # ---------------------------------------------------------------------------
 
 def _async_push_flush_order(rank, world):
+    # REPLICATED global mode: pull has no leader-tier wire, so ranks may
+    # pull in different orders — what must stay matched are the DEFERRED
+    # push collectives, which flush in deterministic (priority, seq)
+    # order at the first pull regardless of which key it names.
     kv = _mk(num_parties=2)
+    kv.global_mode = "replicated"
     kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
     torch.manual_seed(0)
     keys = ["a", "b", "c", "d", "e"]
     for k in keys:
         kv.init(k, torch.full((8,), 1.0))
-    # same push order everywhere (SPMD contract), varied priorities
     for i, k in enumerate(keys):
         kv.push(k, torch.full((8,), float(rank + 1)), priority=-i)
     assert len(kv._pending) == len(keys)
-    # each rank pulls in a DIFFERENT order: the first pull flushes ALL
-    # pending keys in deterministic (priority, seq) order, so the
-    # deferred collectives still match across ranks
-    order = keys[rank:] + keys[:rank]
+    order = keys[rank:] + keys[:rank]   # divergent pull orders
     outs = {}
     for k in order:
         o = torch.empty(8)
@@ -491,5 +492,34 @@ def _async_push_flush_order(rank, world):
                               atol=1e-5), (k, outs[k][0])
 
 
+def test_async_push_flush_order_ws2():
+    run_dist(2, _async_push_flush_order)
+
+
+def _async_push_sharded_same_order(rank, world):
+    # sharded mode keeps the SPMD contract (same pull order on every
+    # rank); interleaved repushes exercise the flush-on-repush path
+    kv = _mk(num_parties=2)
+    kv.set_optimizer(OptimizerSpec(name="sgd", lr=0.1))
+    keys = ["a", "b", "c"]
+    for k in keys:
+        kv.init(k, torch.full((4,), 1.0))
+    for i, k in enumerate(keys):
+        kv.push(k, torch.full((4,), float(rank + 1)), priority=-i)
+    kv.push("a", torch.full((4,), float(rank + 1)))  # repush -> flush
+    outs = {}
+    for k in keys:
+        o = torch.empty(4)
+        kv.pull(k, o)
+        outs[k] = o
+    total = sum(r + 1 for r in range(world))
+    assert torch.allclose(outs["a"], torch.full((4,), 1.0 - 0.2 * total),
+                          atol=1e-5)
+    for k in ("b", "c"):
+        assert torch.allclose(outs[k], torch.full((4,), 1.0 - 0.1 * total),
+                              atol=1e-5)
+
+
 def test_async_push_flush_order_ws4():
-    run_dist(4, _async_push_flush_order)
+    run_dist(4, _async_push_sharded_same_order)
+
