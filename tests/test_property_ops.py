@@ -137,3 +137,19 @@ def test_recordio_property_roundtrip(n, shape, seed, tmp_path_factory):
     for i in [0, n // 2, n - 1]:
         x, y = rd[i]
         assert torch.equal(x, ts[i]) and y == i
+
+
+@given(st.integers(8, 4000), st.sampled_from([64, 128, 256]),
+       st.floats(0.1, 0.9), st.integers(0, 3))
+@settings(max_examples=30, deadline=None)
+def test_dgt_wire_equals_transform(n, chunk, k, seed):
+    """Property: the wire form (compress->decompress) reconstructs
+    exactly what transform() applies locally, for any shape/ratio."""
+    from geomx_amd.kvstore.dgt import DGTState
+    torch.manual_seed(seed)
+    g = torch.randn(n)
+    a = DGTState(n, "cpu", chunk_elems=chunk, k=k, mode=3)
+    b = DGTState(n, "cpu", chunk_elems=chunk, k=k, mode=3)
+    ref, _ = a.transform(g.clone())
+    got = b.decompress(*b.compress(g.clone()))
+    assert torch.allclose(got, ref, atol=1e-5)
