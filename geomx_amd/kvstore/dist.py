@@ -297,7 +297,7 @@ class KVStoreDist(KVStoreBase):
         self.flush()  # pending pushes complete under the OLD scheme
         params = dict(compression_params)
         ctype = params.get("type")
-        if ctype not in ("2bit", "bsc", "fp16", "mpq", "dgt"):
+        if ctype not in ("2bit", "bsc", "fp16", "mpq", "dgt", "bsc_dgt"):
             raise ValueError(f"unknown compression type {ctype!r}")
         if ctype == "2bit":
             params.setdefault("threshold", self.cfg.threshold)
@@ -312,7 +312,7 @@ class KVStoreDist(KVStoreBase):
                     and self.cfg.async_transport == "store":
                 raise ValueError("2bit compression is not supported with "
                                  "async_transport='store'")
-        if ctype in ("bsc", "mpq"):
+        if ctype in ("bsc", "mpq", "bsc_dgt"):
             params.setdefault("threshold", self.cfg.bsc_ratio)
             params.setdefault("size_lower_bound", self.cfg.size_lower_bound)
         self.compression = params
@@ -567,11 +567,31 @@ class KVStoreDist(KVStoreBase):
         ratio = float(self.compression.get("threshold", self.cfg.bsc_ratio)) \
             if self.compression else self.cfg.bsc_ratio
 
-        if ctype == "bsc":
+        if ctype in ("bsc", "bsc_dgt"):
             if st.bsc_u is None:
                 st.bsc_u = torch.zeros(st.numel, device=self._device)
                 st.bsc_v = torch.zeros(st.numel, device=self._device)
             vals, idx = ops.bsc_compress(party_sum, st.bsc_u, st.bsc_v, ratio)
+            if ctype == "bsc_dgt":
+                # config-5 composition: DGT's 4-bit tier rides the packed
+                # VALUES (the reference's DGT chunks whatever bytes a
+                # push carries, kv_app.h:917-995); indices stay exact
+                dg = self._dgt_state(st, vals.device, numel=vals.numel())
+                vals = vals.masked_fill(idx < 0, 0.0)
+                dg.residual.zero_()  # packed positions shift every step
+                payload = dg.compress(vals) + (idx,)
+                gathered = [[torch.empty_like(t) for _ in range(P)]
+                            for t in payload]
+                for lst, t in zip(gathered, payload):
+                    comm.all_gather(lst, t, group=group)
+                self.wan.charge(cross_party_bytes(
+                    "all_gather", dg.wire_bytes() + idx.numel() * 4, P))
+                dense = []
+                for pi in range(P):
+                    v_p = dg.decompress(*(lst[pi] for lst in gathered[:4]))
+                    dense.append(ops.bsc_decompress(
+                        v_p.contiguous(), gathered[4][pi], st.numel))
+                return dense
             payload = vals.numel() * 4 + idx.numel() * 4
             vlist = [torch.empty_like(vals) for _ in range(P)]
             ilist = [torch.empty_like(idx) for _ in range(P)]
@@ -767,7 +787,7 @@ class KVStoreDist(KVStoreBase):
             return
         ctype = self._effective_ctype(st)
         need_wire = self.global_mode == "sharded" and not (
-            ctype in ("bsc", "fp16", "dgt", "2bit")
+            ctype in ("bsc", "fp16", "dgt", "2bit", "bsc_dgt")
             or self.cfg.mode == "dist_async")
         if not need_wire:
             return  # all leaders already hold the result (replayed update)
@@ -937,7 +957,7 @@ class KVStoreDist(KVStoreBase):
             return
         ctype = self._effective_ctype(st)
         need_wire = self.global_mode == "sharded" and not (
-            ctype in ("bsc", "fp16", "dgt", "2bit")
+            ctype in ("bsc", "fp16", "dgt", "2bit", "bsc_dgt")
             or self.cfg.mode == "dist_async")
         if not need_wire:
             return
@@ -1002,14 +1022,15 @@ class KVStoreDist(KVStoreBase):
     # ------------------------------------------------------------------
     # DGT transform (shared DGTState, kvstore/dgt.py)
     # ------------------------------------------------------------------
-    def _dgt_state(self, st: _KeyState, device):
+    def _dgt_state(self, st: _KeyState, device, numel=None):
         if not hasattr(self, "_dgt_states"):
             self._dgt_states = {}
-        key = id(st)
+        n = numel if numel is not None else st.numel
+        key = (id(st), n)
         dg = self._dgt_states.get(key)
-        if dg is None or dg.numel != st.numel:
+        if dg is None or dg.numel != n:
             from .dgt import DGTState
-            dg = DGTState(st.numel, device,
+            dg = DGTState(n, device,
                           chunk_elems=max(64, self.cfg.dgt_block_size // 4),
                           k=self.cfg.dgt_k, alpha=self.cfg.dgt_alpha,
                           mode=self.cfg.enable_dgt or 3)
