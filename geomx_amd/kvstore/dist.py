@@ -547,12 +547,18 @@ class KVStoreDist(KVStoreBase):
         if not self.compression:
             return None
         ctype = self.compression["type"]
+        bound = int(self.compression.get("size_lower_bound",
+                                         self.cfg.size_lower_bound))
         if ctype == "mpq":
             # MPQ: small tensors -> fp16, large -> bsc
             # (size gate kvstore_dist_server.h:841,879)
-            bound = int(self.compression.get("size_lower_bound",
-                                             self.cfg.size_lower_bound))
             return "fp16" if st.numel < bound else "bsc"
+        if ctype in ("bsc", "bsc_dgt") and st.numel < bound:
+            # the reference pushes small tensors PLAIN on the bsc path
+            # too (DataPushToGlobalServersBSCompressed size gate, :879);
+            # without this a tiny key (a bias) has bsc_capacity 0 and
+            # its gradient would vanish entirely
+            return None
         return ctype
 
     def _global_exchange_push(self, key, st: _KeyState,

@@ -1,12 +1,17 @@
-"""GeoConv5: 5x5 stride-1 convolution on the hand-written gfx950 MFMA
-kernel (csrc/conv.hip) for the flagship CNN's shapes.
+"""GeoConv5 / GeoConv5Pool: 5x5 stride-1 convolution stages on the
+hand-written gfx950 MFMA kernels (csrc/conv.hip, convwrw2.hip).
 
-Forward and the data gradient run on the custom implicit-GEMM kernel
-(NHWC bf16, weights pre-packed into per-lane MFMA B-fragment order).
-The weight/bias gradients go through ATen's convolution_backward by
-default; a custom wrw kernel (csrc/convwrw.hip) exists and is verified
-but disabled (see WRW_ENABLED). Anything outside the supported geometry
-falls back to F.conv2d.
+r02 state — every op of both flagship conv stages is custom:
+  * forward: fused conv+bias+ReLU+maxpool kernels (k_conv5_pool*):
+    block-shared LDS row staging, pooling in-register, only the pooled
+    output + argmax-quadrant mask reach HBM.
+  * data gradient: LDS-staged direct conv (k_conv5_lds_nhwc) with
+    in-kernel virtual padding and an XOR bank swizzle (conv2 class);
+    interior direct conv (k_conv5_nhwc) for conv1-class shapes.
+  * weight+bias gradients: ds_read_b64_tr_b16 fragment kernels
+    (convwrw2.hip) with the bias fused as an all-ones tap tile.
+Anything outside the supported geometry falls back to ATen/F.conv2d
+(same math, library kernels).
 """
 
 from __future__ import annotations

@@ -345,7 +345,21 @@ class GeoTrainer:
         ctype = self.cfg.compression
         for b in self.buckets:
             if topo.is_leader:
-                if ctype == "bsc_dgt":
+                small = b.flat.numel() < self.cfg.size_lower_bound
+                if ctype in ("bsc", "bsc_dgt", "mpq") and small:
+                    # reference size gate (kvstore_dist_server.h:879):
+                    # small buckets ship plain fp16 (mpq) / dense
+                    if ctype == "mpq":
+                        h = b.flat.to(torch.float16)
+                        comm.all_reduce(h, group=topo.leader_group)
+                        self.wan.charge(cross_party_bytes(
+                            "all_reduce", h.numel() * 2, P))
+                        b.flat.copy_(h.float())
+                    else:
+                        comm.all_reduce(b.flat, group=topo.leader_group)
+                        self.wan.charge(cross_party_bytes(
+                            "all_reduce", b.flat.numel() * 4, P))
+                elif ctype == "bsc_dgt":
                     # BASELINE config 5 composition: Bi-Sparse selects the
                     # content, DGT's 4-bit tier rides the packed VALUES
                     # (the reference's DGT chunks whatever bytes a push
@@ -387,8 +401,7 @@ class GeoTrainer:
                                            b.flat.numel(), out=acc,
                                            accumulate=True)
                     b.flat.copy_(acc)
-                elif ctype == "bsc" or (
-                        ctype == "mpq" and b.flat.numel() >= self.cfg.size_lower_bound):
+                elif ctype == "bsc" or ctype == "mpq":
                     if b.bsc_u is None:
                         b.bsc_u = torch.zeros_like(b.flat)
                         b.bsc_v = torch.zeros_like(b.flat)

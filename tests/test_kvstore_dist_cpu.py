@@ -98,7 +98,10 @@ def test_update_on_server_sharded_ws4():
 
 def _bsc_push_pull(rank, world):
     kv = _mk(num_parties=2)
-    kv.set_gradient_compression({"type": "bsc", "threshold": 0.05})
+    # size_lower_bound below n: the reference gates SMALL tensors to a
+    # plain push even on the bsc path (kvstore_dist_server.h:879)
+    kv.set_gradient_compression({"type": "bsc", "threshold": 0.05,
+                                 "size_lower_bound": 1000})
     n = 2000
     kv.init("w", torch.zeros(n))
     torch.manual_seed(7)  # identical grads on all ranks for predictability
@@ -527,7 +530,8 @@ def test_async_push_flush_order_ws4():
 
 def _bsc_dgt_kv(rank, world):
     kv = _mk(num_parties=2, dgt_block_size=256, dgt_k=0.5)
-    kv.set_gradient_compression({"type": "bsc_dgt", "threshold": 0.25})
+    kv.set_gradient_compression({"type": "bsc_dgt", "threshold": 0.25,
+                                 "size_lower_bound": 256})
     torch.manual_seed(0)
     n = 512
     kv.init("w", torch.zeros(n))
@@ -551,3 +555,20 @@ def _bsc_dgt_kv(rank, world):
 
 def test_bsc_dgt_kvstore_ws4():
     run_dist(4, _bsc_dgt_kv)
+
+
+def _bsc_small_key_gate(rank, world):
+    """Tiny keys (biases) under bsc must ship PLAIN (reference size
+    gate): before r02's fix their bsc_capacity was 0 and the gradient
+    silently vanished."""
+    kv = _mk(num_parties=2)
+    kv.set_gradient_compression({"type": "bsc", "threshold": 0.01})
+    kv.init("bias", torch.zeros(16))  # capacity would be int(16*0.01)=0
+    kv.push("bias", torch.ones(16))
+    out = torch.empty(16)
+    kv.pull("bias", out)
+    assert torch.allclose(out, torch.full((16,), float(world)), atol=1e-5)
+
+
+def test_bsc_small_key_plain_ws4():
+    run_dist(4, _bsc_small_key_gate)
