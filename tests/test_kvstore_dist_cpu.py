@@ -572,3 +572,45 @@ def _bsc_small_key_gate(rank, world):
 
 def test_bsc_small_key_plain_ws4():
     run_dist(4, _bsc_small_key_gate)
+
+
+def _bsc_dgt_learns(rank, world):
+    """The composed bsc_dgt kvstore path must LEARN, not just run:
+    update-on-worker training on separable data drops the loss."""
+    kv = _mk(num_parties=2, dgt_block_size=256, dgt_k=0.5)
+    kv.set_gradient_compression({"type": "bsc_dgt", "threshold": 0.1,
+                                 "size_lower_bound": 64})
+    torch.manual_seed(0)
+    net = torch.nn.Sequential(torch.nn.Linear(32, 64), torch.nn.ReLU(),
+                              torch.nn.Linear(64, 4))
+    params = [p for p in net.parameters()]
+    for i, p in enumerate(params):
+        kv.init(i, p.data)
+        out = torch.empty_like(p.data)
+        kv.pull(i, out)
+        with torch.no_grad():
+            p.copy_(out)
+    opt = torch.optim.SGD(net.parameters(), lr=0.05)
+    g = torch.Generator().manual_seed(rank)
+    x = torch.randn(64, 32, generator=g)
+    y = (x[:, 0] > 0).long() + 2 * (x[:, 1] > 0).long()
+    first = last = None
+    for it in range(15):
+        loss = torch.nn.functional.cross_entropy(net(x), y)
+        opt.zero_grad()
+        loss.backward()
+        for i, p in enumerate(params):
+            kv.push(i, p.grad / world, priority=-i)
+        for i, p in enumerate(params):
+            gagg = torch.empty_like(p.grad)
+            kv.pull(i, gagg, priority=-i)
+            p.grad.copy_(gagg)
+        opt.step()
+        if first is None:
+            first = float(loss)
+        last = float(loss)
+    assert last < first * 0.7, (first, last)
+
+
+def test_bsc_dgt_learns_ws2():
+    run_dist(2, _bsc_dgt_learns)
