@@ -595,7 +595,7 @@ def _bsc_dgt_learns(rank, world):
     x = torch.randn(64, 32, generator=g)
     y = (x[:, 0] > 0).long() + 2 * (x[:, 1] > 0).long()
     first = last = None
-    for it in range(15):
+    for it in range(60):
         loss = torch.nn.functional.cross_entropy(net(x), y)
         opt.zero_grad()
         loss.backward()
@@ -609,7 +609,7 @@ def _bsc_dgt_learns(rank, world):
         if first is None:
             first = float(loss)
         last = float(loss)
-    assert last < first * 0.7, (first, last)
+    assert last < first * 0.5, (first, last)
 
 
 def test_bsc_dgt_learns_ws2():
