@@ -614,3 +614,44 @@ def _bsc_dgt_learns(rank, world):
 
 def test_bsc_dgt_learns_ws2():
     run_dist(2, _bsc_dgt_learns)
+
+
+def _2bit_learns(rank, world):
+    """2bit error-feedback on BOTH tiers must learn (the residual
+    carries the quantization error across steps)."""
+    kv = _mk(num_parties=2)
+    kv.set_gradient_compression({"type": "2bit", "threshold": 0.05})
+    torch.manual_seed(0)
+    net = torch.nn.Sequential(torch.nn.Linear(32, 64), torch.nn.ReLU(),
+                              torch.nn.Linear(64, 4))
+    params = list(net.parameters())
+    for i, p in enumerate(params):
+        kv.init(i, p.data)
+        out = torch.empty_like(p.data)
+        kv.pull(i, out)
+        with torch.no_grad():
+            p.copy_(out)
+    opt = torch.optim.SGD(net.parameters(), lr=0.05)
+    g = torch.Generator().manual_seed(rank)
+    x = torch.randn(64, 32, generator=g)
+    y = (x[:, 0] > 0).long() + 2 * (x[:, 1] > 0).long()
+    first = last = None
+    for it in range(60):
+        loss = torch.nn.functional.cross_entropy(net(x), y)
+        opt.zero_grad()
+        loss.backward()
+        for i, p in enumerate(params):
+            kv.push(i, p.grad / world, priority=-i)
+        for i, p in enumerate(params):
+            gagg = torch.empty_like(p.grad)
+            kv.pull(i, gagg, priority=-i)
+            p.grad.copy_(gagg)
+        opt.step()
+        if first is None:
+            first = float(loss)
+        last = float(loss)
+    assert last < first * 0.5, (first, last)
+
+
+def test_2bit_learns_ws4():
+    run_dist(4, _2bit_learns)
